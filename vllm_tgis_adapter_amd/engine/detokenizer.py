@@ -1,0 +1,108 @@
+"""Incremental detokenization + stop-sequence scanning.
+
+Streamed text must be decoded token-by-token without re-decoding the whole
+output, while never emitting the replacement char of a half-finished UTF-8 /
+byte-level merge, and while holding back text that could be the prefix of a
+stop sequence (TGIS contract: ≤6 stop seqs of ≤240 bytes; validation.py:10-11).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+from .request import Request, RequestStatus
+
+
+class Detokenizer:
+    def __init__(self, tokenizer):
+        self.tokenizer = tokenizer
+
+    def _convert(self, ids: list[int], skip_special: bool) -> list[str]:
+        toks = self.tokenizer.convert_ids_to_tokens(ids)
+        if skip_special:
+            special = set(self.tokenizer.all_special_tokens)
+            return [t if t not in special else "" for t in toks]
+        return toks
+
+    def append_token(self, req: Request, token_id: int) -> str:
+        """Incrementally decode one new token; returns the new text fragment."""
+        skip = req.sampling_params.skip_special_tokens
+        new_tok = self._convert([token_id], skip)[0]
+        req.prev_token_texts.append(new_tok)
+
+        toks = req.prev_token_texts
+        if req.read_offset == 0 and req.prefix_offset == 0 and len(toks) > 1:
+            pass  # state already consistent
+
+        prefix_text = self.tokenizer.convert_tokens_to_string(
+            [t for t in toks[req.prefix_offset:req.read_offset] if t]
+        )
+        full_text = self.tokenizer.convert_tokens_to_string(
+            [t for t in toks[req.prefix_offset:] if t]
+        )
+        if len(full_text) > len(prefix_text) and not full_text.endswith("�"):
+            new_text = full_text[len(prefix_text):]
+            req.prefix_offset = req.read_offset
+            req.read_offset = len(toks)
+            req.output_text += new_text
+            return new_text
+        return ""
+
+
+class StopChecker:
+    """Applies max/min token limits, EOS and stop-string semantics."""
+
+    def __init__(self, max_model_len: int):
+        self.max_model_len = max_model_len
+
+    def check(self, req: Request, token_id: int, new_text: str) -> None:
+        p = req.sampling_params
+
+        # EOS (suppressed below min_tokens by the sampler; double-check here)
+        if (
+            req.eos_token_id is not None
+            and token_id == req.eos_token_id
+            and req.num_output_tokens >= p.min_tokens
+        ):
+            # eos token excluded from output text by skip_special_tokens
+            req.finish(RequestStatus.FINISHED_STOPPED, stop_reason=None)
+            return
+
+        # stop strings
+        if p.stop and req.num_output_tokens >= p.min_tokens:
+            window_start = max(0, len(req.output_text) - len(new_text) - 240)
+            window = req.output_text[window_start:]
+            best: Optional[tuple[int, str]] = None
+            for s in p.stop:
+                idx = window.find(s)
+                if idx != -1 and (best is None or idx < best[0]):
+                    best = (idx, s)
+            if best is not None:
+                idx, s = best
+                end = window_start + idx + (len(s) if p.include_stop_str_in_output else 0)
+                req.output_text = req.output_text[:end]
+                req.holdback_len = 0
+                req.finish(RequestStatus.FINISHED_STOPPED, stop_reason=s)
+                return
+            # hold back a possible stop-string prefix from streaming
+            req.holdback_len = _longest_stop_prefix(req.output_text, p.stop)
+
+        # length limits
+        if p.max_tokens is not None and req.num_output_tokens >= p.max_tokens:
+            req.finish(RequestStatus.FINISHED_LENGTH)
+            return
+        if req.num_tokens >= self.max_model_len:
+            req.finish(RequestStatus.FINISHED_LENGTH)
+            return
+
+
+def _longest_stop_prefix(text: str, stops: list[str]) -> int:
+    """Length of the longest proper prefix of any stop string that is a
+    suffix of ``text`` (the bytes we must not stream yet)."""
+    best = 0
+    for s in stops:
+        for plen in range(min(len(s) - 1, len(text)), 0, -1):
+            if text.endswith(s[:plen]):
+                best = max(best, plen)
+                break
+    return best

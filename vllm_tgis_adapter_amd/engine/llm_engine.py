@@ -1,0 +1,126 @@
+"""Synchronous engine: scheduler + worker + postprocessing per step.
+
+This is the layer the reference delegates to vLLM for (EngineClient's
+engine side); here it is built natively: continuous batching (E1), paged KV
+(E5), incremental detokenization + stop handling (E10), per-step sampling
+(E7/E8) and request metrics (E21).
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+from ..parallel import get_tp_rank, init_distributed
+from .block_manager import BlockManager
+from .config import EngineConfig
+from .detokenizer import Detokenizer, StopChecker
+from .request import Request, RequestStatus
+from .scheduler import Scheduler
+from .tokenizer import get_tokenizer
+from .types import LoRARequest, RequestOutput, SamplingParams
+from .worker import Worker
+
+
+class LLMEngine:
+    def __init__(self, config: EngineConfig):
+        self.config = config
+        self.model_config = config.model_config
+        device = config.resolve_device()
+        init_distributed(config.tensor_parallel_size, device=device)
+        self.rank = get_tp_rank()
+
+        self.tokenizer = get_tokenizer(self.model_config)
+        self.worker = Worker(config)
+        num_blocks = self.worker.init_kv_cache()
+        self.block_manager = BlockManager(num_blocks, config.cache_config.block_size)
+        self.scheduler = Scheduler(config.scheduler_config, self.block_manager)
+        self.detokenizer = Detokenizer(self.tokenizer)
+        self.stop_checker = StopChecker(self.model_config.max_model_len)
+        self.eos_token_id = self.tokenizer.eos_token_id
+
+    # ------------------------------------------------------------------
+    def add_request(
+        self,
+        request_id: str,
+        prompt: Optional[str],
+        prompt_token_ids: list[int],
+        sampling_params: SamplingParams,
+        arrival_time: Optional[float] = None,
+        lora_request: Optional[LoRARequest] = None,
+        trace_headers: Optional[dict] = None,
+    ) -> Request:
+        req = Request(
+            request_id=request_id,
+            prompt=prompt,
+            prompt_token_ids=prompt_token_ids,
+            sampling_params=sampling_params,
+            arrival_time=arrival_time,
+            lora_request=lora_request,
+            trace_headers=trace_headers,
+        )
+        req.eos_token_id = self.eos_token_id
+        if sampling_params.structured_outputs is not None:
+            from .guided import build_guided_state
+
+            req.guided_state = build_guided_state(
+                sampling_params.structured_outputs, self.tokenizer
+            )
+        self.scheduler.add_request(req)
+        return req
+
+    def abort_request(self, request_id: str) -> Optional[RequestOutput]:
+        req = self.scheduler.abort_request(request_id)
+        if req is None:
+            return None
+        return req.make_output()
+
+    def has_unfinished(self) -> bool:
+        return self.scheduler.has_unfinished()
+
+    # ------------------------------------------------------------------
+    def step(self) -> list[RequestOutput]:
+        sched = self.scheduler.schedule()
+        if sched.is_empty:
+            return []
+        result = self.worker.execute(sched)
+        for it in sched.items:
+            it.request.num_computed_tokens += it.num_new_tokens
+
+        now = time.time()
+        sampler_out = result.sampler_output
+        for it, token_id, lp in zip(
+            self.worker._sampling_items, sampler_out.token_ids, sampler_out.logprobs
+        ):
+            req = it.request
+            if req.status.is_finished:  # aborted mid-step
+                continue
+            if req.metrics.first_token_time is None:
+                req.metrics.first_token_time = now
+            req.metrics.last_token_time = now
+            req.output_token_ids.append(token_id)
+            if req.logprobs is not None:
+                req.logprobs.append(lp)
+                if lp and token_id in lp:
+                    req.cumulative_logprob += lp[token_id].logprob
+            if req.guided_state is not None:
+                req.guided_state.advance(token_id)
+            new_text = self.detokenizer.append_token(req, token_id)
+            self.stop_checker.check(req, token_id, new_text)
+
+        outputs: list[RequestOutput] = []
+        seen = set()
+        for it in sched.items:
+            req = it.request
+            if id(req) in seen:
+                continue
+            seen.add(id(req))
+            out = req.make_output()
+            if out is not None:
+                outputs.append(out)
+            if req.status.is_finished:
+                self.scheduler.finish_request(req)
+        return outputs
+
+    def shutdown(self) -> None:
+        self.worker.stop_workers()

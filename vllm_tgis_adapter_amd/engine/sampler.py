@@ -1,0 +1,201 @@
+"""Sampling pipeline.
+
+Per-request temperature / top-k / top-p / seeded RNG / repetition penalty /
+min-tokens EOS suppression / logits processors / logprob+rank+top-N
+extraction, vectorised over the step's sampling rows.  All tensor math runs
+on-device; the per-position logprob dicts the TGIS wire format needs
+(grpc_server.py:701-756) are assembled host-side from one batched top-k.
+
+The fused HIP sampling kernel will take over the penalty/filter/sample path;
+this module stays the orchestration + host assembly layer.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+
+from .request import Request
+from .types import Logprob, PosLogprobs
+
+_NEG_INF = float("-inf")
+
+
+@dataclass
+class SamplerOutput:
+    token_ids: list[int]
+    # per sampling row: logprob info for the sampled token (or None)
+    logprobs: list[Optional[PosLogprobs]]
+
+
+def _apply_top_k_top_p(logits: torch.Tensor, top_k: torch.Tensor, top_p: torch.Tensor) -> torch.Tensor:
+    """Mask logits outside top-k / nucleus top-p, rows vectorised."""
+    vocab = logits.shape[-1]
+    sorted_logits, sorted_idx = torch.sort(logits, dim=-1, descending=True)
+    # top-k mask
+    ranks = torch.arange(vocab, device=logits.device).unsqueeze(0)
+    k = top_k.clamp(min=1, max=vocab).unsqueeze(1)
+    mask = ranks >= k
+    # top-p mask on the sorted probabilities
+    probs = torch.softmax(sorted_logits, dim=-1)
+    cum = probs.cumsum(dim=-1)
+    pmask = (cum - probs) > top_p.unsqueeze(1)  # keep tokens whose prefix-mass < p
+    mask |= pmask
+    mask[:, 0] = False  # always keep the argmax token
+    sorted_logits = sorted_logits.masked_fill(mask, _NEG_INF)
+    return torch.full_like(logits, _NEG_INF).scatter_(-1, sorted_idx, sorted_logits)
+
+
+class Sampler:
+    def __init__(self, device: str, max_logprobs: int = 11):
+        self.device = device
+        self.max_logprobs = max_logprobs
+
+    def _generator_for(self, req: Request) -> Optional[torch.Generator]:
+        if req.sampling_params.seed is None:
+            return None
+        if req.generator is None:
+            g = torch.Generator(device=self.device)
+            g.manual_seed(req.sampling_params.seed)
+            req.generator = g
+        return req.generator
+
+    @torch.inference_mode()
+    def sample(self, logits: torch.Tensor, requests: list[Request]) -> SamplerOutput:
+        """logits: [N, vocab] raw lm-head outputs for the N sampling rows."""
+        n, vocab = logits.shape
+        assert n == len(requests)
+        logits = logits.float()
+
+        # --- per-request host-side adjustments (rare paths) -------------
+        for i, req in enumerate(requests):
+            p = req.sampling_params
+            # repetition penalty over prompt + generated tokens
+            if p.repetition_penalty != 1.0:
+                ids = torch.tensor(
+                    sorted(set(req.all_token_ids)), device=logits.device, dtype=torch.long
+                )
+                row = logits[i]
+                vals = row[ids]
+                row[ids] = torch.where(
+                    vals > 0, vals / p.repetition_penalty, vals * p.repetition_penalty
+                )
+            # custom logits processors (typical_p, length penalty, ...)
+            if p.logits_processors:
+                row = logits[i]
+                for proc in p.logits_processors:
+                    row = proc(req.output_token_ids, row)
+                logits[i] = row
+            # guided decoding token mask
+            if req.guided_state is not None:
+                allowed = req.guided_state.allowed_token_ids()
+                if allowed is not None:
+                    mask = torch.full((vocab,), _NEG_INF, device=logits.device)
+                    idx = torch.tensor(allowed, device=logits.device, dtype=torch.long)
+                    mask[idx] = 0.0
+                    logits[i] += mask
+            # min_tokens: suppress EOS until satisfied
+            if p.min_tokens and req.num_output_tokens < p.min_tokens and req.eos_token_id is not None:
+                logits[i, req.eos_token_id] = _NEG_INF
+
+        # --- logprob extraction uses pre-temperature logits --------------
+        need_lp = [i for i, r in enumerate(requests) if r.sampling_params.logprobs is not None]
+        logprob_rows = torch.log_softmax(logits[need_lp], dim=-1) if need_lp else None
+
+        # --- sampling -----------------------------------------------------
+        temps = torch.tensor(
+            [r.sampling_params.temperature for r in requests],
+            device=logits.device, dtype=torch.float,
+        )
+        greedy_mask = temps == 0.0
+        sampled = torch.empty(n, dtype=torch.long, device=logits.device)
+
+        if bool(greedy_mask.any()):
+            sampled[greedy_mask] = torch.argmax(logits[greedy_mask], dim=-1)
+
+        sample_idx = (~greedy_mask).nonzero(as_tuple=True)[0]
+        if len(sample_idx):
+            sub = logits[sample_idx] / temps[sample_idx].unsqueeze(1)
+            top_k = torch.tensor(
+                [
+                    requests[i].sampling_params.top_k
+                    if requests[i].sampling_params.top_k > 0 else vocab
+                    for i in sample_idx.tolist()
+                ],
+                device=logits.device,
+            )
+            top_p = torch.tensor(
+                [requests[i].sampling_params.top_p for i in sample_idx.tolist()],
+                device=logits.device, dtype=torch.float,
+            )
+            if bool((top_k < vocab).any()) or bool((top_p < 1.0).any()):
+                sub = _apply_top_k_top_p(sub, top_k, top_p)
+            probs = torch.softmax(sub, dim=-1)
+            # exponential-race sampling; per-request generator when seeded
+            q = torch.empty_like(probs)
+            rows_with_seed = [
+                (j, self._generator_for(requests[i]))
+                for j, i in enumerate(sample_idx.tolist())
+            ]
+            if all(g is None for _, g in rows_with_seed):
+                q.exponential_()
+            else:
+                for j, g in rows_with_seed:
+                    q[j].exponential_(generator=g)
+            sampled[sample_idx] = torch.argmax(probs / q, dim=-1)
+
+        # --- assemble host results ---------------------------------------
+        sampled_cpu = sampled.tolist()
+        out_logprobs: list[Optional[PosLogprobs]] = [None] * n
+        if need_lp:
+            k = self.max_logprobs
+            lp = logprob_rows
+            chosen = sampled[need_lp]
+            chosen_lp = lp.gather(1, chosen.unsqueeze(1)).squeeze(1)
+            ranks = (lp > chosen_lp.unsqueeze(1)).sum(dim=-1) + 1
+            topv, topi = torch.topk(lp, min(k, vocab), dim=-1)
+            topv = topv.tolist()
+            topi = topi.tolist()
+            chosen_lp = chosen_lp.tolist()
+            ranks = ranks.tolist()
+            for row, i in enumerate(need_lp):
+                req = requests[i]
+                num = req.sampling_params.logprobs
+                d: PosLogprobs = {}
+                for j in range(min(num, len(topi[row]))):
+                    d[topi[row][j]] = Logprob(logprob=topv[row][j], rank=j + 1)
+                tok = sampled_cpu[i]
+                if tok not in d:
+                    d[tok] = Logprob(logprob=chosen_lp[row], rank=ranks[row])
+                out_logprobs[i] = d
+        return SamplerOutput(token_ids=sampled_cpu, logprobs=out_logprobs)
+
+
+def prompt_logprob_dicts(
+    logprob_rows: torch.Tensor,  # [M, vocab] log-softmax rows
+    actual_ids: list[int],
+    num_logprobs: int,
+) -> list[PosLogprobs]:
+    """Build per-position logprob dicts for prompt tokens (E8)."""
+    m, vocab = logprob_rows.shape
+    k = min(max(num_logprobs, 1), vocab)
+    topv, topi = torch.topk(logprob_rows, k, dim=-1)
+    ids_t = torch.tensor(actual_ids, device=logprob_rows.device, dtype=torch.long)
+    actual_lp = logprob_rows.gather(1, ids_t.unsqueeze(1)).squeeze(1)
+    ranks = (logprob_rows > actual_lp.unsqueeze(1)).sum(dim=-1) + 1
+    topv = topv.tolist()
+    topi = topi.tolist()
+    actual_lp = actual_lp.tolist()
+    ranks = ranks.tolist()
+    out = []
+    for r in range(m):
+        d: PosLogprobs = {}
+        for j in range(min(num_logprobs, len(topi[r]))):
+            d[topi[r][j]] = Logprob(logprob=topv[r][j], rank=j + 1)
+        tok = actual_ids[r]
+        if tok not in d:
+            d[tok] = Logprob(logprob=actual_lp[r], rank=ranks[r])
+        out.append(d)
+    return out

@@ -1,0 +1,314 @@
+"""GPU worker: owns the model, the paged KV cache, and step execution.
+
+One worker per GPU (one process per GPU for TP>1).  Rank 0
+drives: it turns a SchedulerOutput into a flat token batch, broadcasts it to
+the other TP ranks (torch.distributed over RCCL/gloo), runs the forward, and
+samples.  Non-zero ranks sit in ``worker_loop`` executing broadcast commands.
+"""
+
+from __future__ import annotations
+
+import gc
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+
+from .. import ops
+from ..models import get_model
+from ..parallel import get_tp_rank, get_tp_world_size, tp_broadcast_object
+from .config import EngineConfig
+from .metadata import AttnMetadata
+from .request import Request
+from .sampler import Sampler, SamplerOutput, prompt_logprob_dicts
+from .scheduler import ScheduledItem, SchedulerOutput
+
+
+@dataclass
+class ExecuteResult:
+    # one entry per sampling item (scheduler items with samples=True, in order)
+    sampler_output: SamplerOutput
+    # item-aligned prompt logprob additions handled directly on Request objects
+
+
+def _pad_block_tables(tables: list[list[int]], device) -> torch.Tensor:
+    if not tables:
+        return torch.empty((0, 0), dtype=torch.int32, device=device)
+    maxb = max(1, max(len(t) for t in tables))
+    out = torch.zeros((len(tables), maxb), dtype=torch.int32)
+    for i, t in enumerate(tables):
+        if t:
+            out[i, : len(t)] = torch.tensor(t, dtype=torch.int32)
+    return out.to(device, non_blocking=True)
+
+
+class Worker:
+    def __init__(self, config: EngineConfig):
+        self.config = config
+        self.model_config = config.model_config
+        self.device = config.resolve_device()
+        self.rank = get_tp_rank()
+        self.tp = get_tp_world_size()
+        if self.device == "cuda":
+            torch.cuda.set_device(self.rank % max(1, torch.cuda.device_count()))
+        torch.manual_seed(config.seed)
+
+        with torch.device(self.device):
+            self.model = get_model(self.model_config).eval()
+        self.model.to(self.device)
+        if self.model_config.weights_path:
+            self._load_weights()
+
+        self.block_size = config.cache_config.block_size
+        self.kv_caches: list[tuple[torch.Tensor, torch.Tensor]] = []
+        self.num_blocks = 0
+        self.sampler = Sampler(self.device)
+        self.graph_runner = None  # set by capture_decode_graphs()
+
+    # ------------------------------------------------------------------
+    def _load_weights(self) -> None:
+        from .weights import load_safetensors_weights
+
+        weights = load_safetensors_weights(self.model_config.weights_path)
+        self.model.load_weights(weights)
+
+    # ------------------------------------------------------------------
+    def init_kv_cache(self) -> int:
+        """Allocate the paged KV cache; returns the number of blocks."""
+        cfg = self.config
+        mc = self.model_config
+        kv_heads_local = max(1, mc.num_kv_heads // self.tp)
+        elt = torch.tensor([], dtype=mc.dtype).element_size()
+        block_bytes = 2 * mc.num_layers * self.block_size * kv_heads_local * mc.head_dim * elt
+
+        if cfg.cache_config.num_gpu_blocks is not None:
+            self.num_blocks = cfg.cache_config.num_gpu_blocks
+        elif self.device == "cuda":
+            self._profile_peak_memory()
+            free, total = torch.cuda.mem_get_info()
+            usable = total * cfg.cache_config.gpu_memory_utilization - (total - free)
+            self.num_blocks = max(64, int(usable // block_bytes))
+        else:
+            # CPU tests: enough for max_num_seqs full-length sequences, capped.
+            per_seq = (mc.max_model_len + self.block_size - 1) // self.block_size
+            self.num_blocks = min(8192, per_seq * cfg.scheduler_config.max_num_seqs + 8)
+
+        shape = (self.num_blocks, self.block_size, kv_heads_local, mc.head_dim)
+        self.kv_caches = [
+            (
+                torch.zeros(shape, dtype=mc.dtype, device=self.device),
+                torch.zeros(shape, dtype=mc.dtype, device=self.device),
+            )
+            for _ in range(mc.num_layers)
+        ]
+        return self.num_blocks
+
+    @torch.inference_mode()
+    def _profile_peak_memory(self) -> None:
+        """One forward at the max token budget to materialise activations."""
+        t = min(self.config.scheduler_config.max_num_batched_tokens, 8192)
+        mc = self.model_config
+        ids = torch.zeros(t, dtype=torch.long, device=self.device)
+        pos = torch.zeros(t, dtype=torch.long, device=self.device)
+        meta = AttnMetadata(
+            num_prefill_seqs=1, num_prefill_tokens=t, num_decode_seqs=0,
+            slot_mapping=torch.arange(t, dtype=torch.long, device=self.device),
+            prefill_query_start_loc=torch.tensor([0, t], dtype=torch.int32, device=self.device),
+            prefill_seq_lens=torch.tensor([t], dtype=torch.int32, device=self.device),
+            prefill_block_tables=torch.arange(
+                (t + self.block_size - 1) // self.block_size, dtype=torch.int32,
+                device=self.device).unsqueeze(0),
+            max_prefill_query_len=t, max_prefill_seq_len=t,
+            decode_seq_lens=torch.empty(0, dtype=torch.int32, device=self.device),
+            decode_block_tables=torch.empty((0, 0), dtype=torch.int32, device=self.device),
+            max_decode_seq_len=0,
+        )
+        kv_heads_local = max(1, mc.num_kv_heads // self.tp)
+        nb = (t + self.block_size - 1) // self.block_size
+        shape = (nb, self.block_size, kv_heads_local, mc.head_dim)
+        tmp_cache = [
+            (torch.zeros(shape, dtype=mc.dtype, device=self.device),
+             torch.zeros(shape, dtype=mc.dtype, device=self.device))
+            for _ in range(mc.num_layers)
+        ]
+        hidden = self.model(ids, pos, tmp_cache, meta)
+        self.model.compute_logits(hidden[-1:])
+        del tmp_cache, hidden
+        gc.collect()
+        torch.cuda.empty_cache()
+
+    # ------------------------------------------------------------------
+    def build_batch(self, sched: SchedulerOutput) -> dict:
+        """Rank-0: flatten a SchedulerOutput into broadcastable CPU arrays."""
+        prefills = [i for i in sched.items if i.num_new_tokens > 1]
+        decodes = [i for i in sched.items if i.num_new_tokens == 1]
+        ordered = prefills + decodes
+
+        token_ids: list[int] = []
+        positions: list[int] = []
+        slot_mapping: list[int] = []
+        qsl = [0]
+        prefill_seq_lens = []
+        prefill_tables = []
+        decode_seq_lens = []
+        decode_tables = []
+        sample_rows = []   # row in batch producing the sampled logits, per sampling item
+        extra_rows = []    # rows needed for prompt logprobs
+        bs = self.block_size
+
+        for it in ordered:
+            req = it.request
+            s = req.num_computed_tokens
+            e = s + it.num_new_tokens
+            toks = req.all_token_ids[s:e]
+            token_ids.extend(toks)
+            positions.extend(range(s, e))
+            for pos in range(s, e):
+                b = req.block_ids[pos // bs]
+                slot_mapping.append(b * bs + pos % bs)
+            if it.num_new_tokens > 1:
+                qsl.append(qsl[-1] + it.num_new_tokens)
+                prefill_seq_lens.append(e)
+                prefill_tables.append(req.block_ids)
+            else:
+                decode_seq_lens.append(e)
+                decode_tables.append(req.block_ids)
+
+        row = 0
+        sampling_items: list[ScheduledItem] = []
+        # (item, rows-in-extra-space, prompt token indices, carry_out row or None)
+        prompt_lp_specs = []
+        for it in ordered:
+            req = it.request
+            nrow = row + it.num_new_tokens
+            if it.samples:
+                sample_rows.append(nrow - 1)
+                sampling_items.append(it)
+            if (
+                req.sampling_params.prompt_logprobs is not None
+                and req.num_computed_tokens < req.num_prompt_tokens
+            ):
+                s = req.num_computed_tokens
+                e = min(s + it.num_new_tokens, req.num_prompt_tokens)
+                # rows for logits at positions [s, e-2] predict tokens s+1..e-1
+                lp_rows = list(range(row, row + (e - s) - 1))
+                extra_rows.extend(lp_rows)
+                carry_out = None
+                if e < req.num_prompt_tokens:
+                    # still prefilling after this chunk: also need logits at
+                    # position e-1 to cover token e next chunk
+                    carry_out = row + (e - s) - 1
+                    extra_rows.append(carry_out)
+                prompt_lp_specs.append((it, lp_rows, list(range(s + 1, e)), carry_out))
+            row = nrow
+
+        batch = dict(
+            token_ids=token_ids,
+            positions=positions,
+            slot_mapping=slot_mapping,
+            qsl=qsl if len(qsl) > 1 else [0],
+            prefill_seq_lens=prefill_seq_lens,
+            prefill_tables=prefill_tables,
+            decode_seq_lens=decode_seq_lens,
+            decode_tables=decode_tables,
+            logit_rows=sample_rows + extra_rows,
+            num_sample_rows=len(sample_rows),
+        )
+        # host-only bookkeeping (not broadcast)
+        self._sampling_items = sampling_items
+        self._prompt_lp_specs = prompt_lp_specs
+        self._extra_row_base = len(sample_rows)
+        return batch
+
+    # ------------------------------------------------------------------
+    @torch.inference_mode()
+    def execute_batch(self, batch: dict) -> Optional[torch.Tensor]:
+        """All ranks: run the forward; returns logits rows on every rank."""
+        dev = self.device
+        ids = torch.tensor(batch["token_ids"], dtype=torch.long, device=dev)
+        pos = torch.tensor(batch["positions"], dtype=torch.long, device=dev)
+        np_seqs = len(batch["prefill_seq_lens"])
+        npt = batch["qsl"][-1] if np_seqs else 0
+        meta = AttnMetadata(
+            num_prefill_seqs=np_seqs,
+            num_prefill_tokens=npt,
+            num_decode_seqs=len(batch["decode_seq_lens"]),
+            slot_mapping=torch.tensor(batch["slot_mapping"], dtype=torch.long, device=dev),
+            prefill_query_start_loc=torch.tensor(batch["qsl"], dtype=torch.int32, device=dev),
+            prefill_seq_lens=torch.tensor(batch["prefill_seq_lens"], dtype=torch.int32, device=dev),
+            prefill_block_tables=_pad_block_tables(batch["prefill_tables"], dev),
+            max_prefill_query_len=max(
+                (b - a for a, b in zip(batch["qsl"], batch["qsl"][1:])), default=0),
+            max_prefill_seq_len=max(batch["prefill_seq_lens"], default=0),
+            decode_seq_lens=torch.tensor(batch["decode_seq_lens"], dtype=torch.int32, device=dev),
+            decode_block_tables=_pad_block_tables(batch["decode_tables"], dev),
+            max_decode_seq_len=max(batch["decode_seq_lens"], default=0),
+        )
+        hidden = self.model(ids, pos, self.kv_caches, meta)
+        logit_rows = batch["logit_rows"]
+        if not logit_rows:
+            return None
+        rows = torch.tensor(logit_rows, dtype=torch.long, device=dev)
+        return self.model.compute_logits(hidden[rows])
+
+    # ------------------------------------------------------------------
+    def execute(self, sched: SchedulerOutput) -> ExecuteResult:
+        """Rank-0 entry point for one engine step."""
+        batch = self.build_batch(sched)
+        if self.tp > 1:
+            tp_broadcast_object(("execute", batch))
+        logits = self.execute_batch(batch)
+
+        ns = batch["num_sample_rows"]
+        sampling_reqs = [it.request for it in self._sampling_items]
+        if ns:
+            sampler_out = self.sampler.sample(logits[:ns], sampling_reqs)
+        else:
+            sampler_out = SamplerOutput(token_ids=[], logprobs=[])
+
+        # prompt logprobs (E8): extra logits rows follow the sampling rows,
+        # in the order build_batch appended them to extra_rows.
+        off = self._extra_row_base
+        for it, lp_rows, token_idx, carry_out in self._prompt_lp_specs:
+            req = it.request
+            num_lp = req.sampling_params.prompt_logprobs or 1
+            s = req.num_computed_tokens
+            # token at chunk start: logits carried from the previous chunk
+            if s > 0 and req._plp_carry is not None:
+                req.prompt_logprobs.extend(
+                    prompt_logprob_dicts(
+                        req._plp_carry.unsqueeze(0), [req.prompt_token_ids[s]], num_lp
+                    )
+                )
+            req._plp_carry = None
+            n = len(lp_rows)
+            if n:
+                chunk_rows = logits[off: off + n].float().log_softmax(dim=-1)
+                off += n
+                actual = [req.prompt_token_ids[j] for j in token_idx]
+                req.prompt_logprobs.extend(
+                    prompt_logprob_dicts(chunk_rows, actual, num_lp)
+                )
+            if carry_out is not None:
+                req._plp_carry = logits[off].float().log_softmax(dim=-1)
+                off += 1
+
+        return ExecuteResult(sampler_output=sampler_out)
+
+    # ------------------------------------------------------------------
+    def worker_loop(self) -> None:
+        """Non-zero TP ranks: execute broadcast commands until stopped."""
+        assert self.rank != 0
+        while True:
+            cmd = tp_broadcast_object(None)
+            if cmd is None:
+                continue
+            kind = cmd[0]
+            if kind == "execute":
+                self.execute_batch(cmd[1])
+            elif kind == "stop":
+                return
+
+    def stop_workers(self) -> None:
+        if self.tp > 1 and self.rank == 0:
+            tp_broadcast_object(("stop",))

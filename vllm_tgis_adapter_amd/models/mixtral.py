@@ -1,0 +1,111 @@
+"""Mixtral-class MoE decoder.
+
+Same attention stack as llama; the MLP is a top-k routed mixture of SwiGLU
+experts.  Experts are TP-sharded on the intermediate dimension (every rank
+holds a slice of every expert), so the only collective per MoE block is the
+same single all-reduce a dense row-parallel MLP needs — the right trade at
+xGMI's per-link bandwidth for the TP=4 baseline config.  The expert compute
+is batched per expert (sort tokens by expert, one GEMM per expert); the
+grouped-GEMM HIP kernel will replace the per-expert loop.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .. import ops
+from ..engine.config import ModelConfig
+from ..parallel import divide, get_tp_world_size, tp_all_reduce
+from ..parallel.layers import ParallelLMHead, VocabParallelEmbedding, _init_weight
+from .llama import Attention, RMSNorm
+
+
+class MoEBlock(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        tp = get_tp_world_size()
+        self.num_experts = cfg.num_experts
+        self.top_k = cfg.num_experts_per_tok
+        self.hidden = cfg.hidden_size
+        self.inter = divide(cfg.intermediate_size, tp)
+        self.gate = _init_weight((cfg.num_experts, cfg.hidden_size), cfg.dtype, std=0.02)
+        # w13: [E, 2*I/tp, H] fused gate+up; w2: [E, H, I/tp]
+        self.w13 = _init_weight((cfg.num_experts, 2 * self.inter, cfg.hidden_size), cfg.dtype)
+        self.w2 = _init_weight((cfg.num_experts, cfg.hidden_size, self.inter), cfg.dtype)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:  # [T, H]
+        t = x.shape[0]
+        router_logits = F.linear(x, self.gate)           # [T, E]
+        weights, ids = ops.topk_softmax(router_logits, self.top_k)  # [T, k]
+        out = torch.zeros_like(x)
+        flat_ids = ids.reshape(-1)                        # [T*k]
+        flat_w = weights.reshape(-1).to(x.dtype)
+        token_idx = torch.arange(t, device=x.device).repeat_interleave(self.top_k)
+        for e in range(self.num_experts):
+            sel = flat_ids == e
+            if not torch.any(sel):
+                continue
+            rows = token_idx[sel]
+            xe = x[rows]
+            h = F.linear(xe, self.w13[e])
+            h = ops.silu_and_mul(h)
+            ye = F.linear(h, self.w2[e])
+            out.index_add_(0, rows, ye * flat_w[sel, None])
+        return tp_all_reduce(out)
+
+
+class MixtralDecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, layer_idx: int):
+        super().__init__()
+        self.input_norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, cfg.dtype)
+        self.attn = Attention(cfg, layer_idx)
+        self.post_norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, cfg.dtype)
+        self.moe = MoEBlock(cfg)
+
+    def forward(self, hidden, residual, positions, kv_cache, meta, cos_sin_cache):
+        if residual is None:
+            residual = hidden
+            hidden = self.input_norm(hidden)
+        else:
+            hidden, residual = self.input_norm(hidden, residual)
+        hidden = self.attn(hidden, positions, kv_cache, meta, cos_sin_cache)
+        hidden, residual = self.post_norm(hidden, residual)
+        hidden = self.moe(hidden)
+        return hidden, residual
+
+
+class MixtralForCausalLM(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed = VocabParallelEmbedding(cfg.vocab_size, cfg.hidden_size, dtype=cfg.dtype)
+        self.layers = nn.ModuleList(
+            [MixtralDecoderLayer(cfg, i) for i in range(cfg.num_layers)]
+        )
+        self.final_norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, cfg.dtype)
+        self.lm_head = ParallelLMHead(cfg.vocab_size, cfg.hidden_size, dtype=cfg.dtype)
+        if cfg.tie_word_embeddings:
+            self.lm_head.tie_to(self.embed)
+        self.register_buffer(
+            "cos_sin_cache",
+            ops.make_cos_sin_cache(
+                cfg.head_dim, cfg.max_model_len, cfg.rope_theta, cfg.dtype,
+                cfg.rope_scaling,
+            ),
+            persistent=False,
+        )
+
+    def forward(self, input_ids, positions, kv_caches, meta):
+        hidden = self.embed(input_ids)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            hidden, residual = layer(
+                hidden, residual, positions, kv_caches[i], meta, self.cos_sin_cache
+            )
+        hidden, _ = self.final_norm(hidden, residual)
+        return hidden
+
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        return self.lm_head(hidden)

@@ -1,0 +1,173 @@
+"""Custom-op dispatch.
+
+GPU tensors run the hand-written CDNA4 HIP kernels from the in-tree ``_C``
+extension (kernels/ — built by ``__graft_entry__.build()`` for gfx950).  CPU
+tensors run the plain-PyTorch reference (ops/reference.py), which keeps the
+full engine runnable in CPU-only CI like the reference's CPU vLLM build
+(SURVEY.md §4).  On a GPU box a missing extension is a hard error — there is
+deliberately no silent eager fallback.
+"""
+
+from __future__ import annotations
+
+import os
+
+import torch
+
+from . import reference
+
+_C = None
+_load_error: str | None = None
+
+
+def _try_load_native() -> None:
+    global _C, _load_error
+    if _C is not None:
+        return
+    try:
+        from vllm_tgis_adapter_amd import _C as ext  # in-tree built .so
+
+        _C = ext
+    except ImportError as e:
+        _load_error = str(e)
+
+
+_try_load_native()
+
+_FORCE_REFERENCE = os.environ.get("VTA_FORCE_REFERENCE", "0") == "1"
+
+
+def has_native() -> bool:
+    return _C is not None
+
+
+def _native(t: torch.Tensor) -> bool:
+    if t.device.type != "cuda":
+        return False
+    if _FORCE_REFERENCE:
+        return False
+    if _C is None:
+        raise RuntimeError(
+            "vllm_tgis_adapter_amd._C HIP extension is not built but a GPU "
+            f"tensor reached a custom op (import error: {_load_error}). "
+            "Run `python -c 'import __graft_entry__; __graft_entry__.build()'` "
+            "or kernels/build.py first."
+        )
+    return True
+
+
+# ---------------------------------------------------------------------------
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if _native(x):
+        out = torch.empty_like(x)
+        _C.rms_norm(out, x, weight, eps)
+        return out
+    return reference.rms_norm(x, weight, eps)
+
+
+def fused_add_rms_norm(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """x <- rmsnorm(x + residual); residual <- x + residual (in place on GPU)."""
+    if _native(x):
+        _C.fused_add_rms_norm(x, residual, weight, eps)
+        return x, residual
+    return reference.fused_add_rms_norm(x, residual, weight, eps)
+
+
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if _native(x):
+        d = x.shape[-1] // 2
+        out = torch.empty(*x.shape[:-1], d, dtype=x.dtype, device=x.device)
+        _C.silu_and_mul(out, x)
+        return out
+    return reference.silu_and_mul(x)
+
+
+def rotary_embedding(
+    positions: torch.Tensor,
+    q: torch.Tensor,
+    k: torch.Tensor,
+    head_dim: int,
+    cos_sin_cache: torch.Tensor,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """NeoX rotary; in place on GPU, out-of-place on CPU."""
+    if _native(q):
+        _C.rotary_embedding(positions, q, k, head_dim, cos_sin_cache)
+        return q, k
+    return reference.rotary_embedding(positions, q, k, head_dim, cos_sin_cache)
+
+
+def reshape_and_cache(
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    if _native(k):
+        _C.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+        return
+    reference.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+
+
+def paged_attention_decode(
+    q: torch.Tensor,            # [N, num_heads, head_dim] — one token per seq
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,  # [N, max_blocks] int32, device
+    seq_lens: torch.Tensor,      # [N] int32, device
+    scale: float,
+    max_seq_len: int,
+    out: torch.Tensor | None = None,
+) -> torch.Tensor:
+    if _native(q):
+        if out is None:
+            out = torch.empty_like(q)
+        _C.paged_attention_decode(
+            out, q, k_cache, v_cache, block_tables, seq_lens, scale
+        )
+        return out
+    n = q.shape[0]
+    qsl = torch.arange(n + 1, dtype=torch.int32)
+    r = reference.paged_attention(
+        q, k_cache, v_cache, block_tables.cpu(), qsl, seq_lens.cpu(), scale
+    )
+    if out is not None:
+        out.copy_(r)
+        return out
+    return r
+
+
+def paged_attention_prefill(
+    q: torch.Tensor,               # [T, num_heads, head_dim]
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,    # [num_seqs, max_blocks] int32 device
+    query_start_loc: torch.Tensor,  # [num_seqs+1] int32 device
+    seq_lens: torch.Tensor,         # [num_seqs] int32 device
+    scale: float,
+    max_query_len: int,
+    max_seq_len: int,
+) -> torch.Tensor:
+    if _native(q):
+        out = torch.empty_like(q)
+        _C.paged_attention_prefill(
+            out, q, k_cache, v_cache, block_tables, query_start_loc, seq_lens,
+            scale, max_query_len
+        )
+        return out
+    return reference.paged_attention(
+        q, k_cache, v_cache, block_tables.cpu(), query_start_loc.cpu(),
+        seq_lens.cpu(), scale
+    )
+
+
+def topk_softmax(gate_logits: torch.Tensor, top_k: int):
+    # Router math is tiny; torch ops are fine on both devices for now.
+    return reference.topk_softmax(gate_logits, top_k)
+
+
+make_cos_sin_cache = reference.make_cos_sin_cache
