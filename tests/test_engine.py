@@ -1,0 +1,201 @@
+"""Engine-level unit/behavior tests on the CPU build."""
+
+from __future__ import annotations
+
+import pytest
+import torch
+
+from vllm_tgis_adapter_amd.engine import (
+    EngineConfig,
+    LLMEngine,
+    ModelConfig,
+    SamplingParams,
+)
+from vllm_tgis_adapter_amd.engine.config import CacheConfig, SchedulerConfig
+
+
+def make_engine(max_batched=512, num_blocks=None, seed=0, max_num_seqs=16):
+    mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+    cfg = EngineConfig(
+        model_config=mc,
+        cache_config=CacheConfig(block_size=16, num_gpu_blocks=num_blocks),
+        scheduler_config=SchedulerConfig(
+            max_num_seqs=max_num_seqs, max_num_batched_tokens=max_batched
+        ),
+        seed=seed,
+    )
+    return LLMEngine(cfg)
+
+
+def run_to_completion(engine, max_steps=200):
+    outs = []
+    steps = 0
+    while engine.has_unfinished():
+        outs.extend(engine.step())
+        steps += 1
+        assert steps < max_steps, "engine did not finish"
+    return outs
+
+
+@pytest.fixture(scope="module")
+def engine():
+    return make_engine()
+
+
+def greedy(engine, text, max_tokens=8, **kw):
+    ids = engine.tokenizer(text).input_ids
+    engine.add_request(
+        f"req-{text[:10]}-{max_tokens}", text, ids,
+        SamplingParams(temperature=0.0, max_tokens=max_tokens, **kw),
+    )
+    outs = run_to_completion(engine)
+    return [o for o in outs if o.finished][0]
+
+
+def test_greedy_deterministic(engine):
+    a = greedy(engine, "determinism test")
+    b = greedy(engine, "determinism test")
+    assert a.outputs[0].token_ids == b.outputs[0].token_ids
+
+
+def test_chunked_prefill_equivalence():
+    """A prompt longer than the token budget must chunk across steps and
+    produce the same greedy continuation as an unchunked run."""
+    text = "chunked prefill equivalence test " * 8
+    e_small = make_engine(max_batched=16)
+    e_big = make_engine(max_batched=512)
+    a = greedy(e_small, text, max_tokens=5)
+    b = greedy(e_big, text, max_tokens=5)
+    assert a.outputs[0].token_ids == b.outputs[0].token_ids
+
+
+def test_preemption_recovery():
+    """With a tiny KV pool, concurrent requests force preemption; greedy
+    output must match an uncontended run."""
+    e = make_engine(num_blocks=14, max_batched=128)
+    texts = [f"preemption test prompt number {i} with some length" for i in range(3)]
+    ids = [e.tokenizer(t).input_ids for t in texts]
+    for i, (t, tid) in enumerate(zip(texts, ids)):
+        e.add_request(f"p{i}", t, tid, SamplingParams(temperature=0.0, max_tokens=12))
+    outs = {o.request_id: o for o in run_to_completion(e, max_steps=500) if o.finished}
+    assert len(outs) == 3
+    # reference: run each alone with plenty of blocks
+    for i, (t, tid) in enumerate(zip(texts, ids)):
+        e2 = make_engine(max_batched=512)
+        ref = greedy(e2, t, max_tokens=12)
+        assert outs[f"p{i}"].outputs[0].token_ids == ref.outputs[0].token_ids, i
+
+
+def test_abort_frees_blocks(engine):
+    free_before = engine.block_manager.num_free_blocks
+    ids = engine.tokenizer("abort me please").input_ids
+    engine.add_request("ab1", "abort me please", ids,
+                       SamplingParams(temperature=0.0, max_tokens=50))
+    engine.step()
+    out = engine.abort_request("ab1")
+    assert out is not None and out.finished
+    assert out.outputs[0].finish_reason == "abort"
+    assert engine.block_manager.num_free_blocks == free_before
+    run_to_completion(engine)
+
+
+def test_max_model_len_cap():
+    e = make_engine()
+    ids = list(range(4, 504))  # 500 tokens of a 512 ctx
+    e.add_request("cap", "x", ids, SamplingParams(temperature=0.0, max_tokens=100))
+    outs = run_to_completion(e)
+    final = [o for o in outs if o.finished][0]
+    assert final.outputs[0].finish_reason == "length"
+    assert len(final.outputs[0].token_ids) == 12  # 512 - 500
+
+
+def test_min_tokens_suppresses_eos(engine):
+    ids = engine.tokenizer("short").input_ids
+    engine.add_request(
+        "min1", "short", ids,
+        SamplingParams(temperature=0.0, max_tokens=6, min_tokens=6),
+    )
+    final = [o for o in run_to_completion(engine) if o.finished][0]
+    assert len(final.outputs[0].token_ids) == 6
+
+
+def test_prompt_logprobs(engine):
+    ids = engine.tokenizer("prompt logprob test").input_ids
+    engine.add_request(
+        "plp", "prompt logprob test", ids,
+        SamplingParams(temperature=0.0, max_tokens=2, prompt_logprobs=3, logprobs=3),
+    )
+    final = [o for o in run_to_completion(engine) if o.finished][0]
+    plp = final.prompt_logprobs
+    assert plp is not None
+    assert len(plp) == len(ids)
+    assert plp[0] is None
+    for i, d in enumerate(plp[1:], start=1):
+        assert ids[i] in d
+        assert d[ids[i]].rank >= 1
+
+
+def test_prompt_logprobs_chunked_matches_unchunked():
+    text = "chunked prompt logprobs " * 6
+    results = []
+    for budget in (8, 512):
+        e = make_engine(max_batched=budget)
+        ids = e.tokenizer(text).input_ids
+        e.add_request("plp", text, ids,
+                      SamplingParams(temperature=0.0, max_tokens=1, prompt_logprobs=2))
+        final = [o for o in run_to_completion(e, 500) if o.finished][0]
+        results.append(final.prompt_logprobs)
+    a, b = results
+    assert len(a) == len(b)
+    for da, db in zip(a[1:], b[1:]):
+        assert set(da.keys()) == set(db.keys())
+        for k in da:
+            assert abs(da[k].logprob - db[k].logprob) < 1e-3
+
+
+def test_stop_string_excluded_and_included():
+    e = make_engine()
+    probe = greedy(e, "stop test", max_tokens=3)
+    tok_text = e.detokenizer._convert([probe.outputs[0].token_ids[0]], True)[0]
+    stop = tok_text * 2
+    ids = e.tokenizer("stop test").input_ids
+    e.add_request("s1", "stop test", ids,
+                  SamplingParams(temperature=0.0, max_tokens=20, stop=[stop],
+                                 include_stop_str_in_output=False))
+    final = [o for o in run_to_completion(e) if o.finished][0]
+    assert final.outputs[0].finish_reason == "stop"
+    assert final.outputs[0].stop_reason == stop
+    assert not final.outputs[0].text.endswith(stop)
+
+    e.add_request("s2", "stop test", ids,
+                  SamplingParams(temperature=0.0, max_tokens=20, stop=[stop],
+                                 include_stop_str_in_output=True))
+    final = [o for o in run_to_completion(e) if o.finished][0]
+    assert final.outputs[0].text.endswith(stop)
+
+
+def test_seeded_sampling_reproducible(engine):
+    def run(req_id):
+        ids = engine.tokenizer("seeded").input_ids
+        engine.add_request(req_id, "seeded", ids,
+                           SamplingParams(temperature=1.0, seed=99, max_tokens=8))
+        return [o for o in run_to_completion(engine) if o.finished][0]
+
+    a, b = run("sd1"), run("sd2")
+    assert a.outputs[0].token_ids == b.outputs[0].token_ids
+
+
+def test_delta_outputs_concatenate_to_full():
+    from vllm_tgis_adapter_amd.engine.types import RequestOutputKind
+
+    e = make_engine()
+    ids = e.tokenizer("delta stream check").input_ids
+    e.add_request("d1", "delta stream check", ids,
+                  SamplingParams(temperature=0.0, max_tokens=10,
+                                 output_kind=RequestOutputKind.DELTA))
+    outs = run_to_completion(e)
+    text = "".join(o.outputs[0].text for o in outs if o.request_id == "d1")
+    tokens = [t for o in outs if o.request_id == "d1" for t in o.outputs[0].token_ids]
+    ref = greedy(e, "delta stream check", max_tokens=10)
+    assert tokens == ref.outputs[0].token_ids
+    assert text == ref.outputs[0].text

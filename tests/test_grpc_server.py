@@ -1,0 +1,271 @@
+"""End-to-end gRPC behavior tests against the live dual server
+(coverage modeled on reference tests/test_grpc_server.py)."""
+
+from __future__ import annotations
+
+import grpc
+import pytest
+
+from vllm_tgis_adapter_amd.grpc import proto
+
+SR = proto.StopReasonValue
+
+
+def _gen_req(text="The answer to life the universe and everything is ",
+             max_new=10, **params):
+    req = proto.BatchedGenerationRequest(
+        model_id="m", requests=[proto.GenerationRequest(text=text)]
+    )
+    req.params.stopping.max_new_tokens = max_new
+    for k, v in params.items():
+        setattr(req.params.response, k, v)
+    return req
+
+
+def test_generation_request(grpc_client):
+    resp = grpc_client.Generate(_gen_req(), timeout=60)
+    assert len(resp.responses) == 1
+    r = resp.responses[0]
+    assert r.generated_token_count == 10
+    assert r.stop_reason == SR.MAX_TOKENS
+    assert r.input_token_count > 0
+    assert r.text
+
+
+def test_batched_generation_request(grpc_client):
+    req = proto.BatchedGenerationRequest(
+        model_id="m",
+        requests=[
+            proto.GenerationRequest(text="first prompt"),
+            proto.GenerationRequest(text="a second, longer prompt with more tokens"),
+            proto.GenerationRequest(text="third"),
+        ],
+    )
+    req.params.stopping.max_new_tokens = 7
+    resp = grpc_client.Generate(req, timeout=60)
+    assert len(resp.responses) == 3
+    for r in resp.responses:
+        assert r.generated_token_count == 7
+        assert r.stop_reason == SR.MAX_TOKENS
+
+
+def test_generation_request_stream(grpc_client):
+    req = proto.SingleGenerationRequest(
+        model_id="m", request=proto.GenerationRequest(text="stream this prompt")
+    )
+    req.params.stopping.max_new_tokens = 10
+    messages = list(grpc_client.GenerateStream(req, timeout=60))
+    # N generated tokens => N+1 messages, first carries only input details
+    assert len(messages) == 11
+    first = messages[0]
+    assert first.input_token_count > 0
+    assert first.generated_token_count == 0
+    assert first.text == ""
+    total_text = "".join(m.text for m in messages)
+    assert total_text
+    assert messages[-1].stop_reason == SR.MAX_TOKENS
+
+
+def test_input_text_echo(grpc_client):
+    req = _gen_req(text="echo me", input_text=True)
+    r = grpc_client.Generate(req, timeout=60).responses[0]
+    assert r.text.startswith("echo me")
+
+
+def test_token_details_and_top_n(grpc_client):
+    req = _gen_req(
+        generated_tokens=True, token_logprobs=True, token_ranks=True, top_n_tokens=3
+    )
+    req.params.response.input_tokens = True
+    r = grpc_client.Generate(req, timeout=60).responses[0]
+    assert len(r.tokens) == r.generated_token_count
+    for t in r.tokens:
+        assert t.logprob <= 0.0
+        assert t.rank >= 1
+        assert 1 <= len(t.top_tokens) <= 3
+    # input token details: first has no logprob entry
+    assert len(r.input_tokens) == r.input_token_count
+    assert all(t.rank >= 1 for t in r.input_tokens[1:])
+
+
+def test_stop_sequence(grpc_client):
+    # greedy decode emits repeated text; use a prefix as the stop sequence
+    probe = _gen_req(max_new=4)
+    text = grpc_client.Generate(probe, timeout=60).responses[0].text
+    assert len(text) >= 2
+    tok_text = text[0]
+    req = _gen_req(max_new=20)
+    req.params.stopping.stop_sequences.append(tok_text * 2)
+    r = grpc_client.Generate(req, timeout=60).responses[0]
+    assert r.stop_reason == SR.STOP_SEQUENCE
+    assert r.stop_sequence == tok_text * 2
+    # default include_stop_seqs=True keeps the stop sequence in the text
+    assert r.text.endswith(tok_text * 2)
+
+
+def test_tokenize(grpc_client):
+    req = proto.BatchedTokenizeRequest(
+        model_id="m",
+        requests=[proto.TokenizeRequest(text="abc def"),
+                  proto.TokenizeRequest(text="xyz")],
+        return_tokens=True,
+    )
+    resp = grpc_client.Tokenize(req, timeout=30)
+    assert len(resp.responses) == 2
+    assert resp.responses[0].token_count == len(resp.responses[0].tokens)
+    assert resp.responses[0].token_count > 0
+
+
+def test_tokenize_truncation_keeps_last(grpc_client):
+    full = grpc_client.Tokenize(
+        proto.BatchedTokenizeRequest(
+            model_id="m", requests=[proto.TokenizeRequest(text="abcdefgh")],
+            return_tokens=True,
+        ),
+        timeout=30,
+    ).responses[0]
+    trunc = grpc_client.Tokenize(
+        proto.BatchedTokenizeRequest(
+            model_id="m", requests=[proto.TokenizeRequest(text="abcdefgh")],
+            return_tokens=True, truncate_input_tokens=3,
+        ),
+        timeout=30,
+    ).responses[0]
+    assert trunc.token_count == 3
+    assert list(trunc.tokens) == list(full.tokens)[-3:]
+
+
+def test_tokenize_offsets(grpc_client):
+    resp = grpc_client.Tokenize(
+        proto.BatchedTokenizeRequest(
+            model_id="m", requests=[proto.TokenizeRequest(text="hello")],
+            return_tokens=True, return_offsets=True,
+        ),
+        timeout=30,
+    ).responses[0]
+    assert len(resp.offsets) == resp.token_count
+
+
+def test_model_info(grpc_client):
+    resp = grpc_client.ModelInfo(proto.ModelInfoRequest(model_id="m"), timeout=30)
+    assert resp.max_sequence_length == 512
+    assert resp.max_new_tokens == 1024
+    assert resp.model_kind == 0  # DECODER_ONLY
+
+
+def test_time_limit(grpc_client):
+    req = proto.SingleGenerationRequest(
+        model_id="m", request=proto.GenerationRequest(text="time limited")
+    )
+    req.params.stopping.time_limit_millis = 60
+    messages = list(grpc_client.GenerateStream(req, timeout=60))
+    assert messages[-1].stop_reason == SR.TIME_LIMIT
+
+
+def test_sampling_seed_determinism(grpc_client):
+    def run():
+        req = _gen_req(generated_tokens=True)
+        req.params.method = proto.SAMPLE
+        req.params.sampling.temperature = 0.9
+        req.params.sampling.seed = 12345
+        return [t.text for t in grpc_client.Generate(req, timeout=60).responses[0].tokens]
+
+    assert run() == run()
+
+
+@pytest.mark.parametrize(
+    ("case", "msg"),
+    [
+        ("top_p", "top_p must be > 0.0 and <= 1.0"),
+        ("typical_p", "typical_p must be <= 1.0"),
+        ("rep", "repetition_penalty must be > 0.0 and <= 2.0"),
+        ("length_pen", "length_penalty.decay_factor must be >= 1.0 and <= 10.0"),
+        ("max_new", "max_new_tokens must be <= 1024"),
+        ("min_new", "min_new_tokens must be <= max_new_tokens"),
+        ("stop_seqs", "can specify at most 6 non-empty stop sequences, each "
+                      "not more than 240 UTF8 bytes"),
+        ("top_n", "top_n_tokens (11) must be <= 10"),
+        ("detail", "must request input and/or generated tokens to request extra "
+                   "token detail"),
+    ],
+)
+def test_validation_errors(grpc_client, case, msg):
+    req = _gen_req()
+    p = req.params
+    if case == "top_p":
+        p.method = proto.SAMPLE
+        p.sampling.top_p = 1.5
+    elif case == "typical_p":
+        p.method = proto.SAMPLE
+        p.sampling.typical_p = 1.5
+    elif case == "rep":
+        p.decoding.repetition_penalty = 3.0
+    elif case == "length_pen":
+        p.decoding.length_penalty.start_index = 0
+        p.decoding.length_penalty.decay_factor = 100.0
+    elif case == "max_new":
+        p.stopping.max_new_tokens = 5000
+    elif case == "min_new":
+        p.stopping.min_new_tokens = 20
+        p.stopping.max_new_tokens = 10
+    elif case == "stop_seqs":
+        p.stopping.stop_sequences.extend(["a"] * 7)
+    elif case == "top_n":
+        p.response.top_n_tokens = 11
+        p.response.generated_tokens = True
+    elif case == "detail":
+        p.response.token_logprobs = True
+    with pytest.raises(grpc.RpcError) as e:
+        grpc_client.Generate(req, timeout=30)
+    assert e.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+    assert e.value.details() == msg
+
+
+def test_guided_regex(grpc_client):
+    req = _gen_req(max_new=8, generated_tokens=True)
+    req.params.decoding.regex = "[0-9]+"
+    r = grpc_client.Generate(req, timeout=120).responses[0]
+    assert r.text
+    assert all(c in "0123456789" for c in r.text)
+
+
+def test_guided_choice(grpc_client):
+    req = _gen_req(max_new=8)
+    req.params.decoding.choice.choices.extend(["yes", "no"])
+    r = grpc_client.Generate(req, timeout=120).responses[0]
+    assert r.text in ("yes", "no", "ye", "y", "n")  # may hit token limit mid-choice
+
+
+def test_correlation_id_used_as_request_id(grpc_client, _servers):
+    req = _gen_req(max_new=2)
+    resp = grpc_client.Generate(
+        req, timeout=60, metadata=(("x-correlation-id", "my-corr-id"),)
+    )
+    assert resp.responses[0].generated_token_count == 2
+
+
+def test_lora_adapter_request(grpc_client):
+    req = _gen_req(max_new=4)
+    req.adapter_id = "tiny-lora"
+    r = grpc_client.Generate(req, timeout=120).responses[0]
+    assert r.generated_token_count == 4
+    base = grpc_client.Generate(_gen_req(max_new=4), timeout=60).responses[0]
+    assert base.generated_token_count == 4
+
+
+def test_unknown_adapter_errors(grpc_client):
+    req = _gen_req(max_new=2)
+    req.adapter_id = "does-not-exist"
+    with pytest.raises(grpc.RpcError) as e:
+        grpc_client.Generate(req, timeout=30)
+    assert e.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+    assert "can't retrieve adapter" in e.value.details()
+
+
+def test_adapter_path_traversal_rejected(grpc_client):
+    req = _gen_req(max_new=2)
+    req.adapter_id = "../../etc"
+    with pytest.raises(grpc.RpcError) as e:
+        grpc_client.Generate(req, timeout=30)
+    assert e.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+    assert "Invalid adapter id" in e.value.details()

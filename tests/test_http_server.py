@@ -1,0 +1,78 @@
+"""HTTP front-end tests (coverage modeled on reference tests/test_http_server.py)."""
+
+from __future__ import annotations
+
+import json
+import urllib.request
+
+
+def _get(url: str):
+    return urllib.request.urlopen(url, timeout=30)
+
+
+def _post_json(url: str, payload: dict, headers: dict | None = None):
+    req = urllib.request.Request(
+        url,
+        data=json.dumps(payload).encode(),
+        headers={"Content-Type": "application/json", **(headers or {})},
+    )
+    return urllib.request.urlopen(req, timeout=60)
+
+
+def test_health(http_base):
+    assert _get(f"{http_base}/health").status == 200
+
+
+def test_version(http_base):
+    data = json.load(_get(f"{http_base}/version"))
+    assert "version" in data
+
+
+def test_models_and_completions(http_base):
+    models = json.load(_get(f"{http_base}/v1/models"))
+    model_id = models["data"][0]["id"]
+    assert model_id
+    resp = json.load(_post_json(
+        f"{http_base}/v1/completions",
+        {"model": model_id, "prompt": "The capital of France is", "max_tokens": 5},
+    ))
+    assert len(resp["choices"]) == 1
+    assert resp["usage"]["completion_tokens"] == 5
+    assert resp["choices"][0]["finish_reason"] == "length"
+
+
+def test_completions_multi_prompt(http_base):
+    resp = json.load(_post_json(
+        f"{http_base}/v1/completions",
+        {"model": "m", "prompt": ["one", "two"], "max_tokens": 3},
+    ))
+    assert len(resp["choices"]) == 2
+    assert {c["index"] for c in resp["choices"]} == {0, 1}
+
+
+def test_completions_streaming(http_base):
+    r = _post_json(
+        f"{http_base}/v1/completions",
+        {"model": "m", "prompt": "stream", "max_tokens": 4, "stream": True},
+    )
+    body = r.read().decode()
+    events = [line for line in body.split("\n\n") if line.startswith("data: ")]
+    assert events[-1] == "data: [DONE]"
+    chunks = [json.loads(e[len("data: "):]) for e in events[:-1]]
+    assert chunks
+
+
+def test_chat_completions(http_base):
+    resp = json.load(_post_json(
+        f"{http_base}/v1/chat/completions",
+        {"model": "m", "messages": [{"role": "user", "content": "hi"}],
+         "max_tokens": 4},
+    ))
+    assert resp["choices"][0]["message"]["role"] == "assistant"
+
+
+def test_metrics(http_base):
+    body = _get(f"{http_base}/metrics").read().decode()
+    assert "tgis_amd:request_success" in body
+    assert "tgis_amd:generation_tokens" in body
+    assert "tgis_amd:time_to_first_token_seconds" in body

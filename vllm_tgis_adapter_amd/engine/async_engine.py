@@ -89,6 +89,14 @@ class AsyncLLMEngine:
         self._cmds.put(("abort", request_id))
         self._wakeup.set()
 
+    async def add_lora(self, lora_request: LoRARequest) -> None:
+        import concurrent.futures
+
+        fut: concurrent.futures.Future = concurrent.futures.Future()
+        self._cmds.put(("add_lora", lora_request, fut))
+        self._wakeup.set()
+        await asyncio.wrap_future(fut)
+
     def generate(
         self,
         prompt=None,
@@ -186,6 +194,13 @@ class AsyncLLMEngine:
                 arrival_time=arrival, lora_request=lora_request,
                 trace_headers=trace_headers,
             )
+        elif kind == "add_lora":
+            _, lora_request, fut = cmd
+            try:
+                self.engine.add_lora(lora_request)
+                fut.set_result(None)
+            except BaseException as e:
+                fut.set_exception(e)
         elif kind == "abort":
             request_id = cmd[1]
             out = self.engine.abort_request(request_id)

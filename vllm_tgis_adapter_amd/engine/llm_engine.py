@@ -38,6 +38,9 @@ class LLMEngine:
         self.detokenizer = Detokenizer(self.tokenizer)
         self.stop_checker = StopChecker(self.model_config.max_model_len)
         self.eos_token_id = self.tokenizer.eos_token_id
+        from .metrics import EngineMetrics
+
+        self.metrics = EngineMetrics(self.model_config.model)
 
     # ------------------------------------------------------------------
     def add_request(
@@ -68,6 +71,9 @@ class LLMEngine:
             )
         self.scheduler.add_request(req)
         return req
+
+    def add_lora(self, lora_request: LoRARequest) -> None:
+        self.worker.add_lora(lora_request.lora_path, lora_request.lora_int_id)
 
     def abort_request(self, request_id: str) -> Optional[RequestOutput]:
         req = self.scheduler.abort_request(request_id)
@@ -120,6 +126,22 @@ class LLMEngine:
                 outputs.append(out)
             if req.status.is_finished:
                 self.scheduler.finish_request(req)
+                m = req.metrics
+                self.metrics.request_success.inc()
+                self.metrics.prompt_tokens.inc(req.num_prompt_tokens)
+                self.metrics.generation_tokens.inc(req.num_output_tokens)
+                if m.first_token_time and m.first_scheduled_time:
+                    self.metrics.ttft.observe(m.first_token_time - m.arrival_time)
+                    if req.num_output_tokens > 1 and m.last_token_time:
+                        self.metrics.time_per_output_token.observe(
+                            (m.last_token_time - m.first_token_time)
+                            / (req.num_output_tokens - 1)
+                        )
+        self.metrics.num_running.set(len(self.scheduler.running))
+        self.metrics.num_waiting.set(len(self.scheduler.waiting))
+        self.metrics.kv_usage.set(
+            1.0 - self.block_manager.num_free_blocks / max(1, self.block_manager.num_blocks)
+        )
         return outputs
 
     def shutdown(self) -> None:

@@ -1,0 +1,158 @@
+"""Multi-LoRA runtime.
+
+Hot-loads PEFT LoRA adapters (adapter_config.json + adapter_model.safetensors)
+onto the serving model and applies them per request in mixed batches.
+
+Application model: each step batch carries a per-token adapter id; for every
+active adapter the delta  x @ A^T @ B^T * (alpha/r)  is added onto the
+matching projection outputs.  Today this runs as batched torch GEMMs grouped
+by adapter (shrink+expand per adapter — fine for a handful of live adapters);
+the batched SGMV HIP kernel replaces the inner loop next.
+
+TP sharding follows the base layer: column-parallel projections
+(q/k/v/gate/up) shard B on the output dim; row-parallel projections (o/down)
+shard A on the input dim.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+
+from ..parallel import get_tp_rank, get_tp_world_size
+
+# projection name -> (parallel kind, hf targets)
+_COLUMN = {"q_proj", "k_proj", "v_proj", "gate_proj", "up_proj"}
+_ROW = {"o_proj", "down_proj"}
+_SUPPORTED = _COLUMN | _ROW
+
+
+@dataclass
+class LoRAAdapter:
+    lora_int_id: int
+    rank: int
+    scaling: float
+    # (layer_idx, proj) -> (A [r, in_local], B [out_local, r])
+    weights: dict[tuple[int, str], tuple[torch.Tensor, torch.Tensor]]
+
+
+@dataclass
+class LoRAContext:
+    """Per-step batch context set by the worker before the forward."""
+
+    token_lora_ids: Optional[torch.Tensor] = None  # [T] int32 device (0 = none)
+    adapters: dict[int, LoRAAdapter] = field(default_factory=dict)
+
+    @property
+    def active(self) -> bool:
+        return self.token_lora_ids is not None and bool(self.adapters)
+
+
+CTX = LoRAContext()
+
+
+def set_context(token_lora_ids: Optional[torch.Tensor], adapters: dict[int, LoRAAdapter]) -> None:
+    CTX.token_lora_ids = token_lora_ids
+    CTX.adapters = adapters
+
+
+def clear_context() -> None:
+    CTX.token_lora_ids = None
+    CTX.adapters = {}
+
+
+def apply_lora(
+    layer_idx: int,
+    projs: list[tuple[str, int, int]],  # (proj name, out-slice start, out-slice end)
+    x: torch.Tensor,     # [T, in_local] input of the base linear
+    out: torch.Tensor,   # [T, sum(out_local)] output to update in place
+) -> torch.Tensor:
+    if not CTX.active:
+        return out
+    token_ids = CTX.token_lora_ids
+    for lora_id, adapter in CTX.adapters.items():
+        sel = (token_ids == lora_id).nonzero(as_tuple=True)[0]
+        if sel.numel() == 0:
+            continue
+        x_sub = x[sel]
+        for name, start, end in projs:
+            ab = adapter.weights.get((layer_idx, name))
+            if ab is None:
+                continue
+            a, b = ab
+            delta = (x_sub @ a.t()) @ b.t()
+            out[sel, start:end] += (delta * adapter.scaling).to(out.dtype)
+    return out
+
+
+def load_lora_adapter(
+    path: str,
+    lora_int_id: int,
+    *,
+    device: str,
+    dtype: torch.dtype,
+    max_lora_rank: int,
+) -> LoRAAdapter:
+    cfg_path = os.path.join(path, "adapter_config.json")
+    with open(cfg_path) as f:
+        cfg = json.load(f)
+    if cfg.get("peft_type") != "LORA":
+        raise ValueError(f"unsupported peft_type {cfg.get('peft_type')}")
+    r = int(cfg["r"])
+    if r > max_lora_rank:
+        raise ValueError(f"LoRA rank {r} exceeds max_lora_rank {max_lora_rank}")
+    alpha = float(cfg.get("lora_alpha", r))
+    scaling = alpha / r
+
+    weights_file = os.path.join(path, "adapter_model.safetensors")
+    if os.path.exists(weights_file):
+        from safetensors.torch import load_file
+
+        raw = load_file(weights_file)
+    else:
+        bin_file = os.path.join(path, "adapter_model.bin")
+        raw = torch.load(bin_file, map_location="cpu", weights_only=True)
+
+    tp = get_tp_world_size()
+    rank_idx = get_tp_rank()
+    weights: dict[tuple[int, str], tuple[torch.Tensor, torch.Tensor]] = {}
+    pending: dict[tuple[int, str], dict[str, torch.Tensor]] = {}
+    for key, tensor in raw.items():
+        # e.g. base_model.model.model.layers.3.self_attn.q_proj.lora_A.weight
+        parts = key.split(".")
+        try:
+            li = parts.index("layers")
+            layer_idx = int(parts[li + 1])
+        except (ValueError, IndexError):
+            continue
+        proj = next((p for p in parts if p in _SUPPORTED), None)
+        if proj is None:
+            continue
+        ab = "A" if "lora_A" in key else ("B" if "lora_B" in key else None)
+        if ab is None:
+            continue
+        pending.setdefault((layer_idx, proj), {})[ab] = tensor
+
+    for (layer_idx, proj), d in pending.items():
+        if "A" not in d or "B" not in d:
+            continue
+        a = d["A"].to(torch.float32)  # [r, in]
+        b = d["B"].to(torch.float32)  # [out, r]
+        if tp > 1:
+            if proj in _COLUMN:
+                shard = b.shape[0] // tp
+                b = b[rank_idx * shard:(rank_idx + 1) * shard]
+            else:
+                shard = a.shape[1] // tp
+                a = a[:, rank_idx * shard:(rank_idx + 1) * shard]
+        weights[(layer_idx, proj)] = (
+            a.to(device=device, dtype=dtype),
+            b.to(device=device, dtype=dtype),
+        )
+    if not weights:
+        raise ValueError(f"adapter at {path} has no supported LoRA weights")
+    return LoRAAdapter(lora_int_id=lora_int_id, rank=r, scaling=scaling, weights=weights)

@@ -64,6 +64,19 @@ class Worker:
         self.num_blocks = 0
         self.sampler = Sampler(self.device)
         self.graph_runner = None  # set by capture_decode_graphs()
+        self.loras: dict[int, object] = {}  # lora_int_id -> LoRAAdapter
+
+    # ------------------------------------------------------------------
+    def add_lora(self, lora_path: str, lora_int_id: int) -> None:
+        from .lora import load_lora_adapter
+
+        if self.tp > 1 and self.rank == 0:
+            tp_broadcast_object(("add_lora", lora_path, lora_int_id))
+        self.loras[lora_int_id] = load_lora_adapter(
+            lora_path, lora_int_id,
+            device=self.device, dtype=self.model_config.dtype,
+            max_lora_rank=self.config.max_lora_rank,
+        )
 
     # ------------------------------------------------------------------
     def _load_weights(self) -> None:
@@ -147,6 +160,7 @@ class Worker:
         token_ids: list[int] = []
         positions: list[int] = []
         slot_mapping: list[int] = []
+        lora_ids: list[int] = []
         qsl = [0]
         prefill_seq_lens = []
         prefill_tables = []
@@ -163,6 +177,8 @@ class Worker:
             toks = req.all_token_ids[s:e]
             token_ids.extend(toks)
             positions.extend(range(s, e))
+            lid = req.lora_request.lora_int_id if req.lora_request else 0
+            lora_ids.extend([lid] * (e - s))
             for pos in range(s, e):
                 b = req.block_ids[pos // bs]
                 slot_mapping.append(b * bs + pos % bs)
@@ -213,6 +229,7 @@ class Worker:
             decode_tables=decode_tables,
             logit_rows=sample_rows + extra_rows,
             num_sample_rows=len(sample_rows),
+            lora_ids=lora_ids if any(lora_ids) else None,
         )
         # host-only bookkeeping (not broadcast)
         self._sampling_items = sampling_items
@@ -244,7 +261,17 @@ class Worker:
             decode_block_tables=_pad_block_tables(batch["decode_tables"], dev),
             max_decode_seq_len=max(batch["decode_seq_lens"], default=0),
         )
-        hidden = self.model(ids, pos, self.kv_caches, meta)
+        from . import lora as lora_rt
+
+        if batch.get("lora_ids"):
+            lora_rt.set_context(
+                torch.tensor(batch["lora_ids"], dtype=torch.int32, device=dev),
+                self.loras,
+            )
+        try:
+            hidden = self.model(ids, pos, self.kv_caches, meta)
+        finally:
+            lora_rt.clear_context()
         logit_rows = batch["logit_rows"]
         if not logit_rows:
             return None
@@ -306,6 +333,8 @@ class Worker:
             kind = cmd[0]
             if kind == "execute":
                 self.execute_batch(cmd[1])
+            elif kind == "add_lora":
+                self.add_lora(cmd[1], cmd[2])
             elif kind == "stop":
                 return
 

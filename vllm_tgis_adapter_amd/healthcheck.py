@@ -1,0 +1,71 @@
+"""Standalone gRPC health-probe client, installed as ``grpc_healthcheck``
+(reference healthcheck.py semantics: default target localhost:8033, service
+fmaas.GenerationService, exit 1 when not SERVING)."""
+
+from __future__ import annotations
+
+import argparse
+import sys
+
+import grpc
+
+from .grpc.proto import HealthCheckRequest
+from .grpc.stubs import HealthStub
+
+
+def health_check(
+    *,
+    server_url: str = "localhost:8033",
+    service: str | None = None,
+    insecure: bool = True,
+    timeout: float = 1,
+) -> bool:
+    print("health check...", end="")
+    request = HealthCheckRequest(service=service or "")
+    try:
+        if insecure:
+            channel = grpc.insecure_channel(server_url)
+        else:
+            channel = grpc.secure_channel(server_url, grpc.ssl_channel_credentials())
+        with channel:
+            response = HealthStub(channel).Check(request, timeout=timeout)
+    except grpc.RpcError as e:
+        print(f"Health.Check failed: code={e.code()}, details={e.details()}")
+        return False
+    print(str(response).strip())
+    return response.status == 1  # SERVING
+
+
+def parse_args() -> argparse.Namespace:
+    parser = argparse.ArgumentParser()
+    parser.formatter_class = argparse.ArgumentDefaultsHelpFormatter
+    group = parser.add_mutually_exclusive_group(required=False)
+    group.add_argument("--insecure", dest="insecure", action="store_true",
+                       help="Use an insecure connection")
+    group.add_argument("--secure", dest="secure", action="store_true",
+                       help="Use a secure connection")
+    group.set_defaults(insecure=True, secure=False)
+    parser.add_argument("--server-url", type=str, default="localhost:8033",
+                        help="grpc server url (`host:port`)")
+    parser.add_argument("--timeout", type=float, default=1,
+                        help="Timeout for healthcheck request")
+    parser.add_argument("--service-name", type=str, required=False,
+                        default="fmaas.GenerationService",
+                        help="Name of the service to check")
+    return parser.parse_args()
+
+
+def cli() -> None:
+    args = parse_args()
+    ok = health_check(
+        server_url=args.server_url,
+        service=args.service_name,
+        insecure=not args.secure,
+        timeout=args.timeout,
+    )
+    if not ok:
+        sys.exit(1)
+
+
+if __name__ == "__main__":
+    cli()

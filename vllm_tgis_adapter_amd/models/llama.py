@@ -56,6 +56,11 @@ class Attention(nn.Module):
         self.q_local = self.num_heads * self.head_dim
         self.kv_local = self.num_kv_heads * self.head_dim
         self.o_proj = RowParallelLinear(q_size, cfg.hidden_size, dtype=cfg.dtype)
+        q, kv = self.q_local, self.kv_local
+        self._qkv_lora = (layer_idx, [
+            ("q_proj", 0, q), ("k_proj", q, q + kv), ("v_proj", q + kv, q + 2 * kv),
+        ])
+        self._o_lora = (layer_idx, [("o_proj", 0, cfg.hidden_size)])
 
     def forward(
         self,
@@ -65,7 +70,7 @@ class Attention(nn.Module):
         meta: AttnMetadata,
         cos_sin_cache: torch.Tensor,
     ) -> torch.Tensor:
-        qkv = self.qkv_proj(hidden)
+        qkv = self.qkv_proj(hidden, lora=self._qkv_lora)
         q, k, v = qkv.split([self.q_local, self.kv_local, self.kv_local], dim=-1)
         q, k = ops.rotary_embedding(positions, q, k, self.head_dim, cos_sin_cache)
         t = hidden.shape[0]
@@ -91,20 +96,26 @@ class Attention(nn.Module):
                 meta.decode_block_tables, meta.decode_seq_lens,
                 self.scale, meta.max_decode_seq_len, out=out[npt:],
             )
-        return self.o_proj(out.view(t, -1))
+        return self.o_proj(out.view(t, -1), lora=self._o_lora)
 
 
 class LlamaMLP(nn.Module):
-    def __init__(self, cfg: ModelConfig):
+    def __init__(self, cfg: ModelConfig, layer_idx: int):
         super().__init__()
         self.gate_up = MergedColumnParallelLinear(
             cfg.hidden_size, [cfg.intermediate_size, cfg.intermediate_size],
             dtype=cfg.dtype,
         )
         self.down = RowParallelLinear(cfg.intermediate_size, cfg.hidden_size, dtype=cfg.dtype)
+        inter = self.gate_up.shard_sizes[0]
+        self._gu_lora = (layer_idx, [
+            ("gate_proj", 0, inter), ("up_proj", inter, 2 * inter),
+        ])
+        self._down_lora = (layer_idx, [("down_proj", 0, cfg.hidden_size)])
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.down(ops.silu_and_mul(self.gate_up(x)))
+        h = ops.silu_and_mul(self.gate_up(x, lora=self._gu_lora))
+        return self.down(h, lora=self._down_lora)
 
 
 class LlamaDecoderLayer(nn.Module):
@@ -113,7 +124,7 @@ class LlamaDecoderLayer(nn.Module):
         self.input_norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, cfg.dtype)
         self.attn = Attention(cfg, layer_idx)
         self.post_norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, cfg.dtype)
-        self.mlp = LlamaMLP(cfg)
+        self.mlp = LlamaMLP(cfg, layer_idx)
 
     def forward(self, hidden, residual, positions, kv_cache, meta, cos_sin_cache):
         if residual is None:

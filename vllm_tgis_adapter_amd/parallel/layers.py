@@ -63,8 +63,14 @@ class MergedColumnParallelLinear(nn.Module):
         self.weight = _init_weight((total, in_size), dtype)
         self.bias = _init_weight((total,), dtype, std=0.0) if bias else None
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.linear(x, self.weight, self.bias)
+    def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
+        y = F.linear(x, self.weight, self.bias)
+        if lora is not None:
+            from ..engine.lora import apply_lora
+
+            layer_idx, projs = lora
+            apply_lora(layer_idx, projs, x, y)
+        return y
 
     def load_full_weights(self, ws: list[torch.Tensor]) -> None:
         r = get_tp_rank()
@@ -87,8 +93,15 @@ class RowParallelLinear(nn.Module):
         self.weight = _init_weight((out_size, self.in_per_rank), dtype)
         self.bias = _init_weight((out_size,), dtype, std=0.0) if bias else None
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
         y = F.linear(x, self.weight)
+        if lora is not None:
+            # delta added BEFORE the all-reduce: A is input-sharded, so the
+            # per-rank partial deltas sum to the full LoRA delta
+            from ..engine.lora import apply_lora
+
+            layer_idx, projs = lora
+            apply_lora(layer_idx, projs, x, y)
         y = tp_all_reduce(y)
         if self.bias is not None:
             y = y + self.bias
