@@ -1,0 +1,190 @@
+// Torch extension bindings for the CDNA4 kernels.
+// Built in-tree as vllm_tgis_adapter_amd/_C.so by kernels/build.py
+// (hipcc --offload-arch=gfx950).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+// launchers (defined in the .hip translation units)
+template <typename T>
+void launch_rms_norm(T*, const T*, const T*, float, int, int, hipStream_t);
+template <typename T>
+void launch_fused_add_rms_norm(T*, T*, const T*, float, int, int, hipStream_t);
+template <typename T>
+void launch_silu_and_mul(T*, const T*, int, int, hipStream_t);
+template <typename T>
+void launch_rotary_embedding(const long*, T*, T*, const float*, int, int, int,
+                             int, hipStream_t);
+template <typename T>
+void launch_reshape_and_cache(const T*, const T*, T*, T*, const long*, int,
+                              int, hipStream_t);
+template <typename T>
+void launch_paged_decode(T*, const T*, const T*, const T*, const int*,
+                         const int*, float, int, int, int, int, int, int,
+                         hipStream_t);
+template <typename T>
+void launch_paged_prefill(T*, const T*, const T*, const T*, const int*,
+                          const int*, const int*, float, int, int, int, int,
+                          int, int, int, hipStream_t);
+
+namespace {
+
+hipStream_t current_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+template <typename T>
+T* ptr(torch::Tensor& t) {
+  return reinterpret_cast<T*>(t.data_ptr());
+}
+template <typename T>
+const T* cptr(const torch::Tensor& t) {
+  return reinterpret_cast<const T*>(t.data_ptr());
+}
+
+#define DISPATCH_FLOATING(DTYPE, BODY)                                   \
+  switch (DTYPE) {                                                       \
+    case at::ScalarType::BFloat16: {                                     \
+      using scalar_t = __hip_bfloat16;                                   \
+      BODY;                                                              \
+      break;                                                             \
+    }                                                                    \
+    case at::ScalarType::Half: {                                         \
+      using scalar_t = __half;                                           \
+      BODY;                                                              \
+      break;                                                             \
+    }                                                                    \
+    case at::ScalarType::Float: {                                        \
+      using scalar_t = float;                                            \
+      BODY;                                                              \
+      break;                                                             \
+    }                                                                    \
+    default:                                                             \
+      TORCH_CHECK(false, "unsupported dtype");                           \
+  }
+
+void rms_norm(torch::Tensor out, torch::Tensor input, torch::Tensor weight,
+              double eps) {
+  const int hidden = input.size(-1);
+  const int rows = input.numel() / hidden;
+  TORCH_CHECK(input.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  DISPATCH_FLOATING(input.scalar_type(), {
+    launch_rms_norm<scalar_t>(ptr<scalar_t>(out), cptr<scalar_t>(input),
+                              cptr<scalar_t>(weight), (float)eps, rows, hidden,
+                              current_stream());
+  });
+}
+
+void fused_add_rms_norm(torch::Tensor x, torch::Tensor residual,
+                        torch::Tensor weight, double eps) {
+  const int hidden = x.size(-1);
+  const int rows = x.numel() / hidden;
+  TORCH_CHECK(x.is_contiguous() && residual.is_contiguous());
+  DISPATCH_FLOATING(x.scalar_type(), {
+    launch_fused_add_rms_norm<scalar_t>(
+        ptr<scalar_t>(x), ptr<scalar_t>(residual), cptr<scalar_t>(weight),
+        (float)eps, rows, hidden, current_stream());
+  });
+}
+
+void silu_and_mul(torch::Tensor out, torch::Tensor x) {
+  const int d = out.size(-1);
+  const int rows = out.numel() / d;
+  TORCH_CHECK(x.size(-1) == 2 * d);
+  TORCH_CHECK(d % 8 == 0, "intermediate size must be a multiple of 8");
+  DISPATCH_FLOATING(x.scalar_type(), {
+    launch_silu_and_mul<scalar_t>(ptr<scalar_t>(out), cptr<scalar_t>(x), rows,
+                                  d, current_stream());
+  });
+}
+
+void rotary_embedding(torch::Tensor positions, torch::Tensor q,
+                      torch::Tensor k, int64_t head_dim,
+                      torch::Tensor cos_sin_cache) {
+  const int tokens = positions.size(0);
+  const int nq = q.size(-1) / head_dim;
+  const int nk = k.size(-1) / head_dim;
+  TORCH_CHECK(positions.scalar_type() == at::ScalarType::Long);
+  TORCH_CHECK(cos_sin_cache.scalar_type() == at::ScalarType::Float);
+  DISPATCH_FLOATING(q.scalar_type(), {
+    launch_rotary_embedding<scalar_t>(
+        positions.data_ptr<long>(), ptr<scalar_t>(q), ptr<scalar_t>(k),
+        cos_sin_cache.data_ptr<float>(), tokens, nq, nk, (int)head_dim,
+        current_stream());
+  });
+}
+
+void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
+                       torch::Tensor v_cache, torch::Tensor slot_mapping) {
+  const int tokens = k.size(0);
+  const int row_elems = k_cache.size(2) * k_cache.size(3);
+  TORCH_CHECK(slot_mapping.scalar_type() == at::ScalarType::Long);
+  DISPATCH_FLOATING(k.scalar_type(), {
+    launch_reshape_and_cache<scalar_t>(
+        cptr<scalar_t>(k), cptr<scalar_t>(v), ptr<scalar_t>(k_cache),
+        ptr<scalar_t>(v_cache), slot_mapping.data_ptr<long>(), tokens,
+        row_elems, current_stream());
+  });
+}
+
+void paged_attention_decode(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor block_tables, torch::Tensor seq_lens,
+                            double scale) {
+  const int nseq = q.size(0);
+  const int nheads = q.size(1);
+  const int head_dim = q.size(2);
+  const int kvh = k_cache.size(2);
+  const int block_size = k_cache.size(1);
+  const int max_blocks = block_tables.size(1);
+  TORCH_CHECK(block_tables.scalar_type() == at::ScalarType::Int);
+  TORCH_CHECK(seq_lens.scalar_type() == at::ScalarType::Int);
+  DISPATCH_FLOATING(q.scalar_type(), {
+    launch_paged_decode<scalar_t>(
+        ptr<scalar_t>(out), cptr<scalar_t>(q), cptr<scalar_t>(k_cache),
+        cptr<scalar_t>(v_cache), block_tables.data_ptr<int>(),
+        seq_lens.data_ptr<int>(), (float)scale, nseq, nheads, kvh, head_dim,
+        block_size, max_blocks, current_stream());
+  });
+}
+
+void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
+                             torch::Tensor k_cache, torch::Tensor v_cache,
+                             torch::Tensor block_tables,
+                             torch::Tensor query_start_loc,
+                             torch::Tensor seq_lens, double scale,
+                             int64_t max_query_len) {
+  const int nseq = seq_lens.size(0);
+  const int nheads = q.size(1);
+  const int head_dim = q.size(2);
+  const int kvh = k_cache.size(2);
+  const int block_size = k_cache.size(1);
+  const int max_blocks = block_tables.size(1);
+  DISPATCH_FLOATING(q.scalar_type(), {
+    launch_paged_prefill<scalar_t>(
+        ptr<scalar_t>(out), cptr<scalar_t>(q), cptr<scalar_t>(k_cache),
+        cptr<scalar_t>(v_cache), block_tables.data_ptr<int>(),
+        query_start_loc.data_ptr<int>(), seq_lens.data_ptr<int>(),
+        (float)scale, nseq, nheads, kvh, head_dim, block_size, max_blocks,
+        (int)max_query_len, current_stream());
+  });
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm", &rms_norm, "fused RMSNorm (CDNA4)");
+  m.def("fused_add_rms_norm", &fused_add_rms_norm,
+        "in-place residual-add + RMSNorm (CDNA4)");
+  m.def("silu_and_mul", &silu_and_mul, "SwiGLU activation (CDNA4)");
+  m.def("rotary_embedding", &rotary_embedding, "NeoX rotary, in place (CDNA4)");
+  m.def("reshape_and_cache", &reshape_and_cache, "paged KV cache write (CDNA4)");
+  m.def("paged_attention_decode", &paged_attention_decode,
+        "paged decode attention (CDNA4)");
+  m.def("paged_attention_prefill", &paged_attention_prefill,
+        "paged causal prefill attention (CDNA4)");
+}
