@@ -201,6 +201,9 @@ class EngineConfig:
     max_loras: int = 8
     max_lora_rank: int = 64
     seed: int = 0
+    # load deterministic rank-independent synthetic weights (TP equivalence
+    # tests) instead of per-shard random init
+    load_synthetic_weights: bool = False
 
     def resolve_device(self) -> str:
         if self.device != "auto":

@@ -58,6 +58,10 @@ class Worker:
         self.model.to(self.device)
         if self.model_config.weights_path:
             self._load_weights()
+        elif config.load_synthetic_weights:
+            from .weights import synth_llama_weights
+
+            self.model.load_weights(synth_llama_weights(self.model_config, config.seed))
 
         self.block_size = config.cache_config.block_size
         self.kv_caches: list[tuple[torch.Tensor, torch.Tensor]] = []
