@@ -21,9 +21,25 @@
 #include "common.h"
 #include <float.h>
 
+// 8-element per-lane fragment: one 16-B load for 16-bit dtypes, two for f32
+template <typename T>
+struct alignas(16) Elems8 {
+  T data[8];
+};
+
+template <typename T>
+DEVINLINE Elems8<T> load8(const T* p) {
+  return *reinterpret_cast<const Elems8<T>*>(p);
+}
+
+template <typename T>
+DEVINLINE void store8(T* p, const Elems8<T>& v) {
+  *reinterpret_cast<Elems8<T>*>(p) = v;
+}
+
 // dot of 8-element slices held per lane
 template <typename T>
-DEVINLINE float dot8(const float* qf, const Vec16<T>& kv) {
+DEVINLINE float dot8(const float* qf, const Elems8<T>& kv) {
   float acc = 0.f;
 #pragma unroll
   for (int j = 0; j < 8; ++j) acc += qf[j] * ScalarOps<T>::to_f32(kv.data[j]);
@@ -65,7 +81,7 @@ __global__ __launch_bounds__(256) void paged_decode_kernel(
 #pragma unroll
   for (int g = 0; g < GROUP; ++g) {
     const T* q_row = q + ((long)seq * nheads + kv_head * GROUP + g) * HEAD_DIM;
-    Vec16<T> v = load16(q_row + dim_off);
+    Elems8<T> v = load8(q_row + dim_off);
 #pragma unroll
     for (int j = 0; j < 8; ++j) qf[g][j] = ScalarOps<T>::to_f32(v.data[j]);
   }
@@ -85,14 +101,14 @@ __global__ __launch_bounds__(256) void paged_decode_kernel(
   for (int base = 0; base < seq_len; base += slots_per_iter) {
     const int pos = base + wave * SPW + slot_in_wave;
     const bool valid = pos < seq_len;
-    Vec16<T> k8, v8;
+    Elems8<T> k8, v8;
     long row = 0;
     if (valid) {
       const int block = btable[pos / block_size];
       row = ((long)block * block_size + pos % block_size) * row_stride +
             (long)kv_head * HEAD_DIM + dim_off;
-      k8 = load16(k_cache + row);
-      v8 = load16(v_cache + row);
+      k8 = load8(k_cache + row);
+      v8 = load8(v_cache + row);
     }
     float p[GROUP];
 #pragma unroll
@@ -225,7 +241,7 @@ __global__ __launch_bounds__(256) void paged_prefill_kernel(
     qpos[r] = seq_len - q_len + qi;
     if (qvalid[r]) {
       const T* q_row = q + ((long)(q_start + qi) * nheads + head) * HEAD_DIM;
-      Vec16<T> v = load16(q_row + dim_off);
+      Elems8<T> v = load8(q_row + dim_off);
 #pragma unroll
       for (int j = 0; j < 8; ++j) qf[r][j] = ScalarOps<T>::to_f32(v.data[j]);
     } else {
@@ -249,13 +265,13 @@ __global__ __launch_bounds__(256) void paged_prefill_kernel(
   for (int base = 0; base + 0 <= max_pos; base += SPW) {
     const int pos = base + slot_in_wave;
     const bool valid = pos <= max_pos && pos < seq_len;
-    Vec16<T> k8, v8;
+    Elems8<T> k8, v8;
     if (valid) {
       const int block = btable[pos / block_size];
       const long row = ((long)block * block_size + pos % block_size) * row_stride +
                        (long)kv_head * HEAD_DIM + dim_off;
-      k8 = load16(k_cache + row);
-      v8 = load16(v_cache + row);
+      k8 = load8(k_cache + row);
+      v8 = load8(v_cache + row);
     }
     float p[QR];
 #pragma unroll
@@ -300,11 +316,11 @@ __global__ __launch_bounds__(256) void paged_prefill_kernel(
     const float inv_l = l[r] > 0.f ? 1.f / l[r] : 0.f;
     if (lane < LPS) {
       T* out_row = out + ((long)(q_start + qi) * nheads + head) * HEAD_DIM;
-      Vec16<T> o;
+      Elems8<T> o;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         o.data[j] = ScalarOps<T>::from_f32(acc[r][j] * inv_l);
-      store16(out_row + dim_off, o);
+      store8(out_row + dim_off, o);
     }
   }
 }

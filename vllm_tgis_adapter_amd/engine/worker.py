@@ -118,7 +118,31 @@ class Worker:
             )
             for _ in range(mc.num_layers)
         ]
+        self._maybe_capture_graphs()
         return self.num_blocks
+
+    def _maybe_capture_graphs(self) -> None:
+        """hipGraph-capture the decode step per batch bucket (E24).
+
+        TP>1 capture (RCCL collectives inside the graph) is gated behind
+        VTA_GRAPH_TP=1 until validated on a multi-GPU box.
+        """
+        import os
+
+        if self.device != "cuda" or self.config.enforce_eager:
+            return
+        if os.environ.get("VTA_FORCE_REFERENCE", "0") == "1":
+            return  # torch-reference ops sync to host; not capturable
+        if self.tp > 1 and os.environ.get("VTA_GRAPH_TP", "0") != "1":
+            return
+        from .graph_runner import DecodeGraphRunner
+
+        self.graph_runner = DecodeGraphRunner(
+            self.model, self.kv_caches, self.model_config, self.block_size,
+            self.config.scheduler_config.max_num_seqs,
+            self.model_config.max_model_len, self.device,
+        )
+        self.graph_runner.capture()
 
     @torch.inference_mode()
     def _profile_peak_memory(self) -> None:
@@ -249,6 +273,22 @@ class Worker:
         ids = torch.tensor(batch["token_ids"], dtype=torch.long, device=dev)
         pos = torch.tensor(batch["positions"], dtype=torch.long, device=dev)
         np_seqs = len(batch["prefill_seq_lens"])
+
+        # hipGraph replay path: pure-decode batch, every row samples, no LoRA
+        n = len(batch["token_ids"])
+        if (
+            self.graph_runner is not None
+            and np_seqs == 0
+            and not batch.get("lora_ids")
+            and batch["num_sample_rows"] == n
+            and len(batch["logit_rows"]) == n
+            and self.graph_runner.bucket_for(n) is not None
+        ):
+            slots = torch.tensor(batch["slot_mapping"], dtype=torch.long, device=dev)
+            seq_lens = torch.tensor(batch["decode_seq_lens"], dtype=torch.int32, device=dev)
+            bt = _pad_block_tables(batch["decode_tables"], dev)
+            return self.graph_runner.run(ids, pos, slots, seq_lens, bt)
+
         npt = batch["qsl"][-1] if np_seqs else 0
         meta = AttnMetadata(
             num_prefill_seqs=np_seqs,
