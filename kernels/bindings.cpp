@@ -29,6 +29,14 @@ template <typename T>
 void launch_paged_prefill(T*, const T*, const T*, const T*, const int*,
                           const int*, const int*, float, int, int, int, int,
                           int, int, int, hipStream_t);
+void launch_mfma_probe_32(const unsigned short*, const unsigned short*, float*,
+                          hipStream_t);
+void launch_mfma_probe_16(const unsigned short*, const unsigned short*, float*,
+                          hipStream_t);
+void launch_paged_prefill_mfma(__hip_bfloat16*, const __hip_bfloat16*,
+                               const __hip_bfloat16*, const __hip_bfloat16*,
+                               const int*, const int*, const int*, float, int,
+                               int, int, int, int, int, int, hipStream_t);
 
 namespace {
 
@@ -164,6 +172,22 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   const int kvh = k_cache.size(2);
   const int block_size = k_cache.size(1);
   const int max_blocks = block_tables.size(1);
+  static const bool force_valu = [] {
+    const char* e = getenv("VTA_PREFILL_VALU");
+    return e && e[0] == '1';
+  }();
+  if (!force_valu && q.scalar_type() == at::ScalarType::BFloat16 &&
+      (head_dim == 64 || head_dim == 128)) {
+    launch_paged_prefill_mfma(
+        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(k_cache.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(v_cache.data_ptr()),
+        block_tables.data_ptr<int>(), query_start_loc.data_ptr<int>(),
+        seq_lens.data_ptr<int>(), (float)scale, nseq, nheads, kvh, head_dim,
+        block_size, max_blocks, (int)max_query_len, current_stream());
+    return;
+  }
   DISPATCH_FLOATING(q.scalar_type(), {
     launch_paged_prefill<scalar_t>(
         ptr<scalar_t>(out), cptr<scalar_t>(q), cptr<scalar_t>(k_cache),
@@ -174,9 +198,21 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   });
 }
 
+void mfma_probe(torch::Tensor a, torch::Tensor b, torch::Tensor d, int64_t shape) {
+  TORCH_CHECK(a.scalar_type() == at::ScalarType::BFloat16);
+  auto* ap = reinterpret_cast<const unsigned short*>(a.data_ptr());
+  auto* bp = reinterpret_cast<const unsigned short*>(b.data_ptr());
+  auto* dp = d.data_ptr<float>();
+  if (shape == 32)
+    launch_mfma_probe_32(ap, bp, dp, current_stream());
+  else
+    launch_mfma_probe_16(ap, bp, dp, current_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("mfma_probe", &mfma_probe, "MFMA fragment layout probe");
   m.def("rms_norm", &rms_norm, "fused RMSNorm (CDNA4)");
   m.def("fused_add_rms_norm", &fused_add_rms_norm,
         "in-place residual-add + RMSNorm (CDNA4)");

@@ -42,6 +42,12 @@ _PRESETS: dict[str, dict] = {
         head_dim=64, rope_theta=500000.0, rms_norm_eps=1e-5,
         max_model_len=8192, tie_word_embeddings=True,
     ),
+    "opt-125m": dict(  # BASELINE config 1 (facebook/opt-125m shapes)
+        architecture="opt", vocab_size=50272, hidden_size=768,
+        intermediate_size=3072, num_layers=12, num_heads=12, num_kv_heads=12,
+        head_dim=64, rope_theta=0.0, rms_norm_eps=1e-5,
+        max_model_len=2048, tie_word_embeddings=True,
+    ),
     "tiny-llama": dict(  # CPU protocol tests
         architecture="llama", vocab_size=2048, hidden_size=64,
         intermediate_size=128, num_layers=2, num_heads=4, num_kv_heads=2,
@@ -144,6 +150,8 @@ def _from_hf_config(model: str, hf: dict) -> ModelConfig:
     arch = "llama"
     if any("Mixtral" in a for a in arch_list):
         arch = "mixtral"
+    elif any("OPT" in a for a in arch_list):
+        arch = "opt"
     elif any("Qwen" in a for a in arch_list):
         arch = "llama"  # qwen2-class maps onto the llama executor (w/ qkv bias)
     num_heads = hf.get("num_attention_heads", 32)

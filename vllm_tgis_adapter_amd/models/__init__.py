@@ -14,4 +14,8 @@ def get_model(cfg: ModelConfig):
         from .mixtral import MixtralForCausalLM
 
         return MixtralForCausalLM(cfg)
+    if cfg.architecture == "opt":
+        from .opt import OPTForCausalLM
+
+        return OPTForCausalLM(cfg)
     raise ValueError(f"Unsupported architecture: {cfg.architecture}")
