@@ -111,7 +111,9 @@ def main():
                 times.append(time.perf_counter())
             elif kind == "elapsed":
                 # contribute this rank's elapsed to the all-reduce MAX
-                t = torch.tensor([times[-1] - times[-2]], dtype=torch.float64)
+                # (nccl needs a device tensor)
+                dev = "cuda" if device == "cuda" else "cpu"
+                t = torch.tensor([times[-1] - times[-2]], dtype=torch.float64, device=dev)
                 dist.all_reduce(t, op=dist.ReduceOp.MAX)
             elif kind == "stop":
                 return
@@ -157,7 +159,8 @@ def main():
 
     elapsed = t1 - t0
     if tp > 1:
-        t = torch.tensor([elapsed], dtype=torch.float64)
+        dev = "cuda" if device == "cuda" else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         tp_broadcast_object(("elapsed",))
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
