@@ -37,6 +37,12 @@ void launch_paged_prefill_mfma(__hip_bfloat16*, const __hip_bfloat16*,
                                const __hip_bfloat16*, const __hip_bfloat16*,
                                const int*, const int*, const int*, float, int,
                                int, int, int, int, int, int, hipStream_t);
+int decode_mfma_num_partitions(int nseq, int kvh, int max_context);
+void launch_paged_decode_mfma(__hip_bfloat16*, float*, float*,
+                              const __hip_bfloat16*, const __hip_bfloat16*,
+                              const __hip_bfloat16*, const int*, const int*,
+                              float, int, int, int, int, int, int, int,
+                              hipStream_t);
 
 namespace {
 
@@ -151,6 +157,30 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   const int max_blocks = block_tables.size(1);
   TORCH_CHECK(block_tables.scalar_type() == at::ScalarType::Int);
   TORCH_CHECK(seq_lens.scalar_type() == at::ScalarType::Int);
+  const int group = nheads / kvh;
+  static const bool force_valu = [] {
+    const char* e = getenv("VTA_DECODE_VALU");
+    return e && e[0] == '1';
+  }();
+  if (!force_valu && q.scalar_type() == at::ScalarType::BFloat16 &&
+      (head_dim == 64 || head_dim == 128) && group >= 1 && group <= 8 &&
+      nheads == kvh * group && nseq > 0) {
+    const int npart = decode_mfma_num_partitions(nseq, kvh, max_blocks * block_size);
+    const int P = npart * 4;
+    auto opts = q.options().dtype(at::ScalarType::Float);
+    auto part_acc = at::empty({nseq, nheads, P, head_dim}, opts);
+    auto part_ml = at::empty({nseq, nheads, P, 2}, opts);
+    launch_paged_decode_mfma(
+        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+        part_acc.data_ptr<float>(), part_ml.data_ptr<float>(),
+        reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(k_cache.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(v_cache.data_ptr()),
+        block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), (float)scale,
+        nseq, nheads, kvh, head_dim, block_size, max_blocks, npart,
+        current_stream());
+    return;
+  }
   DISPATCH_FLOATING(q.scalar_type(), {
     launch_paged_decode<scalar_t>(
         ptr<scalar_t>(out), cptr<scalar_t>(q), cptr<scalar_t>(k_cache),
