@@ -89,34 +89,17 @@ __global__ __launch_bounds__(256, 2) void paged_decode_mfma_kernel(
 #pragma unroll
   for (int dt = 0; dt < DT; ++dt) acc_o[dt] = f32x16_t{};
 
+  constexpr int VPASS = KVT * HEAD_DIM / 8 / 64;  // V staging passes per wave
+
   for (int t = tile_lo + wave; t < tile_hi; t += NWAVES) {
     const int kv_base = t * KVT;
 
-    // ---- stage this wave's V tile into LDS (coalesced 16B per lane) ------
-    {
-      constexpr int LPR = HEAD_DIM / 8;        // lanes per row
-      constexpr int ROWS_PER_PASS = 64 / LPR;  // rows per pass per wave
-      const int r_in_pass = lane / LPR;
-      const int d8 = (lane % LPR) * 8;
-#pragma unroll
-      for (int pass = 0; pass < KVT / ROWS_PER_PASS; ++pass) {
-        const int s = pass * ROWS_PER_PASS + r_in_pass;
-        const int pos = kv_base + s;
-        bf16x8_t vv{};
-        if (pos < seq_len) {
-          const int block = btable[pos / block_size];
-          const long row = ((long)block * block_size + pos % block_size) *
-                               kv_row_stride +
-                           (long)kv_head * HEAD_DIM + d8;
-          vv = *reinterpret_cast<const bf16x8_t*>(v_cache + row);
-        }
-        *reinterpret_cast<bf16x8_t*>(v_lds + s * HEAD_DIM + d8) = vv;
-      }
-    }
-
-    // ---- K A-fragments straight from HBM: lane = kv row (col) ------------
-    // Per lane: row kv_base+col, chunks ks*16 + half*8; the row's 256 B
-    // (HD=128) is covered by the ks sweep, so L2 absorbs the split.
+    // ---- issue ALL of this tile's HBM loads first (K frags + V rows) -----
+    // K is needed first (QK^T mfma), so its 8 loads go out before V's; the
+    // mfma chain then only waits on vmcnt covering K while V stays in
+    // flight, and the V->LDS writes drain afterwards.  Keeping every load
+    // of the tile outstanding together (instead of load->ds_write per
+    // pass) collapses ~9 HBM round trips per tile into ~1.
     bf16x8_t ka[KCH];
     {
       const int pos = kv_base + col;
@@ -131,6 +114,31 @@ __global__ __launch_bounds__(256, 2) void paged_decode_mfma_kernel(
       } else {
 #pragma unroll
         for (int ks = 0; ks < KCH; ++ks) ka[ks] = bf16x8_t{};
+      }
+    }
+    bf16x8_t vv[VPASS];
+    {
+      constexpr int LPR = HEAD_DIM / 8;        // lanes per row
+      constexpr int ROWS_PER_PASS = 64 / LPR;  // rows per pass per wave
+      const int r_in_pass = lane / LPR;
+      const int d8 = (lane % LPR) * 8;
+#pragma unroll
+      for (int pass = 0; pass < VPASS; ++pass) {
+        const int s = pass * ROWS_PER_PASS + r_in_pass;
+        const int pos = kv_base + s;
+        vv[pass] = bf16x8_t{};
+        if (pos < seq_len) {
+          const int block = btable[pos / block_size];
+          const long row = ((long)block * block_size + pos % block_size) *
+                               kv_row_stride +
+                           (long)kv_head * HEAD_DIM + d8;
+          vv[pass] = *reinterpret_cast<const bf16x8_t*>(v_cache + row);
+        }
+      }
+#pragma unroll
+      for (int pass = 0; pass < VPASS; ++pass) {
+        const int s = pass * ROWS_PER_PASS + r_in_pass;
+        *reinterpret_cast<bf16x8_t*>(v_lds + s * HEAD_DIM + d8) = vv[pass];
       }
     }
 
