@@ -38,6 +38,8 @@ void launch_paged_prefill_mfma(__hip_bfloat16*, const __hip_bfloat16*,
                                const int*, const int*, const int*, float, int,
                                int, int, int, int, int, int, hipStream_t);
 int decode_mfma_num_partitions(int nseq, int kvh, int max_context);
+void launch_gemm_skinny(__hip_bfloat16*, const __hip_bfloat16*,
+                        const __hip_bfloat16*, int, int, int, hipStream_t);
 void launch_paged_decode_mfma(__hip_bfloat16*, float*, float*,
                               const __hip_bfloat16*, const __hip_bfloat16*,
                               const __hip_bfloat16*, const int*, const int*,
@@ -228,6 +230,21 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   });
 }
 
+void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
+  const int M = x.size(0);
+  const int K = x.size(1);
+  const int N = w.size(0);
+  TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16 &&
+              w.scalar_type() == at::ScalarType::BFloat16);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 16 == 0 && K % 128 == 0);
+  TORCH_CHECK(w.size(1) == K && y.size(0) == M && y.size(1) == N);
+  launch_gemm_skinny(reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()), M,
+                     N, K, current_stream());
+}
+
 void mfma_probe(torch::Tensor a, torch::Tensor b, torch::Tensor d, int64_t shape) {
   TORCH_CHECK(a.scalar_type() == at::ScalarType::BFloat16);
   auto* ap = reinterpret_cast<const unsigned short*>(a.data_ptr());
@@ -253,4 +270,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "paged decode attention (CDNA4)");
   m.def("paged_attention_prefill", &paged_attention_prefill,
         "paged causal prefill attention (CDNA4)");
+  m.def("gemm_skinny", &gemm_skinny,
+        "skinny decode GEMM y = x @ w^T, M <= 64 (CDNA4 MFMA)");
 }

@@ -262,3 +262,21 @@ print('TOKENS', final.outputs[0].token_ids)
         raise AssertionError(out.stdout)
 
     assert toks(native) == toks(ref)
+
+
+@pytest.mark.parametrize(("m", "n", "k"), [
+    (1, 4096, 4096), (8, 6144, 4096), (17, 4096, 14336),
+    (33, 1024, 128), (64, 28672, 4096), (64, 128256, 4096),
+])
+def test_gemm_skinny(m, n, k):
+    from vllm_tgis_adapter_amd import ops
+
+    torch.manual_seed(7)
+    x = torch.randn(m, k, dtype=torch.bfloat16, device="cuda") / 8
+    w = torch.randn(n, k, dtype=torch.bfloat16, device="cuda") / 8
+    out = ops.linear(x, w)
+    ref = (x.float() @ w.float().t())
+    assert out.shape == (m, n)
+    assert torch.allclose(out.float(), ref, atol=0.35, rtol=2e-2), (
+        (out.float() - ref).abs().max().item()
+    )

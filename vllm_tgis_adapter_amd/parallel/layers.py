@@ -12,6 +12,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .. import ops
 from . import divide, get_tp_rank, get_tp_world_size, tp_all_gather, tp_all_reduce
 
 
@@ -37,7 +38,7 @@ class ColumnParallelLinear(nn.Module):
         self.bias = _init_weight((self.out_per_rank,), dtype, std=0.0) if bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.linear(x, self.weight, self.bias)
+        return ops.linear(x, self.weight, self.bias)
 
     def load_full_weight(self, w: torch.Tensor) -> None:
         r = get_tp_rank()
@@ -64,7 +65,7 @@ class MergedColumnParallelLinear(nn.Module):
         self.bias = _init_weight((total,), dtype, std=0.0) if bias else None
 
     def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
-        y = F.linear(x, self.weight, self.bias)
+        y = ops.linear(x, self.weight, self.bias)
         if lora is not None:
             from ..engine.lora import apply_lora
 
@@ -94,7 +95,7 @@ class RowParallelLinear(nn.Module):
         self.bias = _init_weight((out_size,), dtype, std=0.0) if bias else None
 
     def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
-        y = F.linear(x, self.weight)
+        y = ops.linear(x, self.weight)
         if lora is not None:
             # delta added BEFORE the all-reduce: A is input-sharded, so the
             # per-rank partial deltas sum to the full LoRA delta
@@ -155,7 +156,7 @@ class ParallelLMHead(nn.Module):
         self.weight = _init_weight((self.per_rank, hidden), dtype, std=0.02)
 
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
-        logits = F.linear(hidden.to(self.weight.dtype), self.weight)
+        logits = ops.linear(hidden.to(self.weight.dtype), self.weight)
         logits = tp_all_gather(logits, dim=-1)
         return logits[..., : self.vocab_size]
 
