@@ -40,6 +40,9 @@ void launch_paged_prefill_mfma(__hip_bfloat16*, const __hip_bfloat16*,
 int decode_mfma_num_partitions(int nseq, int kvh, int max_context);
 void launch_gemm_skinny(__hip_bfloat16*, const __hip_bfloat16*,
                         const __hip_bfloat16*, int, int, int, hipStream_t);
+void launch_gemm_skinny_gated(__hip_bfloat16*, const __hip_bfloat16*,
+                              const __hip_bfloat16*, int, int, int,
+                              hipStream_t);
 void launch_paged_decode_mfma(__hip_bfloat16*, float*, float*,
                               const __hip_bfloat16*, const __hip_bfloat16*,
                               const __hip_bfloat16*, const int*, const int*,
@@ -245,6 +248,23 @@ void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
                      N, K, current_stream());
 }
 
+void gemm_skinny_gated(torch::Tensor y, torch::Tensor x, torch::Tensor w13) {
+  const int M = x.size(0);
+  const int K = x.size(1);
+  const int I = w13.size(0) / 2;
+  TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16 &&
+              w13.scalar_type() == at::ScalarType::BFloat16);
+  TORCH_CHECK(x.is_contiguous() && w13.is_contiguous() && y.is_contiguous());
+  TORCH_CHECK(M >= 1 && M <= 64 && I % 16 == 0 && K % 128 == 0);
+  TORCH_CHECK(w13.size(0) % 2 == 0 && w13.size(1) == K);
+  TORCH_CHECK(y.size(0) == M && y.size(1) == I);
+  launch_gemm_skinny_gated(
+      reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(w13.data_ptr()), M, I, K,
+      current_stream());
+}
+
 void mfma_probe(torch::Tensor a, torch::Tensor b, torch::Tensor d, int64_t shape) {
   TORCH_CHECK(a.scalar_type() == at::ScalarType::BFloat16);
   auto* ap = reinterpret_cast<const unsigned short*>(a.data_ptr());
@@ -272,4 +292,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "paged causal prefill attention (CDNA4)");
   m.def("gemm_skinny", &gemm_skinny,
         "skinny decode GEMM y = x @ w^T, M <= 64 (CDNA4 MFMA)");
+  m.def("gemm_skinny_gated", &gemm_skinny_gated,
+        "fused gate/up skinny GEMM + SiLU-mul, M <= 64 (CDNA4 MFMA)");
 }

@@ -101,3 +101,93 @@ void launch_gemm_skinny(__hip_bfloat16* y, const __hip_bfloat16* x,
   else
     hipLaunchKernelGGL(gemm_skinny_kernel<4>, grid, block, 0, stream, y, x, w, M, N, K);
 }
+
+// ---------------------------------------------------------------------------
+// Gated variant for the SwiGLU MLP up-projection: with w13 = [gate; up]
+// ([2I, K], MergedColumnParallelLinear layout), computes
+//   y[m, i] = silu(x·Wg^T)[m, i] * (x·Wu^T)[m, i]        (y: [M, I])
+// in one pass: each workgroup streams the gate tile AND the matching up tile
+// (sharing the x B-fragments), applies the activation in the epilogue, and
+// never materialises the [M, 2I] intermediate.
+// ---------------------------------------------------------------------------
+
+template <int MT>
+__global__ __launch_bounds__(256, 4) void gemm_skinny_gated_kernel(
+    __hip_bfloat16* __restrict__ y,          // [M, I]
+    const __hip_bfloat16* __restrict__ x,    // [M, K]
+    const __hip_bfloat16* __restrict__ w13,  // [2I, K]
+    const int M,
+    const int I,
+    const int K) {
+  const int n0 = blockIdx.x * 16;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int row = lane & 15;
+  const int kq = lane >> 4;
+
+  const int k_per_wave = K / GS_NWAVES;
+  const int k_lo = wave * k_per_wave;
+
+  const __hip_bfloat16* wg_row = w13 + (long)(n0 + row) * K + k_lo + kq * 8;
+  const __hip_bfloat16* wu_row = w13 + (long)(I + n0 + row) * K + k_lo + kq * 8;
+  f32x4_t acc_g[MT], acc_u[MT];
+#pragma unroll
+  for (int mt = 0; mt < MT; ++mt) {
+    acc_g[mt] = f32x4_t{};
+    acc_u[mt] = f32x4_t{};
+  }
+  const bool xrow_ok[4] = {
+      0 * 16 + (lane & 15) < M, 1 * 16 + (lane & 15) < M,
+      2 * 16 + (lane & 15) < M, 3 * 16 + (lane & 15) < M};
+
+  for (int k = 0; k < k_per_wave; k += 32) {
+    bf16x8_t ag = *reinterpret_cast<const bf16x8_t*>(wg_row + k);
+    bf16x8_t au = *reinterpret_cast<const bf16x8_t*>(wu_row + k);
+#pragma unroll
+    for (int mt = 0; mt < MT; ++mt) {
+      bf16x8_t b{};
+      if (xrow_ok[mt]) {
+        b = *reinterpret_cast<const bf16x8_t*>(
+            x + (long)(mt * 16 + row) * K + k_lo + k + kq * 8);
+      }
+      acc_g[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ag, b, acc_g[mt], 0, 0, 0);
+      acc_u[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(au, b, acc_u[mt], 0, 0, 0);
+    }
+  }
+
+  __shared__ float lds_g[GS_NWAVES][MT * 16][16];
+  __shared__ float lds_u[GS_NWAVES][MT * 16][16];
+#pragma unroll
+  for (int mt = 0; mt < MT; ++mt) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      lds_g[wave][mt * 16 + row][kq * 4 + r] = acc_g[mt][r];
+      lds_u[wave][mt * 16 + row][kq * 4 + r] = acc_u[mt][r];
+    }
+  }
+  __syncthreads();
+
+  for (int idx = tid; idx < MT * 16 * 16; idx += 256) {
+    const int m = idx >> 4;
+    const int n = idx & 15;
+    if (m >= M) continue;
+    const float g = lds_g[0][m][n] + lds_g[1][m][n] + lds_g[2][m][n] + lds_g[3][m][n];
+    const float u = lds_u[0][m][n] + lds_u[1][m][n] + lds_u[2][m][n] + lds_u[3][m][n];
+    const float act = g / (1.f + __expf(-g));
+    y[(long)m * I + n0 + n] = __float2bfloat16(act * u);
+  }
+}
+
+void launch_gemm_skinny_gated(__hip_bfloat16* y, const __hip_bfloat16* x,
+                              const __hip_bfloat16* w13, int M, int I, int K,
+                              hipStream_t stream) {
+  dim3 grid(I / 16);
+  dim3 block(256);
+  if (M <= 16)
+    hipLaunchKernelGGL(gemm_skinny_gated_kernel<1>, grid, block, 0, stream, y, x, w13, M, I, K);
+  else if (M <= 32)
+    hipLaunchKernelGGL(gemm_skinny_gated_kernel<2>, grid, block, 0, stream, y, x, w13, M, I, K);
+  else
+    hipLaunchKernelGGL(gemm_skinny_gated_kernel<4>, grid, block, 0, stream, y, x, w13, M, I, K);
+}

@@ -280,3 +280,20 @@ def test_gemm_skinny(m, n, k):
     assert torch.allclose(out.float(), ref, atol=0.35, rtol=2e-2), (
         (out.float() - ref).abs().max().item()
     )
+
+
+@pytest.mark.parametrize(("m", "inter", "k"), [(1, 14336, 4096), (64, 14336, 4096), (40, 128, 256)])
+def test_gemm_skinny_gated(m, inter, k):
+    from vllm_tgis_adapter_amd import ops
+
+    torch.manual_seed(8)
+    x = torch.randn(m, k, dtype=torch.bfloat16, device="cuda") / 8
+    w13 = torch.randn(2 * inter, k, dtype=torch.bfloat16, device="cuda") / 8
+    out = ops.gated_mlp_up(x, w13)
+    assert out is not None
+    g = x.float() @ w13[:inter].float().t()
+    u = x.float() @ w13[inter:].float().t()
+    ref = torch.nn.functional.silu(g) * u
+    assert torch.allclose(out.float(), ref, atol=0.3, rtol=2e-2), (
+        (out.float() - ref).abs().max().item()
+    )

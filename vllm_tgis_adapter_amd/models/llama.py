@@ -114,6 +114,12 @@ class LlamaMLP(nn.Module):
         self._down_lora = (layer_idx, [("down_proj", 0, cfg.hidden_size)])
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from ..engine.lora import CTX as _lora_ctx
+
+        if not _lora_ctx.active:
+            h = ops.gated_mlp_up(x, self.gate_up.weight)
+            if h is not None:
+                return self.down(h, lora=self._down_lora)
         h = ops.silu_and_mul(self.gate_up(x, lora=self._gu_lora))
         return self.down(h, lora=self._down_lora)
 

@@ -43,14 +43,14 @@ class MoEBlock(nn.Module):
         flat_ids = ids.reshape(-1)                        # [T*k]
         flat_w = weights.reshape(-1).to(x.dtype)
         token_idx = torch.arange(t, device=x.device).repeat_interleave(self.top_k)
+        # Launch-only expert loop: no data-dependent host branches (`any()`
+        # would sync per expert and break hipGraph capture); empty selections
+        # run zero-row GEMMs, which are free.
         for e in range(self.num_experts):
-            sel = flat_ids == e
-            if not torch.any(sel):
-                continue
+            sel = (flat_ids == e).nonzero(as_tuple=True)[0]
             rows = token_idx[sel]
             xe = x[rows]
-            h = F.linear(xe, self.w13[e])
-            h = ops.silu_and_mul(h)
+            h = ops.silu_and_mul(F.linear(xe, self.w13[e]))
             ye = F.linear(h, self.w2[e])
             out.index_add_(0, rows, ye * flat_w[sel, None])
         return tp_all_reduce(out)

@@ -78,6 +78,8 @@ def fused_add_rms_norm(
 
 
 def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if x.numel() == 0:
+        return x[..., : x.shape[-1] // 2].clone()
     if _native(x):
         d = x.shape[-1] // 2
         out = torch.empty(*x.shape[:-1], d, dtype=x.dtype, device=x.device)
@@ -190,6 +192,31 @@ def linear(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None = No
         _C.gemm_skinny(out, x, weight)
         return out
     return torch.nn.functional.linear(x, weight, bias)
+
+
+def gated_mlp_up(x: torch.Tensor, w13: torch.Tensor) -> torch.Tensor | None:
+    """Fused silu(x@Wg^T) * (x@Wu^T) for merged [gate; up] weights.
+
+    Returns None when the fused CDNA4 path doesn't apply (caller falls back
+    to linear + silu_and_mul).
+    """
+    if (
+        x.dim() == 2
+        and x.shape[0] >= 1
+        and x.shape[0] <= 64
+        and x.dtype == torch.bfloat16
+        and w13.dtype == torch.bfloat16
+        and w13.shape[0] % 32 == 0
+        and x.shape[1] % 128 == 0
+        and x.is_contiguous()
+        and w13.is_contiguous()
+        and _native(x)
+    ):
+        inter = w13.shape[0] // 2
+        out = torch.empty((x.shape[0], inter), dtype=x.dtype, device=x.device)
+        _C.gemm_skinny_gated(out, x, w13)
+        return out
+    return None
 
 
 def topk_softmax(gate_logits: torch.Tensor, top_k: int):
