@@ -240,7 +240,7 @@ void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
   TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16 &&
               w.scalar_type() == at::ScalarType::BFloat16);
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
-  TORCH_CHECK(M >= 1 && M <= 64 && N % 16 == 0 && K % 128 == 0);
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 16 == 0 && K % 512 == 0);
   TORCH_CHECK(w.size(1) == K && y.size(0) == M && y.size(1) == N);
   launch_gemm_skinny(reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
                      reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
@@ -255,7 +255,7 @@ void gemm_skinny_gated(torch::Tensor y, torch::Tensor x, torch::Tensor w13) {
   TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16 &&
               w13.scalar_type() == at::ScalarType::BFloat16);
   TORCH_CHECK(x.is_contiguous() && w13.is_contiguous() && y.is_contiguous());
-  TORCH_CHECK(M >= 1 && M <= 64 && I % 16 == 0 && K % 128 == 0);
+  TORCH_CHECK(M >= 1 && M <= 64 && I % 16 == 0 && K % 512 == 0);
   TORCH_CHECK(w13.size(0) % 2 == 0 && w13.size(1) == K);
   TORCH_CHECK(y.size(0) == M && y.size(1) == I);
   launch_gemm_skinny_gated(

@@ -25,7 +25,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 #define GS_NWAVES 4
 
 template <int MT>  // number of 16-row m tiles (M <= MT*16)
-__global__ __launch_bounds__(256, 4) void gemm_skinny_kernel(
+__global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
     __hip_bfloat16* __restrict__ y,        // [M, N]
     const __hip_bfloat16* __restrict__ x,  // [M, K]
     const __hip_bfloat16* __restrict__ w,  // [N, K]
@@ -54,16 +54,33 @@ __global__ __launch_bounds__(256, 4) void gemm_skinny_kernel(
       0 * 16 + (lane & 15) < M, 1 * 16 + (lane & 15) < M,
       2 * 16 + (lane & 15) < M, 3 * 16 + (lane & 15) < M};
 
-  for (int k = 0; k < k_per_wave; k += 32) {
-    bf16x8_t a = *reinterpret_cast<const bf16x8_t*>(w_row + k);
+  // K unrolled by U with all loads issued ahead of the mfmas: one wave per
+  // SIMD must keep ~U*(1+MT) loads in flight to cover HBM latency (the r1
+  // non-unrolled version ran at ~1/3 of hipBLASLt for the N=4096 shapes).
+  constexpr int U = 4;
+  static_assert(true, "");
+  for (int k = 0; k < k_per_wave; k += 32 * U) {
+    bf16x8_t a[U];
 #pragma unroll
-    for (int mt = 0; mt < MT; ++mt) {
-      bf16x8_t b{};
-      if (xrow_ok[mt]) {
-        b = *reinterpret_cast<const bf16x8_t*>(
-            x + (long)(mt * 16 + row) * K + k_lo + k + kq * 8);
+    for (int u = 0; u < U; ++u)
+      a[u] = *reinterpret_cast<const bf16x8_t*>(w_row + k + u * 32);
+    bf16x8_t b[U][MT];
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+#pragma unroll
+      for (int mt = 0; mt < MT; ++mt) {
+        b[u][mt] = bf16x8_t{};
+        if (xrow_ok[mt]) {
+          b[u][mt] = *reinterpret_cast<const bf16x8_t*>(
+              x + (long)(mt * 16 + row) * K + k_lo + k + u * 32 + kq * 8);
+        }
       }
-      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mt], 0, 0, 0);
+    }
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+#pragma unroll
+      for (int mt = 0; mt < MT; ++mt)
+        acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b[u][mt], acc[mt], 0, 0, 0);
     }
   }
 
@@ -112,7 +129,7 @@ void launch_gemm_skinny(__hip_bfloat16* y, const __hip_bfloat16* x,
 // ---------------------------------------------------------------------------
 
 template <int MT>
-__global__ __launch_bounds__(256, 4) void gemm_skinny_gated_kernel(
+__global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
     __hip_bfloat16* __restrict__ y,          // [M, I]
     const __hip_bfloat16* __restrict__ x,    // [M, K]
     const __hip_bfloat16* __restrict__ w13,  // [2I, K]
@@ -141,18 +158,33 @@ __global__ __launch_bounds__(256, 4) void gemm_skinny_gated_kernel(
       0 * 16 + (lane & 15) < M, 1 * 16 + (lane & 15) < M,
       2 * 16 + (lane & 15) < M, 3 * 16 + (lane & 15) < M};
 
-  for (int k = 0; k < k_per_wave; k += 32) {
-    bf16x8_t ag = *reinterpret_cast<const bf16x8_t*>(wg_row + k);
-    bf16x8_t au = *reinterpret_cast<const bf16x8_t*>(wu_row + k);
+  constexpr int U = 4;
+  for (int k = 0; k < k_per_wave; k += 32 * U) {
+    bf16x8_t ag[U], au[U];
 #pragma unroll
-    for (int mt = 0; mt < MT; ++mt) {
-      bf16x8_t b{};
-      if (xrow_ok[mt]) {
-        b = *reinterpret_cast<const bf16x8_t*>(
-            x + (long)(mt * 16 + row) * K + k_lo + k + kq * 8);
+    for (int u = 0; u < U; ++u) {
+      ag[u] = *reinterpret_cast<const bf16x8_t*>(wg_row + k + u * 32);
+      au[u] = *reinterpret_cast<const bf16x8_t*>(wu_row + k + u * 32);
+    }
+    bf16x8_t b[U][MT];
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+#pragma unroll
+      for (int mt = 0; mt < MT; ++mt) {
+        b[u][mt] = bf16x8_t{};
+        if (xrow_ok[mt]) {
+          b[u][mt] = *reinterpret_cast<const bf16x8_t*>(
+              x + (long)(mt * 16 + row) * K + k_lo + k + u * 32 + kq * 8);
+        }
       }
-      acc_g[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ag, b, acc_g[mt], 0, 0, 0);
-      acc_u[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(au, b, acc_u[mt], 0, 0, 0);
+    }
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+#pragma unroll
+      for (int mt = 0; mt < MT; ++mt) {
+        acc_g[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ag[u], b[u][mt], acc_g[mt], 0, 0, 0);
+        acc_u[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(au[u], b[u][mt], acc_u[mt], 0, 0, 0);
+      }
     }
   }
 

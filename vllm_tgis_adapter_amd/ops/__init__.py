@@ -182,7 +182,7 @@ def linear(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None = No
         and weight.dtype == torch.bfloat16
         and 1 <= x.shape[0] <= 64
         and weight.shape[0] % 16 == 0
-        and x.shape[1] % 128 == 0
+        and x.shape[1] % 512 == 0
         and x.is_contiguous()
         and weight.is_contiguous()
     ):
@@ -207,7 +207,7 @@ def gated_mlp_up(x: torch.Tensor, w13: torch.Tensor) -> torch.Tensor | None:
         and x.dtype == torch.bfloat16
         and w13.dtype == torch.bfloat16
         and w13.shape[0] % 32 == 0
-        and x.shape[1] % 128 == 0
+        and x.shape[1] % 512 == 0
         and x.is_contiguous()
         and w13.is_contiguous()
         and _native(x)
