@@ -48,6 +48,12 @@ void launch_paged_decode_mfma(__hip_bfloat16*, float*, float*,
                               const __hip_bfloat16*, const int*, const int*,
                               float, int, int, int, int, int, int, int,
                               hipStream_t);
+template <typename T>
+void launch_lora_shrink(float*, const T*, const T*, const int*, int, int, int,
+                        hipStream_t);
+template <typename T>
+void launch_lora_expand(T*, const float*, const T*, const int*, const float*,
+                        int, int, int, int, int, hipStream_t);
 
 namespace {
 
@@ -265,6 +271,36 @@ void gemm_skinny_gated(torch::Tensor y, torch::Tensor x, torch::Tensor w13) {
       current_stream());
 }
 
+void lora_bgmv(torch::Tensor out, torch::Tensor x, torch::Tensor a_stack,
+               torch::Tensor b_stack, torch::Tensor tmp, torch::Tensor slots,
+               torch::Tensor scales, int64_t off) {
+  const int Tn = x.size(0);
+  const int K = x.size(1);
+  const int R = a_stack.size(1);
+  const int N = b_stack.size(1);
+  const int out_w = out.size(1);
+  TORCH_CHECK(out.size(0) == Tn && tmp.size(0) == Tn && tmp.size(1) == R);
+  TORCH_CHECK(a_stack.size(2) == K && b_stack.size(2) == R);
+  TORCH_CHECK(R <= 64 && K % 8 == 0);
+  TORCH_CHECK(off + N <= out_w);
+  TORCH_CHECK(slots.scalar_type() == at::ScalarType::Int);
+  TORCH_CHECK(tmp.scalar_type() == at::ScalarType::Float &&
+              scales.scalar_type() == at::ScalarType::Float);
+  TORCH_CHECK(x.is_contiguous() && out.is_contiguous() &&
+              a_stack.is_contiguous() && b_stack.is_contiguous());
+  DISPATCH_FLOATING(x.scalar_type(), {
+    launch_lora_shrink<scalar_t>(tmp.data_ptr<float>(), cptr<scalar_t>(x),
+                                 cptr<scalar_t>(a_stack),
+                                 slots.data_ptr<int>(), Tn, R, K,
+                                 current_stream());
+    launch_lora_expand<scalar_t>(ptr<scalar_t>(out), tmp.data_ptr<float>(),
+                                 cptr<scalar_t>(b_stack),
+                                 slots.data_ptr<int>(),
+                                 scales.data_ptr<float>(), Tn, R, N, out_w,
+                                 (int)off, current_stream());
+  });
+}
+
 void mfma_probe(torch::Tensor a, torch::Tensor b, torch::Tensor d, int64_t shape) {
   TORCH_CHECK(a.scalar_type() == at::ScalarType::BFloat16);
   auto* ap = reinterpret_cast<const unsigned short*>(a.data_ptr());
@@ -294,4 +330,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "skinny decode GEMM y = x @ w^T, M <= 64 (CDNA4 MFMA)");
   m.def("gemm_skinny_gated", &gemm_skinny_gated,
         "fused gate/up skinny GEMM + SiLU-mul, M <= 64 (CDNA4 MFMA)");
+  m.def("lora_bgmv", &lora_bgmv,
+        "batched multi-LoRA shrink+expand for mixed-adapter batches (CDNA4)");
 }

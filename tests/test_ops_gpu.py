@@ -297,3 +297,36 @@ def test_gemm_skinny_gated(m, inter, k):
     assert torch.allclose(out.float(), ref, atol=0.3, rtol=2e-2), (
         (out.float() - ref).abs().max().item()
     )
+
+
+@pytest.mark.parametrize(("t", "k", "n", "ranks"), [
+    (7, 4096, 6144, [8, 16]), (64, 4096, 4096, [64]), (3, 512, 256, [4, 4, 4]),
+])
+def test_lora_bgmv(t, k, n, ranks):
+    from vllm_tgis_adapter_amd import ops
+
+    torch.manual_seed(11)
+    dt = torch.bfloat16
+    L = len(ranks)
+    rmax = max(ranks)
+    x = torch.randn(t, k, dtype=dt, device="cuda") / 4
+    out = torch.randn(t, n + 32, dtype=dt, device="cuda")
+    ref = out.float().clone()
+    a_stack = torch.zeros(L, rmax, k, dtype=dt, device="cuda")
+    b_stack = torch.zeros(L, n, rmax, dtype=dt, device="cuda")
+    scales = torch.rand(L, dtype=torch.float32, device="cuda") + 0.5
+    for s, r in enumerate(ranks):
+        a_stack[s, :r] = torch.randn(r, k, dtype=dt, device="cuda") / 8
+        b_stack[s, :, :r] = torch.randn(n, r, dtype=dt, device="cuda") / 8
+    slots = torch.randint(-1, L, (t,), dtype=torch.int32, device="cuda")
+    off = 16
+    ops.lora_bgmv(out, x, a_stack, b_stack, slots, scales, off)
+    for i in range(t):
+        s = int(slots[i])
+        if s < 0:
+            continue
+        delta = (x[i].float() @ a_stack[s].float().t()) @ b_stack[s].float().t()
+        ref[i, off:off + n] += delta * scales[s]
+    assert torch.allclose(out.float(), ref, atol=0.25, rtol=3e-2), (
+        (out.float() - ref).abs().max().item()
+    )

@@ -29,17 +29,27 @@ SHAPES = [
 ]
 
 
-def bench_shape(m: int, n: int, k: int, iters: int, dtype=torch.bfloat16) -> float:
+def _fn(native):
+    if native:
+        from vllm_tgis_adapter_amd import ops
+
+        return ops.linear
+    return torch.nn.functional.linear
+
+
+def bench_shape(m: int, n: int, k: int, iters: int, dtype=torch.bfloat16,
+                native: bool = False) -> float:
+    fn = _fn(native)
     x = torch.randn(m, k, dtype=dtype, device="cuda")
     w = torch.randn(n, k, dtype=dtype, device="cuda")
     for _ in range(20):
-        y = torch.nn.functional.linear(x, w)
+        y = fn(x, w)
     torch.cuda.synchronize()
     t0 = torch.cuda.Event(enable_timing=True)
     t1 = torch.cuda.Event(enable_timing=True)
     t0.record()
     for _ in range(iters):
-        y = torch.nn.functional.linear(x, w)
+        y = fn(x, w)
     t1.record()
     torch.cuda.synchronize()
     del y
@@ -50,11 +60,13 @@ def main():
     p = argparse.ArgumentParser()
     p.add_argument("--m", type=int, default=64)
     p.add_argument("--iters", type=int, default=200)
+    p.add_argument("--native", action="store_true",
+                   help="route through ops.linear (custom skinny kernel)")
     args = p.parse_args()
     assert torch.cuda.is_available()
     results = {}
     for name, n, k in SHAPES:
-        ms = bench_shape(args.m, n, k, args.iters)
+        ms = bench_shape(args.m, n, k, args.iters, native=args.native)
         bytes_moved = (args.m * k + n * k + args.m * n) * 2
         gbs = bytes_moved / (ms * 1e-3) / 1e9
         results[name] = {
@@ -64,6 +76,7 @@ def main():
         }
         print(name, results[name], flush=True)
     print(json.dumps({
+        "native": args.native,
         "m": args.m,
         "tunableop": os.environ.get("PYTORCH_TUNABLEOP_ENABLED", "0"),
         "results": results,

@@ -46,6 +46,9 @@ class LoRAContext:
 
     token_lora_ids: Optional[torch.Tensor] = None  # [T] int32 device (0 = none)
     adapters: dict[int, LoRAAdapter] = field(default_factory=dict)
+    # BGMV kernel context (GPU): slot per token (-1 = none) + slot order
+    token_slots: Optional[torch.Tensor] = None  # [T] int32 device
+    slot_ids: tuple[int, ...] = ()
 
     @property
     def active(self) -> bool:
@@ -54,15 +57,60 @@ class LoRAContext:
 
 CTX = LoRAContext()
 
+# (slot_ids, layer, proj, N, dtype) -> (a_stack, b_stack, scales) for lora_bgmv
+_STACK_CACHE: dict = {}
+
 
 def set_context(token_lora_ids: Optional[torch.Tensor], adapters: dict[int, LoRAAdapter]) -> None:
     CTX.token_lora_ids = token_lora_ids
     CTX.adapters = adapters
+    CTX.slot_ids = tuple(sorted(adapters))
+    CTX.token_slots = None
+    if token_lora_ids is not None and adapters:
+        id2slot = {lid: s for s, lid in enumerate(CTX.slot_ids)}
+        slots = torch.full_like(token_lora_ids, -1)
+        for lid, s in id2slot.items():
+            slots[token_lora_ids == lid] = s
+        CTX.token_slots = slots.to(torch.int32)
 
 
 def clear_context() -> None:
     CTX.token_lora_ids = None
     CTX.adapters = {}
+    CTX.token_slots = None
+    CTX.slot_ids = ()
+
+
+def _get_stacks(layer_idx: int, proj: str, n_out: int, in_size: int,
+                dtype: torch.dtype, device) -> tuple:
+    """Stacked, rank-padded (A, B, scales) tensors for the active adapter set."""
+    key = (CTX.slot_ids, layer_idx, proj, n_out, dtype)
+    hit = _STACK_CACHE.get(key)
+    if hit is not None:
+        return hit
+    adapters = [CTX.adapters[lid] for lid in CTX.slot_ids]
+    rmax = max(
+        (a.weights[(layer_idx, proj)][0].shape[0]
+         for a in adapters if (layer_idx, proj) in a.weights),
+        default=0,
+    )
+    if rmax == 0:
+        _STACK_CACHE[key] = None
+        return None
+    L = len(adapters)
+    a_stack = torch.zeros((L, rmax, in_size), dtype=dtype, device=device)
+    b_stack = torch.zeros((L, n_out, rmax), dtype=dtype, device=device)
+    scales = torch.zeros((L,), dtype=torch.float32, device=device)
+    for s, ad in enumerate(adapters):
+        ab = ad.weights.get((layer_idx, proj))
+        scales[s] = ad.scaling
+        if ab is None:
+            continue
+        a, b = ab
+        a_stack[s, : a.shape[0]] = a
+        b_stack[s, :, : b.shape[1]] = b
+    _STACK_CACHE[key] = (a_stack, b_stack, scales)
+    return _STACK_CACHE[key]
 
 
 def apply_lora(
@@ -72,6 +120,26 @@ def apply_lora(
     out: torch.Tensor,   # [T, sum(out_local)] output to update in place
 ) -> torch.Tensor:
     if not CTX.active:
+        return out
+    from .. import ops
+
+    # GPU: batched BGMV shrink/expand HIP kernels over the mixed batch (E12)
+    if (
+        CTX.token_slots is not None
+        and x.device.type == "cuda"
+        and ops.has_native()
+        and x.is_contiguous()
+        and out.is_contiguous()
+        and x.shape[1] % 8 == 0
+        and all(a.rank <= 64 for a in CTX.adapters.values())
+    ):
+        for name, start, end in projs:
+            stacks = _get_stacks(layer_idx, name, end - start, x.shape[1],
+                                 x.dtype, x.device)
+            if stacks is None:
+                continue
+            a_stack, b_stack, scales = stacks
+            ops.lora_bgmv(out, x, a_stack, b_stack, CTX.token_slots, scales, start)
         return out
     token_ids = CTX.token_lora_ids
     for lora_id, adapter in CTX.adapters.items():

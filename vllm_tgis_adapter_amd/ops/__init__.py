@@ -219,6 +219,23 @@ def gated_mlp_up(x: torch.Tensor, w13: torch.Tensor) -> torch.Tensor | None:
     return None
 
 
+def lora_bgmv(
+    out: torch.Tensor,      # [T, out_w] updated in place
+    x: torch.Tensor,        # [T, K]
+    a_stack: torch.Tensor,  # [L, R, K]
+    b_stack: torch.Tensor,  # [L, N, R]
+    slots: torch.Tensor,    # [T] int32, -1 = no adapter
+    scales: torch.Tensor,   # [L] float32
+    off: int,
+) -> None:
+    """Batched LoRA delta for a mixed-adapter batch (GPU only)."""
+    assert _native(x), "lora_bgmv is the GPU path; CPU uses the torch loop"
+    tmp = torch.empty(
+        (x.shape[0], a_stack.shape[1]), dtype=torch.float32, device=x.device
+    )
+    _C.lora_bgmv(out, x, a_stack, b_stack, tmp, slots, scales, off)
+
+
 def topk_softmax(gate_logits: torch.Tensor, top_k: int):
     # Router math is tiny; torch ops are fine on both devices for now.
     return reference.topk_softmax(gate_logits, top_k)
