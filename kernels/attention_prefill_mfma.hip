@@ -61,9 +61,11 @@ __global__ __launch_bounds__(256, 1) void paged_prefill_mfma_kernel(
   const int* btable = block_tables + (long)seq * max_blocks;
   const long kv_row_stride = (long)kvh * HEAD_DIM;
 
-  // ---- LDS tiles -------------------------------------------------------
-  __shared__ __hip_bfloat16 k_lds[KVT * HEAD_DIM];  // XOR-swizzled rows
-  __shared__ __hip_bfloat16 v_lds[KVT * HEAD_DIM];  // linear
+  // ---- LDS tiles, double-buffered: tile t+1 streams in while the mfma
+  // phase reads tile t, so the per-tile HBM round trip is overlapped and
+  // only one __syncthreads per tile remains.
+  __shared__ __hip_bfloat16 k_lds2[2][KVT * HEAD_DIM];  // XOR-swizzled rows
+  __shared__ __hip_bfloat16 v_lds2[2][KVT * HEAD_DIM];  // linear
 
   // ---- load this wave's Q sub-tile as B fragments ----------------------
   // B[k][q]: lane holds Q[q=col][ks*16 + half*8 + j]
@@ -96,35 +98,54 @@ __global__ __launch_bounds__(256, 1) void paged_prefill_mfma_kernel(
   const int wg_max_pos = seq_len - q_len + tile_base + wg_rows - 1;
   const int kv_limit = min(seq_len, wg_max_pos + 1);
 
-  for (int kv_base = 0; kv_base < kv_limit; kv_base += KVT) {
-    // ---- stage K (swizzled) and V (linear) tiles, whole workgroup ------
-    __syncthreads();
-    {
-      constexpr int LPR = HEAD_DIM / 8;       // lanes per row (16B each)
-      constexpr int ROWS_PER_PASS = 256 / LPR;
-      const int r_in_pass = tid / LPR;
-      const int d8 = (tid % LPR) * 8;
+  constexpr int LPR = HEAD_DIM / 8;       // staging lanes per row (16B each)
+  constexpr int ROWS_PER_PASS = 256 / LPR;
+  constexpr int NPASS = KVT / ROWS_PER_PASS;
+  const int r_in_pass = tid / LPR;
+  const int d8 = (tid % LPR) * 8;
+
+  // stage tile (kv_base) into LDS buffer `buf`: phase 0 = issue loads into
+  // regs, phase 1 = LDS writes (so all of a tile's HBM loads are in flight
+  // together)
+  bf16x8_t st_k[NPASS], st_v[NPASS];
+  auto stage_load = [&](int kv_base) {
 #pragma unroll
-      for (int pass = 0; pass < KVT / ROWS_PER_PASS; ++pass) {
-        const int s = pass * ROWS_PER_PASS + r_in_pass;
-        const int pos = kv_base + s;
-        bf16x8_t kv_k{}, kv_v{};
-        if (pos < kv_limit) {
-          const int block = btable[pos / block_size];
-          const long row = ((long)block * block_size + pos % block_size) *
-                               kv_row_stride +
-                           (long)kv_head * HEAD_DIM + d8;
-          kv_k = *reinterpret_cast<const bf16x8_t*>(k_cache + row);
-          kv_v = *reinterpret_cast<const bf16x8_t*>(v_cache + row);
-        }
-        // K row s at byte offset (s*HD + d8)*2, XOR-swizzled within the row
-        const int k_byte = (s * HEAD_DIM + d8) * 2 ^ ((s & 7) << 4);
-        *reinterpret_cast<bf16x8_t*>(
-            reinterpret_cast<char*>(k_lds) + k_byte) = kv_k;
-        *reinterpret_cast<bf16x8_t*>(v_lds + s * HEAD_DIM + d8) = kv_v;
+    for (int pass = 0; pass < NPASS; ++pass) {
+      const int s = pass * ROWS_PER_PASS + r_in_pass;
+      const int pos = kv_base + s;
+      st_k[pass] = bf16x8_t{};
+      st_v[pass] = bf16x8_t{};
+      if (pos < kv_limit) {
+        const int block = btable[pos / block_size];
+        const long row = ((long)block * block_size + pos % block_size) *
+                             kv_row_stride +
+                         (long)kv_head * HEAD_DIM + d8;
+        st_k[pass] = *reinterpret_cast<const bf16x8_t*>(k_cache + row);
+        st_v[pass] = *reinterpret_cast<const bf16x8_t*>(v_cache + row);
       }
     }
-    __syncthreads();
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int pass = 0; pass < NPASS; ++pass) {
+      const int s = pass * ROWS_PER_PASS + r_in_pass;
+      const int k_byte = (s * HEAD_DIM + d8) * 2 ^ ((s & 7) << 4);
+      *reinterpret_cast<bf16x8_t*>(
+          reinterpret_cast<char*>(k_lds2[buf]) + k_byte) = st_k[pass];
+      *reinterpret_cast<bf16x8_t*>(v_lds2[buf] + s * HEAD_DIM + d8) = st_v[pass];
+    }
+  };
+
+  stage_load(0);
+  stage_write(0);
+
+  for (int kv_base = 0; kv_base < kv_limit; kv_base += KVT) {
+    const __hip_bfloat16* k_lds = k_lds2[(kv_base / KVT) & 1];
+    const __hip_bfloat16* v_lds = v_lds2[(kv_base / KVT) & 1];
+    __syncthreads();  // tile (kv_base) fully staged for all waves
+    // issue next tile's loads now; they complete behind the mfma phase
+    const int next_base = kv_base + KVT;
+    if (next_base < kv_limit) stage_load(next_base);
 
     // ---- S^T = K · Q^T -------------------------------------------------
     f32x16_t acc_s{};
@@ -220,6 +241,10 @@ __global__ __launch_bounds__(256, 1) void paged_prefill_mfma_kernel(
         acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[c], b, acc_o[dt], 0, 0, 0);
       }
     }
+
+    // drain next tile's loads into its LDS buffer (safe: the buffer's last
+    // readers finished before this iteration's top __syncthreads)
+    if (next_base < kv_limit) stage_write((next_base / KVT) & 1);
   }
 
   // ---- epilogue: normalize by l (per output q row) and store -----------

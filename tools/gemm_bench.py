@@ -17,6 +17,11 @@ import argparse
 import json
 import os
 
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
 import torch
 
 # (name, N, K) for y[M,N] = x[M,K] @ W[N,K]^T — llama-3-8b TP=1 decode shapes
