@@ -38,10 +38,12 @@ void launch_paged_prefill_mfma(__hip_bfloat16*, const __hip_bfloat16*,
                                const int*, const int*, const int*, float, int,
                                int, int, int, int, int, int, hipStream_t);
 int decode_mfma_num_partitions(int nseq, int kvh, int max_context);
-void launch_gemm_skinny(__hip_bfloat16*, const __hip_bfloat16*,
-                        const __hip_bfloat16*, int, int, int, hipStream_t);
-void launch_gemm_skinny_gated(__hip_bfloat16*, const __hip_bfloat16*,
-                              const __hip_bfloat16*, int, int, int,
+int gemm_skinny_num_ksplit(int N, int K);
+void launch_gemm_skinny(__hip_bfloat16*, float*, const __hip_bfloat16*,
+                        const __hip_bfloat16*, int, int, int, int,
+                        hipStream_t);
+void launch_gemm_skinny_gated(__hip_bfloat16*, float*, const __hip_bfloat16*,
+                              const __hip_bfloat16*, int, int, int, int,
                               hipStream_t);
 void launch_paged_decode_mfma(__hip_bfloat16*, float*, float*,
                               const __hip_bfloat16*, const __hip_bfloat16*,
@@ -246,12 +248,19 @@ void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
   TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16 &&
               w.scalar_type() == at::ScalarType::BFloat16);
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
-  TORCH_CHECK(M >= 1 && M <= 64 && N % 16 == 0 && K % 512 == 0);
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 128 == 0 && K % 64 == 0);
   TORCH_CHECK(w.size(1) == K && y.size(0) == M && y.size(1) == N);
-  launch_gemm_skinny(reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+  const int KS = gemm_skinny_num_ksplit(N, K);
+  torch::Tensor part;
+  float* pp = nullptr;
+  if (KS > 1) {
+    part = at::empty({KS, M, N}, x.options().dtype(at::ScalarType::Float));
+    pp = part.data_ptr<float>();
+  }
+  launch_gemm_skinny(reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), pp,
                      reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
                      reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()), M,
-                     N, K, current_stream());
+                     N, K, KS, current_stream());
 }
 
 void gemm_skinny_gated(torch::Tensor y, torch::Tensor x, torch::Tensor w13) {
@@ -261,13 +270,20 @@ void gemm_skinny_gated(torch::Tensor y, torch::Tensor x, torch::Tensor w13) {
   TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16 &&
               w13.scalar_type() == at::ScalarType::BFloat16);
   TORCH_CHECK(x.is_contiguous() && w13.is_contiguous() && y.is_contiguous());
-  TORCH_CHECK(M >= 1 && M <= 64 && I % 16 == 0 && K % 512 == 0);
+  TORCH_CHECK(M >= 1 && M <= 64 && I % 128 == 0 && K % 64 == 0);
   TORCH_CHECK(w13.size(0) % 2 == 0 && w13.size(1) == K);
   TORCH_CHECK(y.size(0) == M && y.size(1) == I);
+  const int KS = gemm_skinny_num_ksplit(I, K);
+  torch::Tensor part;
+  float* pp = nullptr;
+  if (KS > 1) {
+    part = at::empty({KS, 2, M, I}, x.options().dtype(at::ScalarType::Float));
+    pp = part.data_ptr<float>();
+  }
   launch_gemm_skinny_gated(
-      reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+      reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), pp,
       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
-      reinterpret_cast<const __hip_bfloat16*>(w13.data_ptr()), M, I, K,
+      reinterpret_cast<const __hip_bfloat16*>(w13.data_ptr()), M, I, K, KS,
       current_stream());
 }
 

@@ -1,21 +1,23 @@
-// Skinny decode GEMM (M <= 64): y[M,N] = x[M,K] · W[N,K]^T, bf16 in, bf16 out.
+// Skinny decode GEMM (M <= 64): y[M,N] = x[M,K] · W[N,K]^T, bf16 in/out.
 //
 // hipBLASLt's picks for the llama decode shapes run at 22-43% of the HBM
-// roofline (tools/gemm_bench.py, r1), because N/MT tiles alone can't fill
-// 256 CUs at M=64 without split-K.  This kernel is shaped for the real
-// bound — streaming W once at full bandwidth:
-//   - grid.x = N/16: each workgroup owns 16 rows of W (one MFMA n-tile).
-//   - the 4 waves split K statically (K/4 each, contiguous), so every CU
-//     holds 4 independent accumulation streams; the partial C tiles meet in
-//     LDS at the end (no global split-K scratch, no atomics).
-//   - per k-step a wave does MT=ceil(M/16) v_mfma_f32_16x16x32_bf16 ops:
-//     A = W rows (streamed, 16B/lane), B = x^T fragments (x is <=32 KB,
-//     L2-resident after the first workgroup touches it).
-//   - XCD-aware: consecutive blockIdx.x land on different XCDs (round-robin
-//     dispatch), so the N dimension spreads its L2 footprint evenly.
+// roofline (tools/gemm_bench.py, r1).  v1 of this kernel read MFMA fragments
+// straight from HBM; per-lane 16B pieces of 16 rows made every access a
+// scattered line request and each workgroup re-read all of x (the W stream
+// evicts x from L2), so useful bandwidth capped at ~1.8 TB/s.  v2 is a real
+// tiled GEMM shaped for the streaming bound:
+//   - grid (N/128, KS): each workgroup owns 128 W rows and 1/KS of K.
+//     KS (pure shape function, hipGraph-safe) tops the grid up to >=512
+//     workgroups; KS>1 writes f32 partials merged by a tiny second kernel.
+//   - per 64-k chunk, W[128, 64] (16 KB) and x[64, 64] (8 KB) tiles are
+//     staged HBM->LDS with 128B-contiguous-per-row coalescing, XOR-swizzled
+//     ((row&7)<<4, guide G4) so the 16-rows-per-lane fragment reads are
+//     bank-conflict-free; double-buffered so chunk c+1 streams while c
+//     computes.  x is read once per WG (amortized over 128 N rows).
+//   - each wave owns NT=2 16-row n-tiles; per 32-k slice it does
+//     NT*MT v_mfma_f32_16x16x32_bf16 with both operands from LDS.
 //
-// x rows beyond M are never read (fragment loads are masked); output rows
-// beyond M are not written.
+// x rows past M are staged as zeros; output rows past M are never written.
 
 #include "common.h"
 
@@ -23,203 +25,360 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 
 #define GS_NWAVES 4
+#define GS_NT 2                      // n-tiles (16 rows) per wave
+#define GS_ROWS (GS_NWAVES * GS_NT * 16)  // 128 W rows per workgroup
+#define GS_KB 64                     // k elems per staged chunk
 
-template <int MT>  // number of 16-row m tiles (M <= MT*16)
+// byte offset of (row, byte_in_row) in a [rows][GS_KB] bf16 LDS tile with the
+// (row&7)<<4 XOR swizzle; row stride = GS_KB*2 = 128 B
+DEVINLINE int swz(int row, int byte_in_row) {
+  return row * (GS_KB * 2) + (byte_in_row ^ ((row & 7) << 4));
+}
+
+template <int MT>  // m tiles of 16 rows (M <= MT*16 <= 64)
 __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
-    __hip_bfloat16* __restrict__ y,        // [M, N]
+    __hip_bfloat16* __restrict__ y,  // [M, N]   (KS == 1)
+    float* __restrict__ part,        // [KS, M, N] f32 (KS > 1; else null)
     const __hip_bfloat16* __restrict__ x,  // [M, K]
     const __hip_bfloat16* __restrict__ w,  // [N, K]
     const int M,
     const int N,
     const int K) {
-  const int n0 = blockIdx.x * 16;
+  const int n_blk = blockIdx.x;      // which 128-row block of W
+  const int ks = blockIdx.y;
+  const int KS = gridDim.y;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int row = lane & 15;        // W row in tile / n col of C
-  const int kq = lane >> 4;         // 0..3: which 8-elem k slice
+  const int row16 = lane & 15;
+  const int kq = lane >> 4;
 
-  // this wave's K range (contiguous quarter)
-  const int k_per_wave = K / GS_NWAVES;
-  const int k_lo = wave * k_per_wave;
+  const int k_per_wg = K / KS;
+  const int k_lo = ks * k_per_wg;
+  const int nchunks = k_per_wg / GS_KB;
+  const long n_base = (long)n_blk * GS_ROWS;
 
-  const __hip_bfloat16* w_row = w + (long)(n0 + row) * K + k_lo + kq * 8;
-  f32x4_t acc[MT];
+  __shared__ __hip_bfloat16 w_lds[2][GS_ROWS * GS_KB];
+  __shared__ __hip_bfloat16 x_lds[2][64 * GS_KB];
+
+  // ---- staging: 256 threads, 8 per row (8 x 16 B = 128 B contiguous) -----
+  const int st_row = tid / 8;        // 0..31 per pass
+  const int st_byte = (tid & 7) * 16;
+  const __hip_bfloat16* w_base = w + (n_base + st_row) * (long)K + k_lo + st_byte / 2;
+  const __hip_bfloat16* x_base = x + (long)st_row * K + k_lo + st_byte / 2;
+
+  bf16x8_t st_w[GS_ROWS / 32];       // 4 passes of W
+  bf16x8_t st_x[2];                  // 2 passes of x (rows 0..63)
+  auto stage_load = [&](int chunk) {
+    const int koff = chunk * GS_KB;
 #pragma unroll
-  for (int mt = 0; mt < MT; ++mt) acc[mt] = f32x4_t{};
-
-  // x fragment rows: for m tile mt, lane reads x[mt*16 + row][k + kq*8 ..+8]
-  // (masked to zero when the m row is past M)
-  const bool xrow_ok[4] = {
-      0 * 16 + (lane & 15) < M, 1 * 16 + (lane & 15) < M,
-      2 * 16 + (lane & 15) < M, 3 * 16 + (lane & 15) < M};
-
-  // K unrolled by U with all loads issued ahead of the mfmas: one wave per
-  // SIMD must keep ~U*(1+MT) loads in flight to cover HBM latency (the r1
-  // non-unrolled version ran at ~1/3 of hipBLASLt for the N=4096 shapes).
-  constexpr int U = 4;
-  static_assert(true, "");
-  for (int k = 0; k < k_per_wave; k += 32 * U) {
-    bf16x8_t a[U];
+    for (int p = 0; p < GS_ROWS / 32; ++p)
+      st_w[p] = *reinterpret_cast<const bf16x8_t*>(w_base + (long)(p * 32) * K + koff);
 #pragma unroll
-    for (int u = 0; u < U; ++u)
-      a[u] = *reinterpret_cast<const bf16x8_t*>(w_row + k + u * 32);
-    bf16x8_t b[U][MT];
+    for (int p = 0; p < 2; ++p) {
+      const int r = p * 32 + st_row;
+      st_x[p] = bf16x8_t{};
+      if (r < M)
+        st_x[p] = *reinterpret_cast<const bf16x8_t*>(x_base + (long)(p * 32) * K + koff);
+    }
+  };
+  auto stage_write = [&](int buf) {
 #pragma unroll
-    for (int u = 0; u < U; ++u) {
+    for (int p = 0; p < GS_ROWS / 32; ++p)
+      *reinterpret_cast<bf16x8_t*>(
+          reinterpret_cast<char*>(w_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_w[p];
+#pragma unroll
+    for (int p = 0; p < 2; ++p)
+      *reinterpret_cast<bf16x8_t*>(
+          reinterpret_cast<char*>(x_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_x[p];
+  };
+
+  f32x4_t acc[GS_NT][MT];
+#pragma unroll
+  for (int nt = 0; nt < GS_NT; ++nt)
+#pragma unroll
+    for (int mt = 0; mt < MT; ++mt) acc[nt][mt] = f32x4_t{};
+
+  stage_load(0);
+  stage_write(0);
+
+  for (int c = 0; c < nchunks; ++c) {
+    const int buf = c & 1;
+    __syncthreads();  // chunk c staged for everyone
+    if (c + 1 < nchunks) stage_load(c + 1);
+
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      const int byte = kc * 64 + kq * 16;
+      bf16x8_t a[GS_NT], b[MT];
+#pragma unroll
+      for (int nt = 0; nt < GS_NT; ++nt) {
+        const int r = (wave * GS_NT + nt) * 16 + row16;
+        a[nt] = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<const char*>(w_lds[buf]) + swz(r, byte));
+      }
 #pragma unroll
       for (int mt = 0; mt < MT; ++mt) {
-        b[u][mt] = bf16x8_t{};
-        if (xrow_ok[mt]) {
-          b[u][mt] = *reinterpret_cast<const bf16x8_t*>(
-              x + (long)(mt * 16 + row) * K + k_lo + k + u * 32 + kq * 8);
-        }
+        b[mt] = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<const char*>(x_lds[buf]) + swz(mt * 16 + row16, byte));
+      }
+#pragma unroll
+      for (int nt = 0; nt < GS_NT; ++nt)
+#pragma unroll
+        for (int mt = 0; mt < MT; ++mt)
+          acc[nt][mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[nt], b[mt], acc[nt][mt], 0, 0, 0);
+    }
+
+    if (c + 1 < nchunks) stage_write(1 - buf);
+  }
+
+  // ---- epilogue: C frag D[n_local = kq*4+reg][m = row16] per (nt, mt) ----
+#pragma unroll
+  for (int nt = 0; nt < GS_NT; ++nt) {
+    const long n0 = n_base + (wave * GS_NT + nt) * 16 + kq * 4;
+#pragma unroll
+    for (int mt = 0; mt < MT; ++mt) {
+      const int m = mt * 16 + row16;
+      if (m >= M) continue;
+      if (part == nullptr) {
+        // pack 4 f32 -> 4 bf16 (8 B) at y[m][n0..n0+4)
+        union { unsigned u[2]; } o;
+        asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2"
+                     : "=v"(o.u[0]) : "v"(acc[nt][mt][0]), "v"(acc[nt][mt][1]));
+        asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2"
+                     : "=v"(o.u[1]) : "v"(acc[nt][mt][2]), "v"(acc[nt][mt][3]));
+        *reinterpret_cast<unsigned*>(y + (long)m * N + n0) = o.u[0];
+        *reinterpret_cast<unsigned*>(y + (long)m * N + n0 + 2) = o.u[1];
+      } else {
+        float* p = part + ((long)ks * M + m) * N + n0;
+        *reinterpret_cast<f32x4_t*>(p) = acc[nt][mt];
       }
     }
-#pragma unroll
-    for (int u = 0; u < U; ++u) {
-#pragma unroll
-      for (int mt = 0; mt < MT; ++mt)
-        acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b[u][mt], acc[mt], 0, 0, 0);
-    }
-  }
-
-  // ---- cross-wave reduction in LDS ---------------------------------------
-  // C frag (16x16x32): lane holds D[n_local = kq*4 + reg][m = lane&15] for
-  // each m tile; store as lds_c[wave][m][n_local].
-  __shared__ float lds_c[GS_NWAVES][MT * 16][16];
-#pragma unroll
-  for (int mt = 0; mt < MT; ++mt) {
-#pragma unroll
-    for (int r = 0; r < 4; ++r)
-      lds_c[wave][mt * 16 + row][kq * 4 + r] = acc[mt][r];
-  }
-  __syncthreads();
-
-  // 256 threads cover the MT*16 x 16 outputs
-  for (int idx = tid; idx < MT * 16 * 16; idx += 256) {
-    const int m = idx >> 4;
-    const int n = idx & 15;
-    if (m >= M) continue;
-    float s = lds_c[0][m][n] + lds_c[1][m][n] + lds_c[2][m][n] + lds_c[3][m][n];
-    y[(long)m * N + n0 + n] = __float2bfloat16(s);
   }
 }
 
-void launch_gemm_skinny(__hip_bfloat16* y, const __hip_bfloat16* x,
-                        const __hip_bfloat16* w, int M, int N, int K,
+// y[m][n] = sum_ks part[ks][m][n], bf16 out
+__global__ void gemm_skinny_merge_kernel(
+    __hip_bfloat16* __restrict__ y,
+    const float* __restrict__ part,
+    const int M,
+    const int N,
+    const int KS) {
+  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long)M * N) return;
+  float s = 0.f;
+  for (int ks = 0; ks < KS; ++ks) s += part[(long)ks * M * N + idx];
+  y[idx] = __float2bfloat16(s);
+}
+
+// KS: pure function of shapes (hipGraph-stable): top the grid up to >=512
+// workgroups while keeping each WG's K range a multiple of 64 and >=256.
+int gemm_skinny_num_ksplit(int N, int K) {
+  const int cols = N / GS_ROWS;
+  int ks = 1;
+  while (ks < 16 && cols * ks * 2 <= 512 && (K / (ks * 2)) % GS_KB == 0 &&
+         K / (ks * 2) >= 256)
+    ks *= 2;
+  return ks;
+}
+
+void launch_gemm_skinny(__hip_bfloat16* y, float* part, const __hip_bfloat16* x,
+                        const __hip_bfloat16* w, int M, int N, int K, int KS,
                         hipStream_t stream) {
-  dim3 grid(N / 16);
+  dim3 grid(N / GS_ROWS, KS);
   dim3 block(256);
-  if (M <= 16)
-    hipLaunchKernelGGL(gemm_skinny_kernel<1>, grid, block, 0, stream, y, x, w, M, N, K);
-  else if (M <= 32)
-    hipLaunchKernelGGL(gemm_skinny_kernel<2>, grid, block, 0, stream, y, x, w, M, N, K);
-  else
-    hipLaunchKernelGGL(gemm_skinny_kernel<4>, grid, block, 0, stream, y, x, w, M, N, K);
+  float* p = KS > 1 ? part : nullptr;
+#define GS_CASE(MT)                                                          \
+  hipLaunchKernelGGL(gemm_skinny_kernel<MT>, grid, block, 0, stream, y, p,   \
+                     x, w, M, N, K)
+  if (M <= 16) GS_CASE(1);
+  else if (M <= 32) GS_CASE(2);
+  else GS_CASE(4);
+#undef GS_CASE
+  if (KS > 1) {
+    const long total = (long)M * N;
+    hipLaunchKernelGGL(gemm_skinny_merge_kernel,
+                       dim3((total + 255) / 256), dim3(256), 0, stream, y,
+                       part, M, N, KS);
+  }
 }
 
 // ---------------------------------------------------------------------------
-// Gated variant for the SwiGLU MLP up-projection: with w13 = [gate; up]
-// ([2I, K], MergedColumnParallelLinear layout), computes
+// Gated variant for the SwiGLU MLP: with w13 = [gate; up] ([2I, K]),
 //   y[m, i] = silu(x·Wg^T)[m, i] * (x·Wu^T)[m, i]        (y: [M, I])
-// in one pass: each workgroup streams the gate tile AND the matching up tile
-// (sharing the x B-fragments), applies the activation in the epilogue, and
-// never materialises the [M, 2I] intermediate.
+// Same tiling; each workgroup streams the gate tile AND the matching up tile
+// (sharing the x tile), so the [M, 2I] intermediate never exists.
 // ---------------------------------------------------------------------------
 
 template <int MT>
 __global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
-    __hip_bfloat16* __restrict__ y,          // [M, I]
+    __hip_bfloat16* __restrict__ y,   // [M, I]  (KS == 1)
+    float* __restrict__ part,         // [KS, 2, M, I] f32 (KS > 1)
     const __hip_bfloat16* __restrict__ x,    // [M, K]
     const __hip_bfloat16* __restrict__ w13,  // [2I, K]
     const int M,
     const int I,
     const int K) {
-  const int n0 = blockIdx.x * 16;
+  const int n_blk = blockIdx.x;
+  const int ks = blockIdx.y;
+  const int KS = gridDim.y;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int row = lane & 15;
+  const int row16 = lane & 15;
   const int kq = lane >> 4;
 
-  const int k_per_wave = K / GS_NWAVES;
-  const int k_lo = wave * k_per_wave;
+  const int k_per_wg = K / KS;
+  const int k_lo = ks * k_per_wg;
+  const int nchunks = k_per_wg / GS_KB;
+  const long n_base = (long)n_blk * GS_ROWS;
 
-  const __hip_bfloat16* wg_row = w13 + (long)(n0 + row) * K + k_lo + kq * 8;
-  const __hip_bfloat16* wu_row = w13 + (long)(I + n0 + row) * K + k_lo + kq * 8;
-  f32x4_t acc_g[MT], acc_u[MT];
-#pragma unroll
-  for (int mt = 0; mt < MT; ++mt) {
-    acc_g[mt] = f32x4_t{};
-    acc_u[mt] = f32x4_t{};
-  }
-  const bool xrow_ok[4] = {
-      0 * 16 + (lane & 15) < M, 1 * 16 + (lane & 15) < M,
-      2 * 16 + (lane & 15) < M, 3 * 16 + (lane & 15) < M};
+  __shared__ __hip_bfloat16 wg_lds[2][GS_ROWS * GS_KB];
+  __shared__ __hip_bfloat16 wu_lds[2][GS_ROWS * GS_KB];
+  __shared__ __hip_bfloat16 x_lds[2][64 * GS_KB];
 
-  constexpr int U = 4;
-  for (int k = 0; k < k_per_wave; k += 32 * U) {
-    bf16x8_t ag[U], au[U];
+  const int st_row = tid / 8;
+  const int st_byte = (tid & 7) * 16;
+  const __hip_bfloat16* wg_base = w13 + (n_base + st_row) * (long)K + k_lo + st_byte / 2;
+  const __hip_bfloat16* wu_base =
+      w13 + ((long)I + n_base + st_row) * K + k_lo + st_byte / 2;
+  const __hip_bfloat16* x_base = x + (long)st_row * K + k_lo + st_byte / 2;
+
+  bf16x8_t st_g[GS_ROWS / 32], st_u[GS_ROWS / 32], st_x[2];
+  auto stage_load = [&](int chunk) {
+    const int koff = chunk * GS_KB;
 #pragma unroll
-    for (int u = 0; u < U; ++u) {
-      ag[u] = *reinterpret_cast<const bf16x8_t*>(wg_row + k + u * 32);
-      au[u] = *reinterpret_cast<const bf16x8_t*>(wu_row + k + u * 32);
+    for (int p = 0; p < GS_ROWS / 32; ++p) {
+      st_g[p] = *reinterpret_cast<const bf16x8_t*>(wg_base + (long)(p * 32) * K + koff);
+      st_u[p] = *reinterpret_cast<const bf16x8_t*>(wu_base + (long)(p * 32) * K + koff);
     }
-    bf16x8_t b[U][MT];
 #pragma unroll
-    for (int u = 0; u < U; ++u) {
+    for (int p = 0; p < 2; ++p) {
+      const int r = p * 32 + st_row;
+      st_x[p] = bf16x8_t{};
+      if (r < M)
+        st_x[p] = *reinterpret_cast<const bf16x8_t*>(x_base + (long)(p * 32) * K + koff);
+    }
+  };
+  auto stage_write = [&](int buf) {
 #pragma unroll
-      for (int mt = 0; mt < MT; ++mt) {
-        b[u][mt] = bf16x8_t{};
-        if (xrow_ok[mt]) {
-          b[u][mt] = *reinterpret_cast<const bf16x8_t*>(
-              x + (long)(mt * 16 + row) * K + k_lo + k + u * 32 + kq * 8);
+    for (int p = 0; p < GS_ROWS / 32; ++p) {
+      *reinterpret_cast<bf16x8_t*>(
+          reinterpret_cast<char*>(wg_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_g[p];
+      *reinterpret_cast<bf16x8_t*>(
+          reinterpret_cast<char*>(wu_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_u[p];
+    }
+#pragma unroll
+    for (int p = 0; p < 2; ++p)
+      *reinterpret_cast<bf16x8_t*>(
+          reinterpret_cast<char*>(x_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_x[p];
+  };
+
+  f32x4_t acc_g[GS_NT][MT], acc_u[GS_NT][MT];
+#pragma unroll
+  for (int nt = 0; nt < GS_NT; ++nt)
+#pragma unroll
+    for (int mt = 0; mt < MT; ++mt) {
+      acc_g[nt][mt] = f32x4_t{};
+      acc_u[nt][mt] = f32x4_t{};
+    }
+
+  stage_load(0);
+  stage_write(0);
+
+  for (int c = 0; c < nchunks; ++c) {
+    const int buf = c & 1;
+    __syncthreads();
+    if (c + 1 < nchunks) stage_load(c + 1);
+
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      const int byte = kc * 64 + kq * 16;
+      bf16x8_t ag[GS_NT], au[GS_NT], b[MT];
+#pragma unroll
+      for (int nt = 0; nt < GS_NT; ++nt) {
+        const int r = (wave * GS_NT + nt) * 16 + row16;
+        ag[nt] = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<const char*>(wg_lds[buf]) + swz(r, byte));
+        au[nt] = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<const char*>(wu_lds[buf]) + swz(r, byte));
+      }
+#pragma unroll
+      for (int mt = 0; mt < MT; ++mt)
+        b[mt] = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<const char*>(x_lds[buf]) + swz(mt * 16 + row16, byte));
+#pragma unroll
+      for (int nt = 0; nt < GS_NT; ++nt)
+#pragma unroll
+        for (int mt = 0; mt < MT; ++mt) {
+          acc_g[nt][mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ag[nt], b[mt], acc_g[nt][mt], 0, 0, 0);
+          acc_u[nt][mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(au[nt], b[mt], acc_u[nt][mt], 0, 0, 0);
         }
-      }
     }
-#pragma unroll
-    for (int u = 0; u < U; ++u) {
-#pragma unroll
-      for (int mt = 0; mt < MT; ++mt) {
-        acc_g[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ag[u], b[u][mt], acc_g[mt], 0, 0, 0);
-        acc_u[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(au[u], b[u][mt], acc_u[mt], 0, 0, 0);
-      }
-    }
+
+    if (c + 1 < nchunks) stage_write(1 - buf);
   }
 
-  __shared__ float lds_g[GS_NWAVES][MT * 16][16];
-  __shared__ float lds_u[GS_NWAVES][MT * 16][16];
 #pragma unroll
-  for (int mt = 0; mt < MT; ++mt) {
+  for (int nt = 0; nt < GS_NT; ++nt) {
+    const long n0 = n_base + (wave * GS_NT + nt) * 16 + kq * 4;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      lds_g[wave][mt * 16 + row][kq * 4 + r] = acc_g[mt][r];
-      lds_u[wave][mt * 16 + row][kq * 4 + r] = acc_u[mt][r];
+    for (int mt = 0; mt < MT; ++mt) {
+      const int m = mt * 16 + row16;
+      if (m >= M) continue;
+      if (part == nullptr) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float g = acc_g[nt][mt][r];
+          const float act = g / (1.f + __expf(-g));
+          y[(long)m * I + n0 + r] = __float2bfloat16(act * acc_u[nt][mt][r]);
+        }
+      } else {
+        float* pg = part + (((long)ks * 2 + 0) * M + m) * I + n0;
+        float* pu = part + (((long)ks * 2 + 1) * M + m) * I + n0;
+        *reinterpret_cast<f32x4_t*>(pg) = acc_g[nt][mt];
+        *reinterpret_cast<f32x4_t*>(pu) = acc_u[nt][mt];
+      }
     }
-  }
-  __syncthreads();
-
-  for (int idx = tid; idx < MT * 16 * 16; idx += 256) {
-    const int m = idx >> 4;
-    const int n = idx & 15;
-    if (m >= M) continue;
-    const float g = lds_g[0][m][n] + lds_g[1][m][n] + lds_g[2][m][n] + lds_g[3][m][n];
-    const float u = lds_u[0][m][n] + lds_u[1][m][n] + lds_u[2][m][n] + lds_u[3][m][n];
-    const float act = g / (1.f + __expf(-g));
-    y[(long)m * I + n0 + n] = __float2bfloat16(act * u);
   }
 }
 
-void launch_gemm_skinny_gated(__hip_bfloat16* y, const __hip_bfloat16* x,
+// y[m][i] = silu(sum_ks g) * (sum_ks u)
+__global__ void gemm_gated_merge_kernel(
+    __hip_bfloat16* __restrict__ y,
+    const float* __restrict__ part,  // [KS, 2, M, I]
+    const int M,
+    const int I,
+    const int KS) {
+  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long)M * I) return;
+  float g = 0.f, u = 0.f;
+  for (int ks = 0; ks < KS; ++ks) {
+    g += part[((long)ks * 2 + 0) * M * I + idx];
+    u += part[((long)ks * 2 + 1) * M * I + idx];
+  }
+  const float act = g / (1.f + __expf(-g));
+  y[idx] = __float2bfloat16(act * u);
+}
+
+void launch_gemm_skinny_gated(__hip_bfloat16* y, float* part,
+                              const __hip_bfloat16* x,
                               const __hip_bfloat16* w13, int M, int I, int K,
-                              hipStream_t stream) {
-  dim3 grid(I / 16);
+                              int KS, hipStream_t stream) {
+  dim3 grid(I / GS_ROWS, KS);
   dim3 block(256);
-  if (M <= 16)
-    hipLaunchKernelGGL(gemm_skinny_gated_kernel<1>, grid, block, 0, stream, y, x, w13, M, I, K);
-  else if (M <= 32)
-    hipLaunchKernelGGL(gemm_skinny_gated_kernel<2>, grid, block, 0, stream, y, x, w13, M, I, K);
-  else
-    hipLaunchKernelGGL(gemm_skinny_gated_kernel<4>, grid, block, 0, stream, y, x, w13, M, I, K);
+  float* p = KS > 1 ? part : nullptr;
+#define GG_CASE(MT)                                                           \
+  hipLaunchKernelGGL(gemm_skinny_gated_kernel<MT>, grid, block, 0, stream, y, \
+                     p, x, w13, M, I, K)
+  if (M <= 16) GG_CASE(1);
+  else if (M <= 32) GG_CASE(2);
+  else GG_CASE(4);
+#undef GG_CASE
+  if (KS > 1) {
+    const long total = (long)M * I;
+    hipLaunchKernelGGL(gemm_gated_merge_kernel,
+                       dim3((total + 255) / 256), dim3(256), 0, stream, y,
+                       part, M, I, KS);
+  }
 }

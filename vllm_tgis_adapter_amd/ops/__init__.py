@@ -181,8 +181,8 @@ def linear(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None = No
         and x.dtype == torch.bfloat16
         and weight.dtype == torch.bfloat16
         and 1 <= x.shape[0] <= 64
-        and weight.shape[0] % 16 == 0
-        and x.shape[1] % 512 == 0
+        and weight.shape[0] % 128 == 0
+        and x.shape[1] % 64 == 0
         and x.is_contiguous()
         and weight.is_contiguous()
     ):
@@ -206,8 +206,8 @@ def gated_mlp_up(x: torch.Tensor, w13: torch.Tensor) -> torch.Tensor | None:
         and x.shape[0] <= 64
         and x.dtype == torch.bfloat16
         and w13.dtype == torch.bfloat16
-        and w13.shape[0] % 32 == 0
-        and x.shape[1] % 512 == 0
+        and w13.shape[0] % 256 == 0
+        and x.shape[1] % 64 == 0
         and x.is_contiguous()
         and w13.is_contiguous()
         and _native(x)
