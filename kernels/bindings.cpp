@@ -38,7 +38,7 @@ void launch_paged_prefill_mfma(__hip_bfloat16*, const __hip_bfloat16*,
                                const int*, const int*, const int*, float, int,
                                int, int, int, int, int, int, hipStream_t);
 int decode_mfma_num_partitions(int nseq, int kvh, int max_context);
-int gemm_skinny_num_ksplit(int N, int K);
+int gemm_skinny_num_ksplit(int N, int K, int M);
 void launch_gemm_skinny(__hip_bfloat16*, float*, const __hip_bfloat16*,
                         const __hip_bfloat16*, int, int, int, int,
                         hipStream_t);
@@ -248,9 +248,9 @@ void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
   TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16 &&
               w.scalar_type() == at::ScalarType::BFloat16);
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
-  TORCH_CHECK(M >= 1 && M <= 64 && N % 128 == 0 && K % 64 == 0);
+  TORCH_CHECK(M >= 1 && N % 128 == 0 && K % 64 == 0);
   TORCH_CHECK(w.size(1) == K && y.size(0) == M && y.size(1) == N);
-  const int KS = gemm_skinny_num_ksplit(N, K);
+  const int KS = gemm_skinny_num_ksplit(N, K, M);
   torch::Tensor part;
   float* pp = nullptr;
   if (KS > 1) {
@@ -270,10 +270,10 @@ void gemm_skinny_gated(torch::Tensor y, torch::Tensor x, torch::Tensor w13) {
   TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16 &&
               w13.scalar_type() == at::ScalarType::BFloat16);
   TORCH_CHECK(x.is_contiguous() && w13.is_contiguous() && y.is_contiguous());
-  TORCH_CHECK(M >= 1 && M <= 64 && I % 128 == 0 && K % 64 == 0);
+  TORCH_CHECK(M >= 1 && I % 128 == 0 && K % 64 == 0);
   TORCH_CHECK(w13.size(0) % 2 == 0 && w13.size(1) == K);
   TORCH_CHECK(y.size(0) == M && y.size(1) == I);
-  const int KS = gemm_skinny_num_ksplit(I, K);
+  const int KS = gemm_skinny_num_ksplit(I, K, M);
   torch::Tensor part;
   float* pp = nullptr;
   if (KS > 1) {

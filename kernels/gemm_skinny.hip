@@ -35,7 +35,7 @@ DEVINLINE int swz(int row, int byte_in_row) {
   return row * (GS_KB * 2) + (byte_in_row ^ ((row & 7) << 4));
 }
 
-template <int MT>  // m tiles of 16 rows (M <= MT*16 <= 64)
+template <int MT>  // m tiles of 16 rows per workgroup (up to 8 = 128 rows)
 __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
     __hip_bfloat16* __restrict__ y,  // [M, N]   (KS == 1)
     float* __restrict__ part,        // [KS, M, N] f32 (KS > 1; else null)
@@ -46,6 +46,7 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
     const int K) {
   const int n_blk = blockIdx.x;      // which 128-row block of W
   const int ks = blockIdx.y;
+  const int m0 = blockIdx.z * (MT * 16);  // this workgroup's m rows
   const int KS = gridDim.y;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -58,27 +59,30 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
   const int nchunks = k_per_wg / GS_KB;
   const long n_base = (long)n_blk * GS_ROWS;
 
+  constexpr int XROWS = MT * 16;
+  constexpr int XPASS = (XROWS + 31) / 32;
   __shared__ __hip_bfloat16 w_lds[2][GS_ROWS * GS_KB];
-  __shared__ __hip_bfloat16 x_lds[2][64 * GS_KB];
+  __shared__ __hip_bfloat16 x_lds[2][XROWS * GS_KB];
 
   // ---- staging: 256 threads, 8 per row (8 x 16 B = 128 B contiguous) -----
   const int st_row = tid / 8;        // 0..31 per pass
   const int st_byte = (tid & 7) * 16;
   const __hip_bfloat16* w_base = w + (n_base + st_row) * (long)K + k_lo + st_byte / 2;
-  const __hip_bfloat16* x_base = x + (long)st_row * K + k_lo + st_byte / 2;
+  const __hip_bfloat16* x_base =
+      x + (long)(m0 + st_row) * K + k_lo + st_byte / 2;
 
   bf16x8_t st_w[GS_ROWS / 32];       // 4 passes of W
-  bf16x8_t st_x[2];                  // 2 passes of x (rows 0..63)
+  bf16x8_t st_x[XPASS];              // x rows m0 .. m0+XROWS
   auto stage_load = [&](int chunk) {
     const int koff = chunk * GS_KB;
 #pragma unroll
     for (int p = 0; p < GS_ROWS / 32; ++p)
       st_w[p] = *reinterpret_cast<const bf16x8_t*>(w_base + (long)(p * 32) * K + koff);
 #pragma unroll
-    for (int p = 0; p < 2; ++p) {
+    for (int p = 0; p < XPASS; ++p) {
       const int r = p * 32 + st_row;
       st_x[p] = bf16x8_t{};
-      if (r < M)
+      if (r < XROWS && m0 + r < M)
         st_x[p] = *reinterpret_cast<const bf16x8_t*>(x_base + (long)(p * 32) * K + koff);
     }
   };
@@ -88,9 +92,12 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
       *reinterpret_cast<bf16x8_t*>(
           reinterpret_cast<char*>(w_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_w[p];
 #pragma unroll
-    for (int p = 0; p < 2; ++p)
-      *reinterpret_cast<bf16x8_t*>(
-          reinterpret_cast<char*>(x_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_x[p];
+    for (int p = 0; p < XPASS; ++p) {
+      const int r = p * 32 + st_row;
+      if (r < XROWS)
+        *reinterpret_cast<bf16x8_t*>(
+            reinterpret_cast<char*>(x_lds[buf]) + swz(r, st_byte)) = st_x[p];
+    }
   };
 
   f32x4_t acc[GS_NT][MT];
@@ -138,7 +145,7 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
     const long n0 = n_base + (wave * GS_NT + nt) * 16 + kq * 4;
 #pragma unroll
     for (int mt = 0; mt < MT; ++mt) {
-      const int m = mt * 16 + row16;
+      const int m = m0 + mt * 16 + row16;
       if (m >= M) continue;
       if (part == nullptr) {
         // pack 4 f32 -> 4 bf16 (8 B) at y[m][n0..n0+4)
@@ -173,10 +180,11 @@ __global__ void gemm_skinny_merge_kernel(
 
 // KS: pure function of shapes (hipGraph-stable): top the grid up to >=512
 // workgroups while keeping each WG's K range a multiple of 64 and >=256.
-int gemm_skinny_num_ksplit(int N, int K) {
+int gemm_skinny_num_ksplit(int N, int K, int M) {
   const int cols = N / GS_ROWS;
+  const int zb = M > 64 ? (M + 127) / 128 : 1;
   int ks = 1;
-  while (ks < 16 && cols * ks * 2 <= 512 && (K / (ks * 2)) % GS_KB == 0 &&
+  while (ks < 16 && cols * zb * ks * 2 <= 512 && (K / (ks * 2)) % GS_KB == 0 &&
          K / (ks * 2) >= 256)
     ks *= 2;
   return ks;
@@ -185,15 +193,15 @@ int gemm_skinny_num_ksplit(int N, int K) {
 void launch_gemm_skinny(__hip_bfloat16* y, float* part, const __hip_bfloat16* x,
                         const __hip_bfloat16* w, int M, int N, int K, int KS,
                         hipStream_t stream) {
-  dim3 grid(N / GS_ROWS, KS);
   dim3 block(256);
   float* p = KS > 1 ? part : nullptr;
-#define GS_CASE(MT)                                                          \
-  hipLaunchKernelGGL(gemm_skinny_kernel<MT>, grid, block, 0, stream, y, p,   \
-                     x, w, M, N, K)
-  if (M <= 16) GS_CASE(1);
-  else if (M <= 32) GS_CASE(2);
-  else GS_CASE(4);
+#define GS_CASE(MT, ZB)                                                      \
+  hipLaunchKernelGGL(gemm_skinny_kernel<MT>, dim3(N / GS_ROWS, KS, ZB),      \
+                     block, 0, stream, y, p, x, w, M, N, K)
+  if (M <= 16) GS_CASE(1, 1);
+  else if (M <= 32) GS_CASE(2, 1);
+  else if (M <= 64) GS_CASE(4, 1);
+  else GS_CASE(8, (M + 127) / 128);
 #undef GS_CASE
   if (KS > 1) {
     const long total = (long)M * N;
@@ -221,6 +229,7 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
     const int K) {
   const int n_blk = blockIdx.x;
   const int ks = blockIdx.y;
+  const int m0 = blockIdx.z * (MT * 16);
   const int KS = gridDim.y;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -233,18 +242,21 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
   const int nchunks = k_per_wg / GS_KB;
   const long n_base = (long)n_blk * GS_ROWS;
 
+  constexpr int XROWS = MT * 16;
+  constexpr int XPASS = (XROWS + 31) / 32;
   __shared__ __hip_bfloat16 wg_lds[2][GS_ROWS * GS_KB];
   __shared__ __hip_bfloat16 wu_lds[2][GS_ROWS * GS_KB];
-  __shared__ __hip_bfloat16 x_lds[2][64 * GS_KB];
+  __shared__ __hip_bfloat16 x_lds[2][XROWS * GS_KB];
 
   const int st_row = tid / 8;
   const int st_byte = (tid & 7) * 16;
   const __hip_bfloat16* wg_base = w13 + (n_base + st_row) * (long)K + k_lo + st_byte / 2;
   const __hip_bfloat16* wu_base =
       w13 + ((long)I + n_base + st_row) * K + k_lo + st_byte / 2;
-  const __hip_bfloat16* x_base = x + (long)st_row * K + k_lo + st_byte / 2;
+  const __hip_bfloat16* x_base =
+      x + (long)(m0 + st_row) * K + k_lo + st_byte / 2;
 
-  bf16x8_t st_g[GS_ROWS / 32], st_u[GS_ROWS / 32], st_x[2];
+  bf16x8_t st_g[GS_ROWS / 32], st_u[GS_ROWS / 32], st_x[XPASS];
   auto stage_load = [&](int chunk) {
     const int koff = chunk * GS_KB;
 #pragma unroll
@@ -253,10 +265,10 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
       st_u[p] = *reinterpret_cast<const bf16x8_t*>(wu_base + (long)(p * 32) * K + koff);
     }
 #pragma unroll
-    for (int p = 0; p < 2; ++p) {
+    for (int p = 0; p < XPASS; ++p) {
       const int r = p * 32 + st_row;
       st_x[p] = bf16x8_t{};
-      if (r < M)
+      if (r < XROWS && m0 + r < M)
         st_x[p] = *reinterpret_cast<const bf16x8_t*>(x_base + (long)(p * 32) * K + koff);
     }
   };
@@ -269,9 +281,12 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
           reinterpret_cast<char*>(wu_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_u[p];
     }
 #pragma unroll
-    for (int p = 0; p < 2; ++p)
-      *reinterpret_cast<bf16x8_t*>(
-          reinterpret_cast<char*>(x_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_x[p];
+    for (int p = 0; p < XPASS; ++p) {
+      const int r = p * 32 + st_row;
+      if (r < XROWS)
+        *reinterpret_cast<bf16x8_t*>(
+            reinterpret_cast<char*>(x_lds[buf]) + swz(r, st_byte)) = st_x[p];
+    }
   };
 
   f32x4_t acc_g[GS_NT][MT], acc_u[GS_NT][MT];
@@ -324,7 +339,7 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
     const long n0 = n_base + (wave * GS_NT + nt) * 16 + kq * 4;
 #pragma unroll
     for (int mt = 0; mt < MT; ++mt) {
-      const int m = mt * 16 + row16;
+      const int m = m0 + mt * 16 + row16;
       if (m >= M) continue;
       if (part == nullptr) {
 #pragma unroll
@@ -365,15 +380,16 @@ void launch_gemm_skinny_gated(__hip_bfloat16* y, float* part,
                               const __hip_bfloat16* x,
                               const __hip_bfloat16* w13, int M, int I, int K,
                               int KS, hipStream_t stream) {
-  dim3 grid(I / GS_ROWS, KS);
   dim3 block(256);
   float* p = KS > 1 ? part : nullptr;
-#define GG_CASE(MT)                                                           \
-  hipLaunchKernelGGL(gemm_skinny_gated_kernel<MT>, grid, block, 0, stream, y, \
+#define GG_CASE(MT, ZB)                                                       \
+  hipLaunchKernelGGL(gemm_skinny_gated_kernel<MT>,                            \
+                     dim3(I / GS_ROWS, KS, ZB), block, 0, stream, y,          \
                      p, x, w13, M, I, K)
-  if (M <= 16) GG_CASE(1);
-  else if (M <= 32) GG_CASE(2);
-  else GG_CASE(4);
+  if (M <= 16) GG_CASE(1, 1);
+  else if (M <= 32) GG_CASE(2, 1);
+  else if (M <= 64) GG_CASE(4, 1);
+  else GG_CASE(8, (M + 127) / 128);
 #undef GG_CASE
   if (KS > 1) {
     const long total = (long)M * I;

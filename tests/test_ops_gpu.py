@@ -267,6 +267,7 @@ print('TOKENS', final.outputs[0].token_ids)
 @pytest.mark.parametrize(("m", "n", "k"), [
     (1, 4096, 4096), (8, 6144, 4096), (17, 4096, 14336),
     (33, 1024, 512), (64, 28672, 4096), (64, 128256, 4096),
+    (128, 4096, 4096), (300, 6144, 4096), (2048, 4096, 14336),
 ])
 def test_gemm_skinny(m, n, k):
     from vllm_tgis_adapter_amd import ops
@@ -282,7 +283,7 @@ def test_gemm_skinny(m, n, k):
     )
 
 
-@pytest.mark.parametrize(("m", "inter", "k"), [(1, 14336, 4096), (64, 14336, 4096), (40, 128, 512)])
+@pytest.mark.parametrize(("m", "inter", "k"), [(1, 14336, 4096), (64, 14336, 4096), (40, 128, 512), (512, 14336, 4096)])
 def test_gemm_skinny_gated(m, inter, k):
     from vllm_tgis_adapter_amd import ops
 

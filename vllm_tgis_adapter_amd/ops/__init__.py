@@ -180,7 +180,7 @@ def linear(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None = No
         and _native(x)
         and x.dtype == torch.bfloat16
         and weight.dtype == torch.bfloat16
-        and 1 <= x.shape[0] <= 64
+        and x.shape[0] >= 1
         and weight.shape[0] % 128 == 0
         and x.shape[1] % 64 == 0
         and x.is_contiguous()
@@ -203,7 +203,6 @@ def gated_mlp_up(x: torch.Tensor, w13: torch.Tensor) -> torch.Tensor | None:
     if (
         x.dim() == 2
         and x.shape[0] >= 1
-        and x.shape[0] <= 64
         and x.dtype == torch.bfloat16
         and w13.dtype == torch.bfloat16
         and w13.shape[0] % 256 == 0
