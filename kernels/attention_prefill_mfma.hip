@@ -136,16 +136,18 @@ __global__ __launch_bounds__(256, 1) void paged_prefill_mfma_kernel(
     }
   };
 
+  // T14 pipeline: write tile t from regs (loaded a full iteration ago),
+  // re-issue loads for t+1, barrier, compute t.
   stage_load(0);
-  stage_write(0);
 
   for (int kv_base = 0; kv_base < kv_limit; kv_base += KVT) {
-    const __hip_bfloat16* k_lds = k_lds2[(kv_base / KVT) & 1];
-    const __hip_bfloat16* v_lds = v_lds2[(kv_base / KVT) & 1];
-    __syncthreads();  // tile (kv_base) fully staged for all waves
-    // issue next tile's loads now; they complete behind the mfma phase
+    const int buf = (kv_base / KVT) & 1;
+    const __hip_bfloat16* k_lds = k_lds2[buf];
+    const __hip_bfloat16* v_lds = v_lds2[buf];
+    stage_write(buf);
     const int next_base = kv_base + KVT;
     if (next_base < kv_limit) stage_load(next_base);
+    __syncthreads();  // tile (kv_base) fully staged for all waves
 
     // ---- S^T = K · Q^T -------------------------------------------------
     f32x16_t acc_s{};
@@ -242,9 +244,6 @@ __global__ __launch_bounds__(256, 1) void paged_prefill_mfma_kernel(
       }
     }
 
-    // drain next tile's loads into its LDS buffer (safe: the buffer's last
-    // readers finished before this iteration's top __syncthreads)
-    if (next_base < kv_limit) stage_write((next_base / KVT) & 1);
   }
 
   // ---- epilogue: normalize by l (per output q row) and store -----------

@@ -106,13 +106,16 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
 #pragma unroll
     for (int mt = 0; mt < MT; ++mt) acc[nt][mt] = f32x4_t{};
 
+  // T14 pipeline (guide §5): write chunk c from regs (its loads had a full
+  // iteration to land), immediately re-issue loads for c+1 into the same
+  // regs, barrier, compute c.  No load-latency stall inside the loop.
   stage_load(0);
-  stage_write(0);
 
   for (int c = 0; c < nchunks; ++c) {
     const int buf = c & 1;
-    __syncthreads();  // chunk c staged for everyone
+    stage_write(buf);
     if (c + 1 < nchunks) stage_load(c + 1);
+    __syncthreads();  // chunk c staged for everyone
 
 #pragma unroll
     for (int kc = 0; kc < 2; ++kc) {
@@ -135,8 +138,6 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
         for (int mt = 0; mt < MT; ++mt)
           acc[nt][mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[nt], b[mt], acc[nt][mt], 0, 0, 0);
     }
-
-    if (c + 1 < nchunks) stage_write(1 - buf);
   }
 
   // ---- epilogue: C frag D[n_local = kq*4+reg][m = row16] per (nt, mt) ----
@@ -299,12 +300,12 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
     }
 
   stage_load(0);
-  stage_write(0);
 
   for (int c = 0; c < nchunks; ++c) {
     const int buf = c & 1;
-    __syncthreads();
+    stage_write(buf);
     if (c + 1 < nchunks) stage_load(c + 1);
+    __syncthreads();
 
 #pragma unroll
     for (int kc = 0; kc < 2; ++kc) {
@@ -330,8 +331,6 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
           acc_u[nt][mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(au[nt], b[mt], acc_u[nt][mt], 0, 0, 0);
         }
     }
-
-    if (c + 1 < nchunks) stage_write(1 - buf);
   }
 
 #pragma unroll

@@ -174,13 +174,14 @@ def linear(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None = No
     roofline (tools/gemm_bench.py); the hand-written MFMA kernel streams the
     weight matrix once with waves splitting K in-workgroup.
     """
+    max_m = int(os.environ.get("VTA_GEMM_MAX_M", "64"))
     if (
         bias is None
         and x.dim() == 2
         and _native(x)
         and x.dtype == torch.bfloat16
         and weight.dtype == torch.bfloat16
-        and x.shape[0] >= 1
+        and 1 <= x.shape[0] <= max_m
         and weight.shape[0] % 128 == 0
         and x.shape[1] % 64 == 0
         and x.is_contiguous()
@@ -200,9 +201,10 @@ def gated_mlp_up(x: torch.Tensor, w13: torch.Tensor) -> torch.Tensor | None:
     Returns None when the fused CDNA4 path doesn't apply (caller falls back
     to linear + silu_and_mul).
     """
+    max_m = int(os.environ.get("VTA_GEMM_MAX_M", "64"))
     if (
         x.dim() == 2
-        and x.shape[0] >= 1
+        and 1 <= x.shape[0] <= max_m
         and x.dtype == torch.bfloat16
         and w13.dtype == torch.bfloat16
         and w13.shape[0] % 256 == 0
