@@ -64,9 +64,32 @@ class Sampler:
 
     @torch.inference_mode()
     def sample(self, logits: torch.Tensor, requests: list[Request]) -> SamplerOutput:
-        """logits: [N, vocab] raw lm-head outputs for the N sampling rows."""
+        """logits: [N, vocab] raw lm-head outputs for the N sampling rows.
+
+        The all-greedy no-extras batch (the serving steady state) stays in
+        the lm-head dtype end to end — one argmax pass, no [N, vocab] f32
+        materialisation (2x 131 MB per step at batch 256 / 128k vocab).
+        """
         n, vocab = logits.shape
         assert n == len(requests)
+
+        plain_greedy = True
+        for req in requests:
+            p = req.sampling_params
+            if (
+                p.temperature != 0.0
+                or p.repetition_penalty != 1.0
+                or p.logits_processors
+                or req.guided_state is not None
+                or p.logprobs is not None
+                or (p.min_tokens and req.num_output_tokens < p.min_tokens)
+            ):
+                plain_greedy = False
+                break
+        if plain_greedy:
+            sampled_cpu = torch.argmax(logits, dim=-1).tolist()
+            return SamplerOutput(token_ids=sampled_cpu, logprobs=[None] * n)
+
         logits = logits.float()
 
         # --- per-request host-side adjustments (rare paths) -------------
