@@ -1,3 +1,4 @@
+#include <cstdlib>
 // Skinny decode GEMM (M <= 64): y[M,N] = x[M,K] · W[N,K]^T, bf16 in/out.
 //
 // hipBLASLt's picks for the llama decode shapes run at 22-43% of the HBM
@@ -182,11 +183,17 @@ __global__ void gemm_skinny_merge_kernel(
 // KS: pure function of shapes (hipGraph-stable): top the grid up to >=512
 // workgroups while keeping each WG's K range a multiple of 64 and >=256.
 int gemm_skinny_num_ksplit(int N, int K, int M) {
+  // VTA_GEMM_MIN_KDEPTH tunes the fill-vs-pipeline-depth trade
+  // (k elems each workgroup keeps; deeper = better ramp, fewer WGs).
+  static const int min_depth = [] {
+    const char* e = getenv("VTA_GEMM_MIN_KDEPTH");
+    return e ? atoi(e) : 1024;
+  }();
   const int cols = N / GS_ROWS;
   const int zb = M > 64 ? (M + 127) / 128 : 1;
   int ks = 1;
   while (ks < 16 && cols * zb * ks * 2 <= 512 && (K / (ks * 2)) % GS_KB == 0 &&
-         K / (ks * 2) >= 256)
+         K / (ks * 2) >= min_depth)
     ks *= 2;
   return ks;
 }
