@@ -24,6 +24,7 @@
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+typedef __attribute__((ext_vector_type(16))) float f32x16_t;
 
 #define GS_NWAVES 4
 #define GS_NT 2                      // n-tiles (16 rows) per wave
@@ -166,6 +167,124 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Large-M tiled GEMM (M > 64): classic 128x128 macro-tile on
+// v_mfma_f32_32x32x16_bf16.  The 2x2 wave grid gives each wave a 64x64 C
+// tile (2x2 of 32x32 MFMA tiles, 64 f32 accum regs); x and W chunk tiles
+// ([128, 64k], 16 KB each) are staged with the same XOR-swizzled T14
+// double-buffer pipeline as the skinny kernel.  Replaces hipBLASLt for the
+// prefill and large-decode-batch linears (measured ~34% MFU there, r1).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256, 2) void gemm_tile_kernel(
+    __hip_bfloat16* __restrict__ y,  // [M, N]   (KS == 1)
+    float* __restrict__ part,        // [KS, M, N] f32 (KS > 1; else null)
+    const __hip_bfloat16* __restrict__ x,  // [M, K]
+    const __hip_bfloat16* __restrict__ w,  // [N, K]
+    const int M,
+    const int N,
+    const int K) {
+  const int n_blk = blockIdx.x;
+  const int ks = blockIdx.y;
+  const int m0 = blockIdx.z * 128;
+  const int KS = gridDim.y;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int col32 = lane & 31;
+  const int half = lane >> 5;
+  const int mh = wave >> 1;   // wave's m half (64 rows)
+  const int nh = wave & 1;    // wave's n half (64 cols)
+
+  const int k_per_wg = K / KS;
+  const int k_lo = ks * k_per_wg;
+  const int nchunks = k_per_wg / GS_KB;
+  const long n_base = (long)n_blk * 128;
+
+  __shared__ __hip_bfloat16 w_lds[2][128 * GS_KB];
+  __shared__ __hip_bfloat16 x_lds[2][128 * GS_KB];
+
+  const int st_row = tid / 8;        // 0..31 per pass
+  const int st_byte = (tid & 7) * 16;
+  const __hip_bfloat16* w_base = w + (n_base + st_row) * (long)K + k_lo + st_byte / 2;
+  const __hip_bfloat16* x_base =
+      x + (long)(m0 + st_row) * K + k_lo + st_byte / 2;
+
+  bf16x8_t st_w[4], st_x[4];
+  auto stage_load = [&](int chunk) {
+    const int koff = chunk * GS_KB;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      st_w[p] = *reinterpret_cast<const bf16x8_t*>(w_base + (long)(p * 32) * K + koff);
+      st_x[p] = bf16x8_t{};
+      if (m0 + p * 32 + st_row < M)
+        st_x[p] = *reinterpret_cast<const bf16x8_t*>(x_base + (long)(p * 32) * K + koff);
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      *reinterpret_cast<bf16x8_t*>(
+          reinterpret_cast<char*>(w_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_w[p];
+      *reinterpret_cast<bf16x8_t*>(
+          reinterpret_cast<char*>(x_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_x[p];
+    }
+  };
+
+  f32x16_t acc[2][2];  // [mt][nt] 32x32 tiles
+  acc[0][0] = f32x16_t{}; acc[0][1] = f32x16_t{};
+  acc[1][0] = f32x16_t{}; acc[1][1] = f32x16_t{};
+
+  stage_load(0);
+
+  for (int c = 0; c < nchunks; ++c) {
+    const int buf = c & 1;
+    stage_write(buf);
+    if (c + 1 < nchunks) stage_load(c + 1);
+    __syncthreads();
+
+#pragma unroll
+    for (int kc = 0; kc < 4; ++kc) {  // 16-k slices of the 64-k chunk
+      const int byte = kc * 32 + half * 16;
+      bf16x8_t a[2], b[2];
+#pragma unroll
+      for (int mt = 0; mt < 2; ++mt)
+        a[mt] = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<const char*>(x_lds[buf]) +
+            swz(mh * 64 + mt * 32 + col32, byte));
+#pragma unroll
+      for (int nt = 0; nt < 2; ++nt)
+        b[nt] = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<const char*>(w_lds[buf]) +
+            swz(nh * 64 + nt * 32 + col32, byte));
+#pragma unroll
+      for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+        for (int nt = 0; nt < 2; ++nt)
+          acc[mt][nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a[mt], b[nt], acc[mt][nt], 0, 0, 0);
+    }
+  }
+
+  // epilogue: D[32m][32n]: lane holds D[(r&3)+8*(r>>2)+4*half][col32]
+#pragma unroll
+  for (int mt = 0; mt < 2; ++mt) {
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      const long n = n_base + nh * 64 + nt * 32 + col32;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int m = m0 + mh * 64 + mt * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+        if (m >= M) continue;
+        if (part == nullptr)
+          y[(long)m * N + n] = __float2bfloat16(acc[mt][nt][r]);
+        else
+          part[((long)ks * M + m) * N + n] = acc[mt][nt][r];
+      }
+    }
+  }
+}
+
 // y[m][n] = sum_ks part[ks][m][n], bf16 out
 __global__ void gemm_skinny_merge_kernel(
     __hip_bfloat16* __restrict__ y,
@@ -209,7 +328,10 @@ void launch_gemm_skinny(__hip_bfloat16* y, float* part, const __hip_bfloat16* x,
   if (M <= 16) GS_CASE(1, 1);
   else if (M <= 32) GS_CASE(2, 1);
   else if (M <= 64) GS_CASE(4, 1);
-  else GS_CASE(8, (M + 127) / 128);
+  else
+    hipLaunchKernelGGL(gemm_tile_kernel,
+                       dim3(N / 128, KS, (M + 127) / 128), block, 0, stream,
+                       y, p, x, w, M, N, K);
 #undef GS_CASE
   if (KS > 1) {
     const long total = (long)M * N;
