@@ -135,17 +135,10 @@ __global__ __launch_bounds__(256, 2) void paged_decode_mfma_kernel(
           vv[pass] = *reinterpret_cast<const bf16x8_t*>(v_cache + row);
         }
       }
-      // fragment-major V image: chunk dt*2+c, lane = col+32h, j = s%8
-      const int stv_dt = d8 >> 5;
-      const int stv_col = d8 & 31;
 #pragma unroll
       for (int pass = 0; pass < VPASS; ++pass) {
         const int s = pass * ROWS_PER_PASS + r_in_pass;
-        short* vb = reinterpret_cast<short*>(v_lds) +
-                    (((stv_dt * 2 + (s >> 4)) * 64 + stv_col + 32 * ((s >> 3) & 1)) * 8) +
-                    (s & 7);
-#pragma unroll
-        for (int i = 0; i < 8; ++i) vb[i * 8] = vv[pass][i];
+        *reinterpret_cast<bf16x8_t*>(v_lds + s * HEAD_DIM + d8) = vv[pass];
       }
     }
 
@@ -221,8 +214,11 @@ __global__ __launch_bounds__(256, 2) void paged_decode_mfma_kernel(
     for (int dt = 0; dt < DT; ++dt) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        bf16x8_t b = *reinterpret_cast<const bf16x8_t*>(
-            reinterpret_cast<const char*>(v_lds) + ((dt * 2 + c) * 64 + lane) * 16);
+        bf16x8_t b;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          b[j] = *reinterpret_cast<const short*>(
+              v_lds + (c * 16 + half * 8 + j) * HEAD_DIM + dt * 32 + col);
         acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[c], b, acc_o[dt], 0, 0, 0);
       }
     }

@@ -12,11 +12,9 @@
 // P is packed to bf16 with v_cvt_pk_bf16_f32 and redistributed with
 // v_permlane32_swap so it feeds the P·V MFMA's A operand directly (T12/T21).
 //
-// LDS: both K and V tiles are stored FRAGMENT-MAJOR — in the exact per-lane
-// order the MFMA fragment reads consume — so every A/B fragment load is one
-// conflict-free ds_read_b128 at ((chunk)*64 + lane)*16B.  K keeps contiguous
-// 16B staging writes; V staging scatters 8x2B per chunk (write-side cost
-// once per tile instead of 64 scalar reads per tile per wave).
+// LDS: the K tile is XOR-swizzled (byte ^= (row&7)<<4, guide G4) because its
+// A-fragment reads walk 32 distinct rows at one column slice; the V tile
+// stays linear because its B-fragment gather reads one row across lanes.
 
 #include "common.h"
 #include <float.h>
@@ -127,25 +125,14 @@ __global__ __launch_bounds__(256, 1) void paged_prefill_mfma_kernel(
       }
     }
   };
-  // fragment-major positions: K (A-frag, 32x32x16): chunk ks = d8/16,
-  // lane = s + 32*((d8%16)/8).  V (B-frag): chunk dt*2+c (dt = d8/32,
-  // c = s/16), lane = (d8%32)+i + 32*((s%16)/8), j = s%8 — 8 2B scatters.
-  const int stk_ks = (d8 >> 4);
-  const int stk_h = (d8 >> 3) & 1;
-  const int stv_dt = d8 >> 5;
-  const int stv_col = d8 & 31;
   auto stage_write = [&](int buf) {
 #pragma unroll
     for (int pass = 0; pass < NPASS; ++pass) {
       const int s = pass * ROWS_PER_PASS + r_in_pass;
+      const int k_byte = (s * HEAD_DIM + d8) * 2 ^ ((s & 7) << 4);
       *reinterpret_cast<bf16x8_t*>(
-          reinterpret_cast<char*>(k_lds2[buf]) +
-          (stk_ks * 64 + s + 32 * stk_h) * 16) = st_k[pass];
-      short* vb = reinterpret_cast<short*>(v_lds2[buf]) +
-                  (((stv_dt * 2 + (s >> 4)) * 64 + stv_col + 32 * ((s >> 3) & 1)) * 8) +
-                  (s & 7);
-#pragma unroll
-      for (int i = 0; i < 8; ++i) vb[i * 8] = st_v[pass][i];
+          reinterpret_cast<char*>(k_lds2[buf]) + k_byte) = st_k[pass];
+      *reinterpret_cast<bf16x8_t*>(v_lds2[buf] + s * HEAD_DIM + d8) = st_v[pass];
     }
   };
 
@@ -166,9 +153,10 @@ __global__ __launch_bounds__(256, 1) void paged_prefill_mfma_kernel(
     f32x16_t acc_s{};
 #pragma unroll
     for (int ks = 0; ks < KCH; ++ks) {
-      // A[kv=col][k = ks*16 + half*8 + j]: fragment-major, one b128 read
+      // A[kv=col][k = ks*16 + half*8 + j] from the swizzled K tile
+      const int byte = (col * HEAD_DIM + ks * 16 + half * 8) * 2 ^ ((col & 7) << 4);
       bf16x8_t a = *reinterpret_cast<const bf16x8_t*>(
-          reinterpret_cast<const char*>(k_lds) + (ks * 64 + lane) * 16);
+          reinterpret_cast<const char*>(k_lds) + byte);
       acc_s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qb[ks], acc_s, 0, 0, 0);
     }
 
@@ -246,9 +234,12 @@ __global__ __launch_bounds__(256, 1) void paged_prefill_mfma_kernel(
     for (int dt = 0; dt < DT; ++dt) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        // B[k=kv][d=col]: fragment-major, one b128 read
-        bf16x8_t b = *reinterpret_cast<const bf16x8_t*>(
-            reinterpret_cast<const char*>(v_lds) + ((dt * 2 + c) * 64 + lane) * 16);
+        // B[k=kv][d=col]: lane gathers V[c*16 + half*8 + j][dt*32 + col]
+        bf16x8_t b;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          b[j] = *reinterpret_cast<const short*>(
+              v_lds + (c * 16 + half * 8 + j) * HEAD_DIM + dt * 32 + col);
         acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[c], b, acc_o[dt], 0, 0, 0);
       }
     }

@@ -31,17 +31,11 @@ typedef __attribute__((ext_vector_type(16))) float f32x16_t;
 #define GS_ROWS (GS_NWAVES * GS_NT * 16)  // 128 W rows per workgroup
 #define GS_KB 64                     // k elems per staged chunk
 
-// Fragment-major LDS images: tiles are stored in exactly the per-lane order
-// the MFMA fragment reads consume, so every fragment load is ONE
-// ds_read_b128 at ((subtile*chunks + chunk)*64 + lane)*16B — consecutive
-// lanes read consecutive 16B, conflict-free — and the staging writes stay
-// single 16B stores.  (The earlier [row][k] image with an XOR swizzle left
-// 4-8-way bank conflicts: row strides are multiples of 64 words.)
-//
-// 16x16x32 frags (16-row subtiles, GS_KB/32 chunks of 32k):
-//   element (r, d) -> chunk d/32, lane r + 16*((d%32)/8), j d%8
-// 32x32x16 frags (32-row subtiles, GS_KB/16 chunks of 16k):
-//   element (r, d) -> chunk d/16, lane r + 32*((d%16)/8), j d%8
+// byte offset of (row, byte_in_row) in a [rows][GS_KB] bf16 LDS tile with the
+// (row&7)<<4 XOR swizzle; row stride = GS_KB*2 = 128 B
+DEVINLINE int swz(int row, int byte_in_row) {
+  return row * (GS_KB * 2) + (byte_in_row ^ ((row & 7) << 4));
+}
 
 template <int MT>  // m tiles of 16 rows per workgroup (up to 8 = 128 rows)
 __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
@@ -94,25 +88,17 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
         st_x[p] = *reinterpret_cast<const bf16x8_t*>(x_base + (long)(p * 32) * K + koff);
     }
   };
-  // 16x16x32 fragment-major positions for this staging thread's 16B chunk:
-  // d8 = (tid&7)*8 -> chunk (tid&7)>>2, kq (tid&7)&3; row r -> subtile r/16,
-  // lane r%16 + 16*kq.  Offsets in 16B units.
-  const int st_kc = (tid & 7) >> 2;
-  const int st_lane16 = (st_row % 16) + 16 * ((tid & 7) & 3);
-  auto fm16 = [&](int sub) { return ((sub * (GS_KB / 32) + st_kc) * 64 + st_lane16) * 16; };
   auto stage_write = [&](int buf) {
 #pragma unroll
-    for (int p = 0; p < GS_ROWS / 32; ++p) {
-      const int row = p * 32 + st_row;
+    for (int p = 0; p < GS_ROWS / 32; ++p)
       *reinterpret_cast<bf16x8_t*>(
-          reinterpret_cast<char*>(w_lds[buf]) + fm16(row / 16)) = st_w[p];
-    }
+          reinterpret_cast<char*>(w_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_w[p];
 #pragma unroll
     for (int p = 0; p < XPASS; ++p) {
       const int r = p * 32 + st_row;
       if (r < XROWS)
         *reinterpret_cast<bf16x8_t*>(
-            reinterpret_cast<char*>(x_lds[buf]) + fm16(r / 16)) = st_x[p];
+            reinterpret_cast<char*>(x_lds[buf]) + swz(r, st_byte)) = st_x[p];
     }
   };
 
@@ -135,19 +121,18 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
 
 #pragma unroll
     for (int kc = 0; kc < 2; ++kc) {
+      const int byte = kc * 64 + kq * 16;
       bf16x8_t a[GS_NT], b[MT];
 #pragma unroll
       for (int nt = 0; nt < GS_NT; ++nt) {
-        const int sub = wave * GS_NT + nt;
+        const int r = (wave * GS_NT + nt) * 16 + row16;
         a[nt] = *reinterpret_cast<const bf16x8_t*>(
-            reinterpret_cast<const char*>(w_lds[buf]) +
-            ((sub * 2 + kc) * 64 + lane) * 16);
+            reinterpret_cast<const char*>(w_lds[buf]) + swz(r, byte));
       }
 #pragma unroll
       for (int mt = 0; mt < MT; ++mt) {
         b[mt] = *reinterpret_cast<const bf16x8_t*>(
-            reinterpret_cast<const char*>(x_lds[buf]) +
-            ((mt * 2 + kc) * 64 + lane) * 16);
+            reinterpret_cast<const char*>(x_lds[buf]) + swz(mt * 16 + row16, byte));
       }
 #pragma unroll
       for (int nt = 0; nt < GS_NT; ++nt)
@@ -237,18 +222,13 @@ __global__ __launch_bounds__(256, 2) void gemm_tile_kernel(
         st_x[p] = *reinterpret_cast<const bf16x8_t*>(x_base + (long)(p * 32) * K + koff);
     }
   };
-  // 32x32x16 fragment-major: d8 = (tid&7)*8 -> chunk ks (tid&7)>>1,
-  // half h (tid&7)&1; row p*32+st_row -> subtile p, lane st_row + 32*h.
-  const int st_ks = (tid & 7) >> 1;
-  const int st_lane32 = st_row + 32 * ((tid & 7) & 1);
-  auto fm32 = [&](int sub) { return ((sub * (GS_KB / 16) + st_ks) * 64 + st_lane32) * 16; };
   auto stage_write = [&](int buf) {
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       *reinterpret_cast<bf16x8_t*>(
-          reinterpret_cast<char*>(w_lds[buf]) + fm32(p)) = st_w[p];
+          reinterpret_cast<char*>(w_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_w[p];
       *reinterpret_cast<bf16x8_t*>(
-          reinterpret_cast<char*>(x_lds[buf]) + fm32(p)) = st_x[p];
+          reinterpret_cast<char*>(x_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_x[p];
     }
   };
 
@@ -266,17 +246,18 @@ __global__ __launch_bounds__(256, 2) void gemm_tile_kernel(
 
 #pragma unroll
     for (int kc = 0; kc < 4; ++kc) {  // 16-k slices of the 64-k chunk
+      const int byte = kc * 32 + half * 16;
       bf16x8_t a[2], b[2];
 #pragma unroll
       for (int mt = 0; mt < 2; ++mt)
         a[mt] = *reinterpret_cast<const bf16x8_t*>(
             reinterpret_cast<const char*>(x_lds[buf]) +
-            (((mh * 2 + mt) * 4 + kc) * 64 + lane) * 16);
+            swz(mh * 64 + mt * 32 + col32, byte));
 #pragma unroll
       for (int nt = 0; nt < 2; ++nt)
         b[nt] = *reinterpret_cast<const bf16x8_t*>(
             reinterpret_cast<const char*>(w_lds[buf]) +
-            (((nh * 2 + nt) * 4 + kc) * 64 + lane) * 16);
+            swz(nh * 64 + nt * 32 + col32, byte));
 #pragma unroll
       for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
@@ -421,24 +402,20 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
         st_x[p] = *reinterpret_cast<const bf16x8_t*>(x_base + (long)(p * 32) * K + koff);
     }
   };
-  const int st_kc = (tid & 7) >> 2;
-  const int st_lane16 = (st_row % 16) + 16 * ((tid & 7) & 3);
-  auto fm16 = [&](int sub) { return ((sub * (GS_KB / 32) + st_kc) * 64 + st_lane16) * 16; };
   auto stage_write = [&](int buf) {
 #pragma unroll
     for (int p = 0; p < GS_ROWS / 32; ++p) {
-      const int sub = (p * 32 + st_row) / 16;
       *reinterpret_cast<bf16x8_t*>(
-          reinterpret_cast<char*>(wg_lds[buf]) + fm16(sub)) = st_g[p];
+          reinterpret_cast<char*>(wg_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_g[p];
       *reinterpret_cast<bf16x8_t*>(
-          reinterpret_cast<char*>(wu_lds[buf]) + fm16(sub)) = st_u[p];
+          reinterpret_cast<char*>(wu_lds[buf]) + swz(p * 32 + st_row, st_byte)) = st_u[p];
     }
 #pragma unroll
     for (int p = 0; p < XPASS; ++p) {
       const int r = p * 32 + st_row;
       if (r < XROWS)
         *reinterpret_cast<bf16x8_t*>(
-            reinterpret_cast<char*>(x_lds[buf]) + fm16(r / 16)) = st_x[p];
+            reinterpret_cast<char*>(x_lds[buf]) + swz(r, st_byte)) = st_x[p];
     }
   };
 
@@ -461,22 +438,20 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
 
 #pragma unroll
     for (int kc = 0; kc < 2; ++kc) {
+      const int byte = kc * 64 + kq * 16;
       bf16x8_t ag[GS_NT], au[GS_NT], b[MT];
 #pragma unroll
       for (int nt = 0; nt < GS_NT; ++nt) {
-        const int sub = wave * GS_NT + nt;
+        const int r = (wave * GS_NT + nt) * 16 + row16;
         ag[nt] = *reinterpret_cast<const bf16x8_t*>(
-            reinterpret_cast<const char*>(wg_lds[buf]) +
-            ((sub * 2 + kc) * 64 + lane) * 16);
+            reinterpret_cast<const char*>(wg_lds[buf]) + swz(r, byte));
         au[nt] = *reinterpret_cast<const bf16x8_t*>(
-            reinterpret_cast<const char*>(wu_lds[buf]) +
-            ((sub * 2 + kc) * 64 + lane) * 16);
+            reinterpret_cast<const char*>(wu_lds[buf]) + swz(r, byte));
       }
 #pragma unroll
       for (int mt = 0; mt < MT; ++mt)
         b[mt] = *reinterpret_cast<const bf16x8_t*>(
-            reinterpret_cast<const char*>(x_lds[buf]) +
-            ((mt * 2 + kc) * 64 + lane) * 16);
+            reinterpret_cast<const char*>(x_lds[buf]) + swz(mt * 16 + row16, byte));
 #pragma unroll
       for (int nt = 0; nt < GS_NT; ++nt)
 #pragma unroll
