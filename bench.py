@@ -35,6 +35,8 @@ def parse_args():
     p.add_argument("--block-size", type=int, default=16)
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--ttft-samples", type=int, default=5)
+    p.add_argument("--timing", action="store_true",
+                   help="print per-phase step timing breakdown to stderr")
     return p.parse_args()
 
 
@@ -147,6 +149,8 @@ def main():
         engine.step()
     for _ in range(args.warmup):
         engine.step()
+    if args.timing:
+        engine.phase_times = {"steps": 0}
 
     rank0_barrier()
     t0 = time.perf_counter()
@@ -158,6 +162,14 @@ def main():
     t1 = time.perf_counter()
 
     elapsed = t1 - t0
+    if args.timing and engine.phase_times:
+        import sys as _sys
+
+        pt = dict(engine.phase_times)
+        n = max(1, pt.pop("steps"))
+        parts = {k: round(v / n * 1e3, 3) for k, v in pt.items()}
+        print(f"[timing] per-step ms over {n} steps: {parts}", file=_sys.stderr)
+        engine.phase_times = None
     if tp > 1:
         dev = "cuda" if device == "cuda" else "cpu"
         t = torch.tensor([elapsed], dtype=torch.float64, device=dev)

@@ -41,6 +41,8 @@ class LLMEngine:
         from .metrics import EngineMetrics
 
         self.metrics = EngineMetrics(self.model_config.model)
+        # optional per-phase step timing (bench --timing): phase -> seconds
+        self.phase_times: Optional[dict] = None
 
     # ------------------------------------------------------------------
     def add_request(
@@ -86,12 +88,27 @@ class LLMEngine:
 
     # ------------------------------------------------------------------
     def step(self) -> list[RequestOutput]:
+        pt = self.phase_times
+        if pt is not None:
+            import torch as _torch
+
+            _sync = _torch.cuda.synchronize if self.worker.device == "cuda" else (lambda: None)
+            _sync()
+            t0 = time.perf_counter()
         sched = self.scheduler.schedule()
         if sched.is_empty:
             return []
+        if pt is not None:
+            t1 = time.perf_counter()
+            pt["schedule"] += t1 - t0
         result = self.worker.execute(sched)
         for it in sched.items:
             it.request.num_computed_tokens += it.num_new_tokens
+        if pt is not None:
+            _sync()
+            t2 = time.perf_counter()
+            pt["execute+sample"] += t2 - t1
+            pt["build_batch"] += getattr(self.worker, "last_build_time", 0.0)
 
         now = time.time()
         sampler_out = result.sampler_output
@@ -137,6 +154,10 @@ class LLMEngine:
                             (m.last_token_time - m.first_token_time)
                             / (req.num_output_tokens - 1)
                         )
+        if pt is not None:
+            t3 = time.perf_counter()
+            pt["postprocess"] += t3 - t2
+            pt["steps"] += 1
         self.metrics.num_running.set(len(self.scheduler.running))
         self.metrics.num_waiting.set(len(self.scheduler.waiting))
         self.metrics.kv_usage.set(

@@ -325,7 +325,11 @@ class Worker:
     # ------------------------------------------------------------------
     def execute(self, sched: SchedulerOutput) -> ExecuteResult:
         """Rank-0 entry point for one engine step."""
+        import time as _time
+
+        _tb = _time.perf_counter()
         batch = self.build_batch(sched)
+        self.last_build_time = _time.perf_counter() - _tb
         if self.tp > 1:
             tp_broadcast_object(("execute", batch))
         logits = self.execute_batch(batch)
