@@ -150,7 +150,9 @@ def main():
     for _ in range(args.warmup):
         engine.step()
     if args.timing:
-        engine.phase_times = {"steps": 0}
+        from collections import defaultdict
+
+        engine.phase_times = defaultdict(float)
 
     rank0_barrier()
     t0 = time.perf_counter()
