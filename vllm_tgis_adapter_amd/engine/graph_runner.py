@@ -18,7 +18,7 @@ from typing import Optional
 
 import torch
 
-_BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256, 384, 512]
+_BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256, 384, 512, 768, 1024]
 
 
 class DecodeGraphRunner:
