@@ -16,18 +16,28 @@ from .request import Request, RequestStatus
 class Detokenizer:
     def __init__(self, tokenizer):
         self.tokenizer = tokenizer
+        self._special = set(tokenizer.all_special_tokens)
+        # id -> (token, token-or-"" when skipping specials); lazily filled
+        self._tok_cache: dict[int, tuple[str, str]] = {}
 
     def _convert(self, ids: list[int], skip_special: bool) -> list[str]:
         toks = self.tokenizer.convert_ids_to_tokens(ids)
         if skip_special:
-            special = set(self.tokenizer.all_special_tokens)
-            return [t if t not in special else "" for t in toks]
+            return [t if t not in self._special else "" for t in toks]
         return toks
+
+    def _one_token(self, token_id: int, skip_special: bool) -> str:
+        hit = self._tok_cache.get(token_id)
+        if hit is None:
+            tok = self.tokenizer.convert_ids_to_tokens([token_id])[0]
+            hit = (tok, "" if tok in self._special else tok)
+            self._tok_cache[token_id] = hit
+        return hit[1] if skip_special else hit[0]
 
     def append_token(self, req: Request, token_id: int) -> str:
         """Incrementally decode one new token; returns the new text fragment."""
         skip = req.sampling_params.skip_special_tokens
-        new_tok = self._convert([token_id], skip)[0]
+        new_tok = self._one_token(token_id, skip)
         req.prev_token_texts.append(new_tok)
 
         toks = req.prev_token_texts
