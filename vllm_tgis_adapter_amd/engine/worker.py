@@ -12,6 +12,7 @@ import gc
 from dataclasses import dataclass, field
 from typing import Optional
 
+import numpy as np
 import torch
 
 from .. import ops
@@ -35,11 +36,11 @@ def _pad_block_tables(tables: list[list[int]], device) -> torch.Tensor:
     if not tables:
         return torch.empty((0, 0), dtype=torch.int32, device=device)
     maxb = max(1, max(len(t) for t in tables))
-    out = torch.zeros((len(tables), maxb), dtype=torch.int32)
+    arr = np.zeros((len(tables), maxb), dtype=np.int32)
     for i, t in enumerate(tables):
         if t:
-            out[i, : len(t)] = torch.tensor(t, dtype=torch.int32)
-    return out.to(device, non_blocking=True)
+            arr[i, : len(t)] = t
+    return torch.from_numpy(arr).to(device, non_blocking=True)
 
 
 class Worker:
@@ -270,8 +271,8 @@ class Worker:
     def execute_batch(self, batch: dict) -> Optional[torch.Tensor]:
         """All ranks: run the forward; returns logits rows on every rank."""
         dev = self.device
-        ids = torch.tensor(batch["token_ids"], dtype=torch.long, device=dev)
-        pos = torch.tensor(batch["positions"], dtype=torch.long, device=dev)
+        ids = torch.from_numpy(np.asarray(batch["token_ids"], dtype=np.int64)).to(dev, non_blocking=True)
+        pos = torch.from_numpy(np.asarray(batch["positions"], dtype=np.int64)).to(dev, non_blocking=True)
         np_seqs = len(batch["prefill_seq_lens"])
 
         # hipGraph replay path: pure-decode batch, every row samples, no LoRA
@@ -284,8 +285,8 @@ class Worker:
             and len(batch["logit_rows"]) == n
             and self.graph_runner.bucket_for(n) is not None
         ):
-            slots = torch.tensor(batch["slot_mapping"], dtype=torch.long, device=dev)
-            seq_lens = torch.tensor(batch["decode_seq_lens"], dtype=torch.int32, device=dev)
+            slots = torch.from_numpy(np.asarray(batch["slot_mapping"], dtype=np.int64)).to(dev, non_blocking=True)
+            seq_lens = torch.from_numpy(np.asarray(batch["decode_seq_lens"], dtype=np.int32)).to(dev, non_blocking=True)
             bt = _pad_block_tables(batch["decode_tables"], dev)
             return self.graph_runner.run(ids, pos, slots, seq_lens, bt)
 
@@ -294,14 +295,14 @@ class Worker:
             num_prefill_seqs=np_seqs,
             num_prefill_tokens=npt,
             num_decode_seqs=len(batch["decode_seq_lens"]),
-            slot_mapping=torch.tensor(batch["slot_mapping"], dtype=torch.long, device=dev),
-            prefill_query_start_loc=torch.tensor(batch["qsl"], dtype=torch.int32, device=dev),
-            prefill_seq_lens=torch.tensor(batch["prefill_seq_lens"], dtype=torch.int32, device=dev),
+            slot_mapping=torch.from_numpy(np.asarray(batch["slot_mapping"], dtype=np.int64)).to(dev, non_blocking=True),
+            prefill_query_start_loc=torch.from_numpy(np.asarray(batch["qsl"], dtype=np.int32)).to(dev, non_blocking=True),
+            prefill_seq_lens=torch.from_numpy(np.asarray(batch["prefill_seq_lens"], dtype=np.int32)).to(dev, non_blocking=True),
             prefill_block_tables=_pad_block_tables(batch["prefill_tables"], dev),
             max_prefill_query_len=max(
                 (b - a for a, b in zip(batch["qsl"], batch["qsl"][1:])), default=0),
             max_prefill_seq_len=max(batch["prefill_seq_lens"], default=0),
-            decode_seq_lens=torch.tensor(batch["decode_seq_lens"], dtype=torch.int32, device=dev),
+            decode_seq_lens=torch.from_numpy(np.asarray(batch["decode_seq_lens"], dtype=np.int32)).to(dev, non_blocking=True),
             decode_block_tables=_pad_block_tables(batch["decode_tables"], dev),
             max_decode_seq_len=max(batch["decode_seq_lens"], default=0),
         )
