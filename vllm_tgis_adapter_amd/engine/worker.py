@@ -111,6 +111,19 @@ class Worker:
             per_seq = (mc.max_model_len + self.block_size - 1) // self.block_size
             self.num_blocks = min(8192, per_seq * cfg.scheduler_config.max_num_seqs + 8)
 
+        if self.tp > 1:
+            # ranks may profile slightly different free memory; the block
+            # manager (rank 0) must never hand out a block id some rank
+            # didn't allocate
+            import torch.distributed as dist
+
+            t = torch.tensor(
+                [self.num_blocks], dtype=torch.int64,
+                device=self.device if self.device == "cuda" else "cpu",
+            )
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            self.num_blocks = int(t.item())
+
         shape = (self.num_blocks, self.block_size, kv_heads_local, mc.head_dim)
         self.kv_caches = [
             (
