@@ -54,7 +54,7 @@ def build_params(args):
     if args.seed is not None:
         p.sampling.seed = args.seed
     if args.guided_regex:
-        p.decoding.guided.regex = args.guided_regex
+        p.decoding.regex = args.guided_regex
     return p
 
 
