@@ -147,6 +147,10 @@ class Worker:
             return
         if os.environ.get("VTA_FORCE_REFERENCE", "0") == "1":
             return  # torch-reference ops sync to host; not capturable
+        if getattr(self.model_config, "num_experts", 0):
+            # MoE routing uses data-dependent shapes (nonzero/index_add);
+            # capturable fixed-capacity routing is tracked for a later round
+            return
         if self.tp > 1 and os.environ.get("VTA_GRAPH_TP", "0") != "1":
             return
         from .graph_runner import DecodeGraphRunner
