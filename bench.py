@@ -106,6 +106,8 @@ def main():
             kind = cmd[0]
             if kind == "execute":
                 worker.execute_batch(cmd[1])
+            elif kind == "execute_packed":
+                worker.execute_batch(worker.recv_packed_batch(cmd[1]))
             elif kind == "add_lora":
                 worker.add_lora(cmd[1], cmd[2])
             elif kind == "barrier":

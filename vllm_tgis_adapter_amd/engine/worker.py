@@ -32,7 +32,12 @@ class ExecuteResult:
     # item-aligned prompt logprob additions handled directly on Request objects
 
 
-def _pad_block_tables(tables: list[list[int]], device) -> torch.Tensor:
+def _pad_block_tables(tables, device) -> torch.Tensor:
+    if isinstance(tables, np.ndarray):  # already padded (TP broadcast path)
+        if tables.size == 0:
+            return torch.empty((0, 0), dtype=torch.int32, device=device)
+        return torch.from_numpy(np.ascontiguousarray(tables, dtype=np.int32)).to(
+            device, non_blocking=True)
     if not tables:
         return torch.empty((0, 0), dtype=torch.int32, device=device)
     maxb = max(1, max(len(t) for t in tables))
@@ -41,6 +46,73 @@ def _pad_block_tables(tables: list[list[int]], device) -> torch.Tensor:
         if t:
             arr[i, : len(t)] = t
     return torch.from_numpy(arr).to(device, non_blocking=True)
+
+
+def _tables_to_array(tables: list[list[int]]) -> np.ndarray:
+    if not tables:
+        return np.zeros((0, 0), dtype=np.int64)
+    maxb = max(1, max(len(t) for t in tables))
+    arr = np.zeros((len(tables), maxb), dtype=np.int64)
+    for i, t in enumerate(tables):
+        if t:
+            arr[i, : len(t)] = t
+    return arr
+
+
+def pack_batch(batch: dict) -> tuple[np.ndarray, np.ndarray]:
+    """Flatten a step batch into (header, payload) int64 arrays for a pair of
+    tensor broadcasts — ~10x cheaper than pickling the dict at batch 256."""
+    pt = _tables_to_array(batch["prefill_tables"])
+    dt = _tables_to_array(batch["decode_tables"])
+    lora = batch["lora_ids"] if batch["lora_ids"] is not None else []
+    parts = [
+        np.asarray(batch["token_ids"], dtype=np.int64),
+        np.asarray(batch["positions"], dtype=np.int64),
+        np.asarray(batch["slot_mapping"], dtype=np.int64),
+        np.asarray(batch["qsl"], dtype=np.int64),
+        np.asarray(batch["prefill_seq_lens"], dtype=np.int64),
+        pt.reshape(-1),
+        np.asarray(batch["decode_seq_lens"], dtype=np.int64),
+        dt.reshape(-1),
+        np.asarray(batch["logit_rows"], dtype=np.int64),
+        np.asarray(lora, dtype=np.int64),
+    ]
+    header = np.array(
+        [batch["num_sample_rows"], len(batch["qsl"]),
+         len(batch["prefill_seq_lens"]), pt.shape[1] if pt.size else 0,
+         len(batch["decode_seq_lens"]), dt.shape[1] if dt.size else 0,
+         1 if batch["lora_ids"] is not None else 0]
+        + [len(p) for p in parts],
+        dtype=np.int64,
+    )
+    return header, np.concatenate(parts) if parts else np.zeros(0, np.int64)
+
+
+def unpack_batch(header: np.ndarray, payload: np.ndarray) -> dict:
+    num_sample_rows, n_qsl, n_pf, pf_maxb, n_dec, dec_maxb, has_lora = header[:7]
+    sizes = header[7:]
+    arrs = []
+    off = 0
+    for n in sizes:
+        arrs.append(payload[off:off + n])
+        off += n
+    (token_ids, positions, slot_mapping, qsl, pf_lens, pf_flat, dec_lens,
+     dec_flat, logit_rows, lora) = arrs
+    return dict(
+        token_ids=token_ids,
+        positions=positions,
+        slot_mapping=slot_mapping,
+        qsl=qsl.tolist(),
+        prefill_seq_lens=pf_lens,
+        prefill_tables=pf_flat.reshape(int(n_pf), int(pf_maxb)) if pf_maxb else
+        np.zeros((0, 0), np.int64),
+        decode_seq_lens=dec_lens,
+        decode_tables=dec_flat.reshape(int(n_dec), int(dec_maxb)) if dec_maxb else
+        np.zeros((0, 0), np.int64),
+        logit_rows=logit_rows.tolist(),
+        num_sample_rows=int(num_sample_rows),
+        lora_ids=lora.tolist() if has_lora else None,
+    )
 
 
 class Worker:
@@ -349,7 +421,13 @@ class Worker:
         batch = self.build_batch(sched)
         self.last_build_time = _time.perf_counter() - _tb
         if self.tp > 1:
-            tp_broadcast_object(("execute", batch))
+            import torch.distributed as dist
+
+            header, payload = pack_batch(batch)
+            tp_broadcast_object(("execute_packed", header.tolist()))
+            t = torch.from_numpy(payload).to(
+                self.device if self.device == "cuda" else "cpu")
+            dist.broadcast(t, src=0)
         logits = self.execute_batch(batch)
 
         ns = batch["num_sample_rows"]
@@ -399,10 +477,24 @@ class Worker:
             kind = cmd[0]
             if kind == "execute":
                 self.execute_batch(cmd[1])
+            elif kind == "execute_packed":
+                self.execute_batch(self.recv_packed_batch(cmd[1]))
             elif kind == "add_lora":
                 self.add_lora(cmd[1], cmd[2])
             elif kind == "stop":
                 return
+
+    def recv_packed_batch(self, header_list: list[int]):
+        """Non-zero ranks: receive the packed step batch broadcast."""
+        import torch.distributed as dist
+
+        header = np.asarray(header_list, dtype=np.int64)
+        total = int(header[7:].sum())
+        t = torch.empty(
+            total, dtype=torch.int64,
+            device=self.device if self.device == "cuda" else "cpu")
+        dist.broadcast(t, src=0)
+        return unpack_batch(header, t.cpu().numpy())
 
     def stop_workers(self) -> None:
         if self.tp > 1 and self.rank == 0:
