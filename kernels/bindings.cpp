@@ -56,6 +56,9 @@ void launch_lora_shrink(float*, const T*, const T*, const int*, int, int, int,
 template <typename T>
 void launch_lora_expand(T*, const float*, const T*, const int*, const float*,
                         int, int, int, int, int, hipStream_t);
+template <typename T>
+void launch_sample_argmax(long*, const T*, const float*, const float*, int,
+                          int, hipStream_t);
 
 namespace {
 
@@ -317,6 +320,28 @@ void lora_bgmv(torch::Tensor out, torch::Tensor x, torch::Tensor a_stack,
   });
 }
 
+void sample_argmax(torch::Tensor out, torch::Tensor logits,
+                   torch::Tensor temps, c10::optional<torch::Tensor> noise) {
+  const int N = logits.size(0);
+  const int V = logits.size(1);
+  TORCH_CHECK(out.scalar_type() == at::ScalarType::Long && out.size(0) == N);
+  TORCH_CHECK(temps.scalar_type() == at::ScalarType::Float && temps.size(0) == N);
+  TORCH_CHECK(logits.is_contiguous());
+  const float* np_ = nullptr;
+  if (noise.has_value()) {
+    TORCH_CHECK(noise->scalar_type() == at::ScalarType::Float &&
+                noise->is_contiguous() && noise->size(0) == N &&
+                noise->size(1) == V);
+    np_ = noise->data_ptr<float>();
+  }
+  DISPATCH_FLOATING(logits.scalar_type(), {
+    launch_sample_argmax<scalar_t>(out.data_ptr<long>(),
+                                   cptr<scalar_t>(logits),
+                                   temps.data_ptr<float>(), np_, N, V,
+                                   current_stream());
+  });
+}
+
 void mfma_probe(torch::Tensor a, torch::Tensor b, torch::Tensor d, int64_t shape) {
   TORCH_CHECK(a.scalar_type() == at::ScalarType::BFloat16);
   auto* ap = reinterpret_cast<const unsigned short*>(a.data_ptr());
@@ -346,6 +371,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "skinny decode GEMM y = x @ w^T, M <= 64 (CDNA4 MFMA)");
   m.def("gemm_skinny_gated", &gemm_skinny_gated,
         "fused gate/up skinny GEMM + SiLU-mul, M <= 64 (CDNA4 MFMA)");
+  m.def("sample_argmax", &sample_argmax,
+        "fused temperature/gumbel-race sampling + greedy argmax (CDNA4)");
   m.def("lora_bgmv", &lora_bgmv,
         "batched multi-LoRA shrink+expand for mixed-adapter batches (CDNA4)");
 }

@@ -336,3 +336,27 @@ def test_lora_bgmv(t, k, n, ranks):
     assert torch.allclose(out.float(), ref, atol=0.25, rtol=3e-2), (
         (out.float() - ref).abs().max().item()
     )
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_sample_argmax(dtype):
+    from vllm_tgis_adapter_amd import ops
+
+    torch.manual_seed(13)
+    n, v = 9, 5000
+    logits = torch.randn(n, v, dtype=dtype, device="cuda") * 3
+    temps = torch.tensor(
+        [0.0, 0.7, 1.3, 0.0, 1.0, 0.5, 2.0, 0.0, 0.9],
+        dtype=torch.float32, device="cuda",
+    )
+    noise = torch.empty(n, v, dtype=torch.float32, device="cuda")
+    noise.exponential_()
+    out = torch.empty(n, dtype=torch.long, device="cuda")
+    ops.sample_argmax(out, logits, temps, noise)
+    lf = logits.float()
+    for i in range(n):
+        if temps[i] == 0:
+            ref = int(torch.argmax(lf[i]))
+        else:
+            ref = int(torch.argmax(lf[i] / temps[i] - noise[i].log()))
+        assert int(out[i]) == ref, (i, int(out[i]), ref)

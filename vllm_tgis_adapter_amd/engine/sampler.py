@@ -73,20 +73,51 @@ class Sampler:
         n, vocab = logits.shape
         assert n == len(requests)
 
-        plain_greedy = True
+        # Fused path (E7): rows that are greedy or plain temperature sampling
+        # (no top-k/top-p filter, penalties, processors, guided masks or
+        # logprob requests) sample in ONE kernel pass via the
+        # exponential-race identity; per-request seeded noise keeps seeded
+        # sampling deterministic.
+        from .. import ops as _ops
+
+        fused_ok = logits.is_cuda and _ops.has_native()
+        simple = True
+        any_sampling = False
         for req in requests:
             p = req.sampling_params
             if (
-                p.temperature != 0.0
-                or p.repetition_penalty != 1.0
+                p.repetition_penalty != 1.0
                 or p.logits_processors
                 or req.guided_state is not None
                 or p.logprobs is not None
                 or (p.min_tokens and req.num_output_tokens < p.min_tokens)
             ):
-                plain_greedy = False
+                simple = False
                 break
-        if plain_greedy:
+            if p.temperature != 0.0:
+                any_sampling = True
+                if not (p.top_k <= 0 or p.top_k >= vocab) or p.top_p < 1.0:
+                    simple = False
+                    break
+        if simple and fused_ok:
+            temps = torch.tensor(
+                [r.sampling_params.temperature for r in requests],
+                dtype=torch.float32, device=logits.device,
+            )
+            noise = None
+            if any_sampling:
+                noise = torch.empty((n, vocab), dtype=torch.float32, device=logits.device)
+                for i, req in enumerate(requests):
+                    if req.sampling_params.temperature != 0.0:
+                        g = self._generator_for(req)
+                        if g is None:
+                            noise[i].exponential_()
+                        else:
+                            noise[i].exponential_(generator=g)
+            out = torch.empty(n, dtype=torch.long, device=logits.device)
+            _ops.sample_argmax(out, logits, temps, noise)
+            return SamplerOutput(token_ids=out.tolist(), logprobs=[None] * n)
+        if simple and not any_sampling:
             sampled_cpu = torch.argmax(logits, dim=-1).tolist()
             return SamplerOutput(token_ids=sampled_cpu, logprobs=[None] * n)
 

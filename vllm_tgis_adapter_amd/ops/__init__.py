@@ -237,6 +237,17 @@ def lora_bgmv(
     _C.lora_bgmv(out, x, a_stack, b_stack, tmp, slots, scales, off)
 
 
+def sample_argmax(
+    out: torch.Tensor,     # [N] int64
+    logits: torch.Tensor,  # [N, V]
+    temps: torch.Tensor,   # [N] float32, 0 = greedy
+    noise: torch.Tensor | None,  # [N, V] float32 Exp(1) draws for temp>0 rows
+) -> None:
+    """Fused temperature + exponential-race sampling / greedy argmax (E7)."""
+    assert _native(logits), "sample_argmax is the GPU path"
+    _C.sample_argmax(out, logits, temps, noise)
+
+
 def topk_softmax(gate_logits: torch.Tensor, top_k: int):
     # Router math is tiny; torch ops are fine on both devices for now.
     return reference.topk_softmax(gate_logits, top_k)
