@@ -80,7 +80,7 @@ class Sampler:
         # sampling deterministic.
         from .. import ops as _ops
 
-        fused_ok = logits.is_cuda and _ops.has_native()
+        fused_ok = _ops.native_enabled(logits)
         simple = True
         any_sampling = False
         for req in requests:

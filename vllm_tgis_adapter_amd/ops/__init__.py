@@ -41,6 +41,12 @@ def has_native() -> bool:
     return _C is not None
 
 
+def native_enabled(t: torch.Tensor) -> bool:
+    """True when the HIP path applies to this tensor (device, build, and the
+    VTA_FORCE_REFERENCE escape hatch all considered) — non-raising."""
+    return t.device.type == "cuda" and not _FORCE_REFERENCE and _C is not None
+
+
 def _native(t: torch.Tensor) -> bool:
     if t.device.type != "cuda":
         return False
