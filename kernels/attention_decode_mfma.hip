@@ -110,8 +110,7 @@ __global__ __launch_bounds__(256, 2) void paged_decode_mfma_kernel(
             (long)kv_head * HEAD_DIM;
 #pragma unroll
         for (int ks = 0; ks < KCH; ++ks)
-          ka[ks] = __builtin_nontemporal_load(
-              reinterpret_cast<const bf16x8_t*>(k_row + ks * 16 + half * 8));
+          ka[ks] = *reinterpret_cast<const bf16x8_t*>(k_row + ks * 16 + half * 8);
       } else {
 #pragma unroll
         for (int ks = 0; ks < KCH; ++ks) ka[ks] = bf16x8_t{};
@@ -133,8 +132,7 @@ __global__ __launch_bounds__(256, 2) void paged_decode_mfma_kernel(
           const long row = ((long)block * block_size + pos % block_size) *
                                kv_row_stride +
                            (long)kv_head * HEAD_DIM + d8;
-          vv[pass] = __builtin_nontemporal_load(
-              reinterpret_cast<const bf16x8_t*>(v_cache + row));
+          vv[pass] = *reinterpret_cast<const bf16x8_t*>(v_cache + row);
         }
       }
 #pragma unroll

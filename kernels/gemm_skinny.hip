@@ -79,8 +79,7 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_kernel(
     const int koff = chunk * GS_KB;
 #pragma unroll
     for (int p = 0; p < GS_ROWS / 32; ++p)
-      st_w[p] = __builtin_nontemporal_load(
-          reinterpret_cast<const bf16x8_t*>(w_base + (long)(p * 32) * K + koff));
+      st_w[p] = *reinterpret_cast<const bf16x8_t*>(w_base + (long)(p * 32) * K + koff);
 #pragma unroll
     for (int p = 0; p < XPASS; ++p) {
       const int r = p * 32 + st_row;
@@ -392,10 +391,8 @@ __global__ __launch_bounds__(256, 2) void gemm_skinny_gated_kernel(
     const int koff = chunk * GS_KB;
 #pragma unroll
     for (int p = 0; p < GS_ROWS / 32; ++p) {
-      st_g[p] = __builtin_nontemporal_load(
-          reinterpret_cast<const bf16x8_t*>(wg_base + (long)(p * 32) * K + koff));
-      st_u[p] = __builtin_nontemporal_load(
-          reinterpret_cast<const bf16x8_t*>(wu_base + (long)(p * 32) * K + koff));
+      st_g[p] = *reinterpret_cast<const bf16x8_t*>(wg_base + (long)(p * 32) * K + koff);
+      st_u[p] = *reinterpret_cast<const bf16x8_t*>(wu_base + (long)(p * 32) * K + koff);
     }
 #pragma unroll
     for (int p = 0; p < XPASS; ++p) {
