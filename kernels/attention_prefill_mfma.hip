@@ -25,7 +25,7 @@ typedef __attribute__((ext_vector_type(16))) float f32x16_t;
 #define KVT 32  // kv slots per tile
 
 template <int HEAD_DIM>
-__global__ __launch_bounds__(256, 1) void paged_prefill_mfma_kernel(
+__global__ __launch_bounds__(256, 2) void paged_prefill_mfma_kernel(
     __hip_bfloat16* __restrict__ out,            // [total_q, nheads, HD]
     const __hip_bfloat16* __restrict__ q,        // [total_q, nheads, HD]
     const __hip_bfloat16* __restrict__ k_cache,  // [nb, bs, kvh, HD]

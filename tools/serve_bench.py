@@ -90,6 +90,35 @@ async def run_load(args) -> dict:
     await asyncio.gather(*(one_stream(stub, text, 4, w) for _ in range(min(8, args.concurrency))))
 
     results: list = []
+    if args.duration > 0:
+        # closed-loop sustained load: C always-on streams, each re-issuing as
+        # it finishes; measure completed tokens between warm and end marks.
+        stop_at = time.perf_counter() + args.duration + args.warmup_s
+        warm_at = time.perf_counter() + args.warmup_s
+
+        async def worker():
+            while time.perf_counter() < stop_at:
+                r: list = []
+                await one_stream(stub, text, args.max_new_tokens, r)
+                now = time.perf_counter()
+                if warm_at < now < stop_at:
+                    results.append(r[0][0] if r else 0)
+
+        t0 = time.perf_counter()
+        await asyncio.gather(*(worker() for _ in range(args.concurrency)))
+        await channel.close()
+        total_tokens = sum(results)
+        return {
+            "metric": "grpc_stream_sustained_tokens_per_s",
+            "value": round(total_tokens / args.duration, 1),
+            "total_tokens": total_tokens,
+            "duration_s": args.duration,
+            "concurrency": args.concurrency,
+            "max_new_tokens": args.max_new_tokens,
+            "prompt_tokens": args.prompt_tokens,
+            "model": args.model,
+        }
+
     t0 = time.perf_counter()
     await asyncio.gather(
         *(one_stream(stub, text, args.max_new_tokens, results)
@@ -124,6 +153,9 @@ def main() -> int:
     ap.add_argument("--port", type=int, default=8033)
     ap.add_argument("--http-port", type=int, default=8000)
     ap.add_argument("--num-gpu-blocks", type=int, default=None)
+    ap.add_argument("--duration", type=float, default=0,
+                    help=">0: closed-loop sustained load for this many seconds")
+    ap.add_argument("--warmup-s", type=float, default=5)
     ap.add_argument("--server-log", default="serve_bench_server.log")
     args = ap.parse_args()
     args.server_log = open(args.server_log, "w")
