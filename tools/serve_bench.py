@@ -75,6 +75,17 @@ async def one_stream(stub, text: str, max_new: int, results: list) -> None:
     results.append((tokens, ttft, time.perf_counter() - t0))
 
 
+def _scrape_gen_tokens(http_port: int) -> float:
+    import urllib.request
+
+    txt = urllib.request.urlopen(
+        f"http://localhost:{http_port}/metrics", timeout=5).read().decode()
+    for line in txt.splitlines():
+        if line.startswith("tgis_amd:generation_tokens_total"):
+            return float(line.split()[-1])
+    return 0.0
+
+
 async def run_load(args) -> dict:
     target = f"localhost:{args.port}"
     await wait_healthy(target)
@@ -104,8 +115,15 @@ async def run_load(args) -> dict:
                 if warm_at < now < stop_at:
                     results.append(r[0][0] if r else 0)
 
+        async def mark(delay):
+            await asyncio.sleep(delay)
+            return _scrape_gen_tokens(args.http_port)
+
         t0 = time.perf_counter()
+        g0_task = asyncio.create_task(mark(args.warmup_s))
+        g1_task = asyncio.create_task(mark(args.warmup_s + args.duration))
         await asyncio.gather(*(worker() for _ in range(args.concurrency)))
+        g0, g1 = await g0_task, await g1_task
         await channel.close()
         total_tokens = sum(results)
         return {
@@ -116,6 +134,7 @@ async def run_load(args) -> dict:
             "concurrency": args.concurrency,
             "max_new_tokens": args.max_new_tokens,
             "prompt_tokens": args.prompt_tokens,
+            "server_side_tokens_per_s": round((g1 - g0) / args.duration, 1),
             "model": args.model,
         }
 
