@@ -28,6 +28,17 @@ class EngineDeadError(RuntimeError):
 _STREAM_END = object()
 
 
+def _dispatch_batch(items: list) -> None:
+    """Runs in the event loop: deliver one step's outputs to their streams.
+    One cross-thread hop per STEP instead of one per request (at batch 256
+    that is 256 loop wakeups per step saved)."""
+    for stream, out in items:
+        stream.queue.put_nowait(out)
+        if out.finished:
+            stream.finished = True
+            stream.queue.put_nowait(_STREAM_END)
+
+
 class _AsyncStream:
     def __init__(self, request_id: str, loop: asyncio.AbstractEventLoop):
         self.request_id = request_id
@@ -166,13 +177,13 @@ class AsyncLLMEngine:
                 if eng.has_unfinished():
                     outputs = eng.step()
                     worked = True
+                    items = []
                     for out in outputs:
                         stream = self._streams.get(out.request_id)
                         if stream is not None:
-                            stream.put_threadsafe(out)
-                            if out.finished:
-                                stream.finished = True
-                                stream.put_threadsafe(_STREAM_END)
+                            items.append((stream, out))
+                    if items:
+                        items[0][0].loop.call_soon_threadsafe(_dispatch_batch, items)
                 if not worked:
                     self._wakeup.wait(timeout=0.05)
                     self._wakeup.clear()
