@@ -192,6 +192,10 @@ class CacheConfig:
 class SchedulerConfig:
     max_num_seqs: int = 256
     max_num_batched_tokens: int = 8192
+    # hold a trickle of arrivals (<batch) up to this age so prefills batch
+    # into fewer non-graphed mixed steps (see scheduler.schedule)
+    prefill_admit_batch: int = 4
+    prefill_admit_delay_s: float = 0.1
     # chunked prefill is always on; a prompt longer than the remaining token
     # budget is split across steps (reference tolerates multiple prompt-only
     # outputs: grpc_server.py:369-373)

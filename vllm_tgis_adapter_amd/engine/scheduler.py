@@ -129,6 +129,22 @@ class Scheduler:
             budget -= chunk
 
         # 2. Admit waiting requests with the remaining budget.
+        #
+        # Prefill-admission batching: with a busy decode batch, a steady
+        # trickle of arrivals would otherwise make nearly EVERY step a mixed
+        # prefill+decode step — which cannot replay the hipGraph decode
+        # buckets and runs eager (~2-3x slower per step under sustained gRPC
+        # load, tools/serve_bench.py r1).  Hold arrivals briefly so prefills
+        # amortize into fewer mixed steps; an idle engine admits immediately.
+        if (
+            self.waiting
+            and out.items
+            and len(self.waiting) < self.config.prefill_admit_batch
+        ):
+            oldest = self.waiting[0]
+            age = time.time() - (oldest.metrics.arrival_time or 0.0)
+            if age < self.config.prefill_admit_delay_s:
+                return out
         while self.waiting and budget > 0 and len(self.running) < self.config.max_num_seqs:
             req = self.waiting[0]
             chunk, samples = self._schedule_chunk(req, budget)
