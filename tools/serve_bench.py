@@ -107,7 +107,10 @@ async def run_load(args) -> dict:
         stop_at = time.perf_counter() + args.duration + args.warmup_s
         warm_at = time.perf_counter() + args.warmup_s
 
+        import random as _random
+
         async def worker():
+            await asyncio.sleep(_random.uniform(0, args.stagger))
             while time.perf_counter() < stop_at:
                 r: list = []
                 await one_stream(stub, text, args.max_new_tokens, r)
@@ -175,6 +178,8 @@ def main() -> int:
     ap.add_argument("--duration", type=float, default=0,
                     help=">0: closed-loop sustained load for this many seconds")
     ap.add_argument("--warmup-s", type=float, default=5)
+    ap.add_argument("--stagger", type=float, default=0,
+                    help="randomize stream start times over this many seconds")
     ap.add_argument("--server-log", default="serve_bench_server.log")
     args = ap.parse_args()
     args.server_log = open(args.server_log, "w")
