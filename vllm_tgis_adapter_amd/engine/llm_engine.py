@@ -43,6 +43,15 @@ class LLMEngine:
         self.metrics = EngineMetrics(self.model_config.model)
         # optional per-phase step timing (bench --timing): phase -> seconds
         self.phase_times: Optional[dict] = None
+        import os as _os
+
+        if _os.environ.get("VTA_STEP_TIMING", "0") == "1":
+            from collections import defaultdict
+
+            self.phase_times = defaultdict(float)
+            self._timing_autoprint = True
+        else:
+            self._timing_autoprint = False
 
     # ------------------------------------------------------------------
     def add_request(
@@ -158,6 +167,20 @@ class LLMEngine:
             t3 = time.perf_counter()
             pt["postprocess"] += t3 - t2
             pt["steps"] += 1
+            pt["tokens"] += len(sampler_out.token_ids)
+            pt["prefill_steps"] += 1 if sched.items and any(
+                i.num_new_tokens > 1 for i in sched.items) else 0
+            if self._timing_autoprint and pt["steps"] >= 128:
+                import sys as _sys
+
+                n = pt.pop("steps")
+                toks = pt.pop("tokens")
+                pf = pt.pop("prefill_steps")
+                parts = {k: round(v / n * 1e3, 2) for k, v in pt.items()}
+                print(f"[step-timing] {n} steps ({pf} with prefill, "
+                      f"{toks} tokens): per-step ms {parts}",
+                      file=_sys.stderr, flush=True)
+                pt.clear()
         self.metrics.num_running.set(len(self.scheduler.running))
         self.metrics.num_waiting.set(len(self.scheduler.waiting))
         self.metrics.kv_usage.set(
