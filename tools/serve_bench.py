@@ -40,7 +40,11 @@ def start_server(args) -> subprocess.Popen:
     ]
     if args.num_gpu_blocks:
         cmd += ["--num-gpu-blocks", str(args.num_gpu_blocks)]
-    return subprocess.Popen(cmd, stdout=args.server_log, stderr=args.server_log)
+    env = dict(os.environ)
+    if args.step_timing:
+        env["VTA_STEP_TIMING"] = "1"
+    return subprocess.Popen(cmd, stdout=args.server_log, stderr=args.server_log,
+                            env=env)
 
 
 async def wait_healthy(target: str, deadline_s: float = 300.0) -> None:
@@ -180,6 +184,8 @@ def main() -> int:
     ap.add_argument("--warmup-s", type=float, default=5)
     ap.add_argument("--stagger", type=float, default=0,
                     help="randomize stream start times over this many seconds")
+    ap.add_argument("--step-timing", action="store_true",
+                    help="run the server with VTA_STEP_TIMING=1")
     ap.add_argument("--server-log", default="serve_bench_server.log")
     args = ap.parse_args()
     args.server_log = open(args.server_log, "w")
