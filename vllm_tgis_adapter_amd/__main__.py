@@ -53,7 +53,18 @@ async def start_servers(args: "argparse.Namespace") -> None:
     sock_addr = (args.host or "", args.port)
     sock = create_server_socket(sock_addr)
 
-    engine = AsyncLLMEngine(engine_config_from_args(args))
+    cfg = engine_config_from_args(args)
+    use_mp = (
+        not getattr(args, "disable_frontend_multiprocessing", False)
+        and cfg.tensor_parallel_size == 1
+        and int(os.environ.get("WORLD_SIZE", "1")) == 1
+    )
+    if use_mp:
+        from .engine.mp_engine import AsyncMPEngine
+
+        engine = AsyncMPEngine(cfg)
+    else:
+        engine = AsyncLLMEngine(cfg)
     tasks: list[asyncio.Task] = []
     try:
         add_logging_wrappers(engine)
@@ -111,12 +122,33 @@ def parse_args(argv=None) -> "argparse.Namespace":
     return postprocess_tgis_args(parser.parse_args(argv))
 
 
+def run_tp_worker(args: "argparse.Namespace") -> None:
+    """Non-zero torchrun ranks: serve broadcast commands from rank 0.
+
+    ``torchrun --nproc-per-node N python -m vllm_tgis_adapter_amd
+    --num-gpus N ...`` — rank 0 runs the dual server + engine; the other
+    ranks build their model shard and execute broadcast step batches
+    (one process per GPU over RCCL).
+    """
+    from .engine.worker import Worker
+    from .parallel import init_distributed
+
+    cfg = engine_config_from_args(args)
+    init_distributed(cfg.tensor_parallel_size, device=cfg.resolve_device())
+    worker = Worker(cfg)
+    worker.init_kv_cache()
+    worker.worker_loop()
+
+
 def main(argv=None) -> None:
     args = parse_args(argv)
     from . import __version__
 
     logger.info("vllm_tgis_adapter_amd version %s", __version__)
     logger.info("args: %s", args)
+    if int(os.environ.get("WORLD_SIZE", "1")) > 1 and int(os.environ.get("RANK", "0")) != 0:
+        run_tp_worker(args)
+        return
     loop = asyncio.new_event_loop()
     task = loop.create_task(start_servers(args))
     run_and_catch_termination_cause(loop, task)

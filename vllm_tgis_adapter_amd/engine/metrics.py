@@ -61,3 +61,40 @@ class EngineMetrics:
             Gauge, "tgis_amd:gpu_cache_usage_perc", "KV cache usage fraction",
             labelnames=self.labelnames,
         ).labels(**labels)
+
+    # -- process-boundary support (mp_engine): ship snapshots parent-ward ---
+    def snapshot(self) -> dict:
+        """Raw values for mirroring into another process's registry."""
+        return {
+            "request_success": self.request_success._value.get(),
+            "prompt_tokens": self.prompt_tokens._value.get(),
+            "generation_tokens": self.generation_tokens._value.get(),
+            "num_running": self.num_running._value.get(),
+            "num_waiting": self.num_waiting._value.get(),
+            "kv_usage": self.kv_usage._value.get(),
+            "ttft_sum": self.ttft._sum.get(),
+            "ttft_count": sum(b.get() for b in self.ttft._buckets),
+            "tpot_sum": self.time_per_output_token._sum.get(),
+            "tpot_count": sum(b.get() for b in self.time_per_output_token._buckets),
+        }
+
+    def apply_snapshot(self, snap: dict, prev: dict) -> None:
+        """Mirror a child-process snapshot (counters by delta, gauges by set)."""
+        self.request_success.inc(snap["request_success"] - prev.get("request_success", 0))
+        self.prompt_tokens.inc(snap["prompt_tokens"] - prev.get("prompt_tokens", 0))
+        self.generation_tokens.inc(snap["generation_tokens"] - prev.get("generation_tokens", 0))
+        self.num_running.set(snap["num_running"])
+        self.num_waiting.set(snap["num_waiting"])
+        self.kv_usage.set(snap["kv_usage"])
+        # histograms: reflect sum/count movement into the +Inf bucket so
+        # rate() and averages stay correct across the process boundary
+        d_ttft = snap["ttft_count"] - prev.get("ttft_count", 0)
+        if d_ttft > 0:
+            avg = (snap["ttft_sum"] - prev.get("ttft_sum", 0.0)) / d_ttft
+            for _ in range(int(d_ttft)):
+                self.ttft.observe(avg)
+        d_tpot = snap["tpot_count"] - prev.get("tpot_count", 0)
+        if d_tpot > 0:
+            avg = (snap["tpot_sum"] - prev.get("tpot_sum", 0.0)) / d_tpot
+            for _ in range(int(d_tpot)):
+                self.time_per_output_token.observe(avg)

@@ -173,6 +173,9 @@ def add_tgis_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
     parser.add_argument("--adapter-cache", type=str)
     parser.add_argument("--prefix-store-path", type=str,
                         help="Deprecated, use --adapter-cache")
+    parser.add_argument("--disable-frontend-multiprocessing", action="store_true",
+                        help="run the engine in the server process instead of "
+                             "its own process (reference flag; MP is the default)")
     parser.add_argument("--speculator-name", type=str)
     parser.add_argument("--speculator-n-candidates", type=int)
     parser.add_argument("--speculator-max-batch-size", type=int)
