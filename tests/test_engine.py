@@ -199,3 +199,62 @@ def test_delta_outputs_concatenate_to_full():
     ref = greedy(e, "delta stream check", max_tokens=10)
     assert tokens == ref.outputs[0].token_ids
     assert text == ref.outputs[0].text
+
+
+def test_pack_unpack_batch_roundtrip():
+    from vllm_tgis_adapter_amd.engine.worker import pack_batch, unpack_batch
+
+    batch = dict(
+        token_ids=[1, 2, 3, 4, 5],
+        positions=[0, 1, 2, 0, 7],
+        slot_mapping=[10, 11, 12, 80, 81],
+        qsl=[0, 3],
+        prefill_seq_lens=[3],
+        prefill_tables=[[0, 1]],
+        decode_seq_lens=[5, 9],
+        decode_tables=[[5], [6, 7]],
+        logit_rows=[2, 3, 4],
+        num_sample_rows=3,
+        lora_ids=None,
+    )
+    header, payload = pack_batch(batch)
+    out = unpack_batch(header, payload)
+    assert list(out["token_ids"]) == batch["token_ids"]
+    assert list(out["positions"]) == batch["positions"]
+    assert list(out["slot_mapping"]) == batch["slot_mapping"]
+    assert out["qsl"] == batch["qsl"]
+    assert list(out["prefill_seq_lens"]) == batch["prefill_seq_lens"]
+    assert out["prefill_tables"].tolist() == [[0, 1]]
+    assert out["decode_tables"].tolist() == [[5, 0], [6, 7]]
+    assert out["logit_rows"] == batch["logit_rows"]
+    assert out["num_sample_rows"] == 3
+    assert out["lora_ids"] is None
+
+    batch["lora_ids"] = [0, 0, 0, 7, 7]
+    header, payload = pack_batch(batch)
+    out = unpack_batch(header, payload)
+    assert out["lora_ids"] == batch["lora_ids"]
+
+
+def test_scheduler_batches_trickling_prefills():
+    """With a busy decode batch, a single fresh arrival waits (up to the
+    admission delay) instead of forcing a mixed prefill step immediately."""
+    eng = make_engine()
+    eng.config.scheduler_config.prefill_admit_batch = 4
+    eng.config.scheduler_config.prefill_admit_delay_s = 30.0
+    for i in range(2):
+        eng.add_request(f"d{i}", None, list(range(10, 26)),
+                        SamplingParams(temperature=0.0, max_tokens=32))
+    # reach decode steady state
+    for _ in range(3):
+        eng.step()
+    eng.add_request("new", None, list(range(30, 46)),
+                    SamplingParams(temperature=0.0, max_tokens=4))
+    for _ in range(2):
+        eng.step()
+    # the new request must still be waiting (admission held)
+    assert any(r.request_id == "new" for r in eng.scheduler.waiting)
+    # drop the delay: admitted on the next step
+    eng.config.scheduler_config.prefill_admit_delay_s = 0.0
+    eng.step()
+    assert not any(r.request_id == "new" for r in eng.scheduler.waiting)
