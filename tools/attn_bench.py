@@ -34,13 +34,17 @@ def main():
     ap.add_argument("--heads", type=int, default=32)
     ap.add_argument("--hd", type=int, default=128)
     ap.add_argument("--bs", type=int, default=16)
+    ap.add_argument("--only", choices=["prefill", "decode"], default=None)
     args = ap.parse_args()
     assert torch.cuda.is_available()
     torch.manual_seed(0)
     dt = torch.bfloat16
     out = {}
 
-    for name, nseq, ctx in [("prefill 16x512", 16, 512), ("prefill 4x2048", 4, 2048)]:
+    prefill_cases = [("prefill 16x512", 16, 512), ("prefill 4x2048", 4, 2048)]
+    if args.only == "decode":
+        prefill_cases = []
+    for name, nseq, ctx in prefill_cases:
         total_q = nseq * ctx
         nb = (ctx + args.bs - 1) // args.bs
         kc = torch.randn(nseq * nb + 8, args.bs, args.kvh, args.hd, dtype=dt, device="cuda")
@@ -57,7 +61,10 @@ def main():
                      "TFLOPs": round(flops / us / 1e6, 1)}
         print(name, out[name], flush=True)
 
-    for name, nseq, ctx in [("decode 256x576", 256, 576), ("decode 64x512", 64, 512)]:
+    decode_cases = [("decode 256x576", 256, 576), ("decode 64x512", 64, 512)]
+    if args.only == "prefill":
+        decode_cases = []
+    for name, nseq, ctx in decode_cases:
         nb = (ctx + args.bs - 1) // args.bs
         kc = torch.randn(nseq * nb + 8, args.bs, args.kvh, args.hd, dtype=dt, device="cuda")
         vc = torch.randn_like(kc)
