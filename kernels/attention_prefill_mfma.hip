@@ -64,14 +64,8 @@ __global__ __launch_bounds__(256, 2) void paged_prefill_mfma_kernel(
   // ---- LDS tiles, double-buffered: tile t+1 streams in while the mfma
   // phase reads tile t, so the per-tile HBM round trip is overlapped and
   // only one __syncthreads per tile remains.
-  //
-  // V is stored TRANSPOSED (v_t[d][kv], rows padded to VSTRIDE elems = 80 B
-  // so the 32-lane b128 reads spread banks): the PV B-fragment is then ONE
-  // aligned ds_read_b128 per (dt, c) instead of 8 scalar 2B gathers — PMC
-  // r1: 64 gathers/tile/wave dominated VALU+LDS issue (SQ_LDS_BANK_CONFLICT
-  // ~34% of LDS cycles).
   __shared__ __hip_bfloat16 k_lds2[2][KVT * HEAD_DIM];  // XOR-swizzled rows
-  __shared__ __hip_bfloat16 v_lds2[2][HEAD_DIM * (KVT + 8)];  // transposed
+  __shared__ __hip_bfloat16 v_lds2[2][KVT * HEAD_DIM];  // linear
 
   // ---- load this wave's Q sub-tile as B fragments ----------------------
   // B[k][q]: lane holds Q[q=col][ks*16 + half*8 + j]
@@ -131,7 +125,6 @@ __global__ __launch_bounds__(256, 2) void paged_prefill_mfma_kernel(
       }
     }
   };
-  constexpr int VSTRIDE = KVT + 8;  // 80 B rows: banks spread for b128 reads
   auto stage_write = [&](int buf) {
 #pragma unroll
     for (int pass = 0; pass < NPASS; ++pass) {
@@ -139,10 +132,7 @@ __global__ __launch_bounds__(256, 2) void paged_prefill_mfma_kernel(
       const int k_byte = (s * HEAD_DIM + d8) * 2 ^ ((s & 7) << 4);
       *reinterpret_cast<bf16x8_t*>(
           reinterpret_cast<char*>(k_lds2[buf]) + k_byte) = st_k[pass];
-      // transposed scatter: element (s, d8+i) -> v_t[d8+i][s] (bit copy)
-      short* vt = reinterpret_cast<short*>(v_lds2[buf]) + d8 * VSTRIDE + s;
-#pragma unroll
-      for (int i = 0; i < 8; ++i) vt[i * VSTRIDE] = st_v[pass][i];
+      *reinterpret_cast<bf16x8_t*>(v_lds2[buf] + s * HEAD_DIM + d8) = st_v[pass];
     }
   };
 
@@ -244,9 +234,12 @@ __global__ __launch_bounds__(256, 2) void paged_prefill_mfma_kernel(
     for (int dt = 0; dt < DT; ++dt) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        // B[k=kv][d=col] from the transposed image: one b128 read
-        bf16x8_t b = *reinterpret_cast<const bf16x8_t*>(
-            v_lds + (dt * 32 + col) * VSTRIDE + c * 16 + half * 8);
+        // B[k=kv][d=col]: lane gathers V[c*16 + half*8 + j][dt*32 + col]
+        bf16x8_t b;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          b[j] = *reinterpret_cast<const short*>(
+              v_lds + (c * 16 + half * 8 + j) * HEAD_DIM + dt * 32 + col);
         acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[c], b, acc_o[dt], 0, 0, 0);
       }
     }
