@@ -59,6 +59,9 @@ void launch_lora_expand(T*, const float*, const T*, const int*, const float*,
 template <typename T>
 void launch_sample_argmax(long*, const T*, const float*, const float*, int,
                           int, hipStream_t);
+void launch_moe_gemm(__hip_bfloat16*, const __hip_bfloat16*,
+                     const __hip_bfloat16*, const int*, int, int, int, int,
+                     bool, hipStream_t);
 
 namespace {
 
@@ -320,6 +323,29 @@ void lora_bgmv(torch::Tensor out, torch::Tensor x, torch::Tensor a_stack,
   });
 }
 
+void moe_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+              torch::Tensor seg_off, bool gated) {
+  const int T = x.size(0);
+  const int K = x.size(1);
+  const int E = w.size(0);
+  const long wn = w.size(1);
+  const int N = gated ? (int)(wn / 2) : (int)wn;
+  TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16 &&
+              w.scalar_type() == at::ScalarType::BFloat16 &&
+              y.scalar_type() == at::ScalarType::BFloat16);
+  TORCH_CHECK(seg_off.scalar_type() == at::ScalarType::Int &&
+              seg_off.numel() == E + 1);
+  TORCH_CHECK(N % 128 == 0 && K % 64 == 0 && w.size(2) == K);
+  TORCH_CHECK(y.size(0) == T && y.size(1) == N);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
+  if (T == 0) return;
+  launch_moe_gemm(reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+                  reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                  reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                  seg_off.data_ptr<int>(), E, N, K, T, gated,
+                  current_stream());
+}
+
 void sample_argmax(torch::Tensor out, torch::Tensor logits,
                    torch::Tensor temps, c10::optional<torch::Tensor> noise) {
   const int N = logits.size(0);
@@ -371,6 +397,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "skinny decode GEMM y = x @ w^T, M <= 64 (CDNA4 MFMA)");
   m.def("gemm_skinny_gated", &gemm_skinny_gated,
         "fused gate/up skinny GEMM + SiLU-mul, M <= 64 (CDNA4 MFMA)");
+  m.def("moe_gemm", &moe_gemm,
+        "grouped MoE GEMM over expert segments (CDNA4 MFMA)");
   m.def("sample_argmax", &sample_argmax,
         "fused temperature/gumbel-race sampling + greedy argmax (CDNA4)");
   m.def("lora_bgmv", &lora_bgmv,

@@ -360,3 +360,34 @@ def test_sample_argmax(dtype):
         else:
             ref = int(torch.argmax(lf[i] / temps[i] - noise[i].log()))
         assert int(out[i]) == ref, (i, int(out[i]), ref)
+
+
+@pytest.mark.parametrize(("t", "e", "h", "inter", "topk_counts"), [
+    (24, 4, 256, 512, None), (130, 8, 512, 1024, None), (5, 4, 128, 128, None),
+])
+def test_moe_gemm(t, e, h, inter, topk_counts):
+    from vllm_tgis_adapter_amd import ops
+
+    torch.manual_seed(21)
+    dt = torch.bfloat16
+    x = torch.randn(t, h, dtype=dt, device="cuda") / 4
+    w13 = torch.randn(e, 2 * inter, h, dtype=dt, device="cuda") / 8
+    w2 = torch.randn(e, h, inter, dtype=dt, device="cuda") / 8
+    ids = torch.randint(0, e, (t,), device="cuda")
+    sort_idx = torch.argsort(ids)
+    xs = x[sort_idx].contiguous()
+    counts = torch.bincount(ids, minlength=e)
+    seg = torch.zeros(e + 1, dtype=torch.int32, device="cuda")
+    seg[1:] = counts.cumsum(0).to(torch.int32)
+
+    hmid = ops.moe_gemm(xs, w13, seg, gated=True)
+    y = ops.moe_gemm(hmid, w2, seg, gated=False)
+
+    ids_sorted = ids[sort_idx]
+    for i in range(t):
+        ee = int(ids_sorted[i])
+        g = xs[i].float() @ w13[ee, :inter].float().t()
+        u = xs[i].float() @ w13[ee, inter:].float().t()
+        hr = torch.nn.functional.silu(g) * u
+        ref = hr @ w2[ee].float().t()
+        assert torch.allclose(y[i].float(), ref, atol=0.5, rtol=3e-2), i

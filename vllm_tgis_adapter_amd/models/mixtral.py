@@ -43,6 +43,19 @@ class MoEBlock(nn.Module):
         flat_ids = ids.reshape(-1)                        # [T*k]
         flat_w = weights.reshape(-1).to(x.dtype)
         token_idx = torch.arange(t, device=x.device).repeat_interleave(self.top_k)
+        if ops.moe_gemm_usable(x, self.hidden, self.inter):
+            # grouped-GEMM path (E16): sort token-expert pairs by expert and
+            # run ONE gated launch + ONE down launch over all segments
+            sort_idx = torch.argsort(flat_ids)
+            rows = token_idx[sort_idx]
+            counts = torch.bincount(flat_ids, minlength=self.num_experts)
+            seg = torch.zeros(self.num_experts + 1, dtype=torch.int32, device=x.device)
+            seg[1:] = counts.cumsum(0).to(torch.int32)
+            xs = x[rows].contiguous()
+            h = ops.moe_gemm(xs, self.w13, seg, gated=True)
+            ye = ops.moe_gemm(h, self.w2, seg, gated=False)
+            out.index_add_(0, rows, ye * flat_w[sort_idx, None])
+            return tp_all_reduce(out)
         # Launch-only expert loop: no data-dependent host branches (`any()`
         # would sync per expert and break hipGraph capture); empty selections
         # run zero-row GEMMs, which are free.
@@ -109,3 +122,42 @@ class MixtralForCausalLM(nn.Module):
 
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
         return self.lm_head(hidden)
+
+    # ------------------------------------------------------------------
+    def load_weights(self, weights: dict[str, torch.Tensor]) -> None:
+        """Load an HF-layout Mixtral state dict (TP-aware slicing).
+
+        HF names: block_sparse_moe.experts.E.{w1,w2,w3} = gate/down/up,
+        block_sparse_moe.gate = router (reference model family parity).
+        """
+        from ..parallel import get_tp_rank
+
+        r = get_tp_rank()
+        for i, layer in enumerate(self.layers):
+            p = f"model.layers.{i}."
+            layer.attn.qkv_proj.load_full_weights([
+                weights[p + "self_attn.q_proj.weight"],
+                weights[p + "self_attn.k_proj.weight"],
+                weights[p + "self_attn.v_proj.weight"],
+            ])
+            layer.attn.o_proj.load_full_weight(weights[p + "self_attn.o_proj.weight"])
+            moe = layer.moe
+            moe.gate.data.copy_(
+                weights[p + "block_sparse_moe.gate.weight"].to(moe.gate.dtype))
+            for e in range(moe.num_experts):
+                ep = f"{p}block_sparse_moe.experts.{e}."
+                w1 = weights[ep + "w1.weight"]  # gate [I, H]
+                w3 = weights[ep + "w3.weight"]  # up   [I, H]
+                w2 = weights[ep + "w2.weight"]  # down [H, I]
+                sl = slice(r * moe.inter, (r + 1) * moe.inter)
+                moe.w13.data[e, :moe.inter].copy_(w1[sl].to(moe.w13.dtype))
+                moe.w13.data[e, moe.inter:].copy_(w3[sl].to(moe.w13.dtype))
+                moe.w2.data[e].copy_(w2[:, sl].to(moe.w2.dtype))
+            layer.input_norm.weight.data.copy_(
+                weights[p + "input_layernorm.weight"].to(self.cfg.dtype))
+            layer.post_norm.weight.data.copy_(
+                weights[p + "post_attention_layernorm.weight"].to(self.cfg.dtype))
+        self.embed.load_full_weight(weights["model.embed_tokens.weight"])
+        self.final_norm.weight.data.copy_(weights["model.norm.weight"].to(self.cfg.dtype))
+        if not self.cfg.tie_word_embeddings and "lm_head.weight" in weights:
+            self.lm_head.load_full_weight(weights["lm_head.weight"])

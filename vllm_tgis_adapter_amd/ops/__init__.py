@@ -254,6 +254,30 @@ def sample_argmax(
     _C.sample_argmax(out, logits, temps, noise)
 
 
+def moe_gemm(
+    x: torch.Tensor,        # [T, K] rows sorted by expert
+    w: torch.Tensor,        # [E, 2N, K] (gated) or [E, N, K]
+    seg_off: torch.Tensor,  # [E+1] int32 device row offsets
+    gated: bool,
+) -> torch.Tensor:
+    """Grouped GEMM over expert segments (E16); gated fuses SiLU(g)*u."""
+    assert _native(x)
+    n = w.shape[1] // 2 if gated else w.shape[1]
+    out = torch.empty((x.shape[0], n), dtype=x.dtype, device=x.device)
+    _C.moe_gemm(out, x, w, seg_off, gated)
+    return out
+
+
+def moe_gemm_usable(x: torch.Tensor, hidden: int, inter: int) -> bool:
+    return (
+        native_enabled(x)
+        and x.dtype == torch.bfloat16
+        and inter % 128 == 0
+        and hidden % 128 == 0
+        and hidden % 64 == 0
+    )
+
+
 def topk_softmax(gate_logits: torch.Tensor, top_k: int):
     # Router math is tiny; torch ops are fine on both devices for now.
     return reference.topk_softmax(gate_logits, top_k)
