@@ -219,10 +219,21 @@ class Worker:
             return
         if os.environ.get("VTA_FORCE_REFERENCE", "0") == "1":
             return  # torch-reference ops sync to host; not capturable
-        if getattr(self.model_config, "num_experts", 0):
-            # MoE routing uses data-dependent shapes (nonzero/index_add);
-            # capturable fixed-capacity routing is tracked for a later round
-            return
+        mc = self.model_config
+        if getattr(mc, "num_experts", 0):
+            # the grouped-GEMM MoE path (argsort/bincount/index_add — all
+            # shape-static) is capturable; the per-expert fallback loop uses
+            # nonzero() and is not
+            inter_local = mc.intermediate_size // max(1, self.tp)
+            grouped_ok = (
+                ops.has_native()
+                and mc.dtype == torch.bfloat16
+                and inter_local % 128 == 0
+                and mc.hidden_size % 128 == 0
+                and mc.hidden_size % 64 == 0
+            )
+            if not grouped_ok:
+                return
         if self.tp > 1 and os.environ.get("VTA_GRAPH_TP", "0") != "1":
             return
         from .graph_runner import DecodeGraphRunner
