@@ -48,7 +48,10 @@ class MoEBlock(nn.Module):
             # run ONE gated launch + ONE down launch over all segments
             sort_idx = torch.argsort(flat_ids)
             rows = token_idx[sort_idx]
-            counts = torch.bincount(flat_ids, minlength=self.num_experts)
+            # capture-safe histogram (torch.bincount host-syncs on its input)
+            counts = torch.zeros(self.num_experts, dtype=torch.int32, device=x.device)
+            counts.scatter_add_(
+                0, flat_ids, torch.ones_like(flat_ids, dtype=torch.int32))
             seg = torch.zeros(self.num_experts + 1, dtype=torch.int32, device=x.device)
             seg[1:] = counts.cumsum(0).to(torch.int32)
             xs = x[rows].contiguous()
