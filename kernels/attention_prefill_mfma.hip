@@ -64,7 +64,11 @@ __global__ __launch_bounds__(256, 2) void paged_prefill_mfma_kernel(
   // ---- LDS tiles, double-buffered: tile t+1 streams in while the mfma
   // phase reads tile t, so the per-tile HBM round trip is overlapped and
   // only one __syncthreads per tile remains.
-  __shared__ __hip_bfloat16 k_lds2[2][KVT * HEAD_DIM];  // XOR-swizzled rows
+  // K rows padded to HEAD_DIM+8 elems (272 B at hd=128): row base banks then
+  // stride 4 per row (gcd 16 with 64 banks) -> A-fragment reads are 2-way
+  // conflicted instead of the 4-way XOR classes, and staging writes stay
+  // contiguous b128 (PMC r1: SQ_LDS_BANK_CONFLICT ~34% of LDS cycles).
+  __shared__ __hip_bfloat16 k_lds2[2][KVT * (HEAD_DIM + 8)];
   __shared__ __hip_bfloat16 v_lds2[2][KVT * HEAD_DIM];  // linear
 
   // ---- load this wave's Q sub-tile as B fragments ----------------------
@@ -129,7 +133,7 @@ __global__ __launch_bounds__(256, 2) void paged_prefill_mfma_kernel(
 #pragma unroll
     for (int pass = 0; pass < NPASS; ++pass) {
       const int s = pass * ROWS_PER_PASS + r_in_pass;
-      const int k_byte = (s * HEAD_DIM + d8) * 2 ^ ((s & 7) << 4);
+      const int k_byte = (s * (HEAD_DIM + 8) + d8) * 2;
       *reinterpret_cast<bf16x8_t*>(
           reinterpret_cast<char*>(k_lds2[buf]) + k_byte) = st_k[pass];
       *reinterpret_cast<bf16x8_t*>(v_lds2[buf] + s * HEAD_DIM + d8) = st_v[pass];
@@ -153,8 +157,8 @@ __global__ __launch_bounds__(256, 2) void paged_prefill_mfma_kernel(
     f32x16_t acc_s{};
 #pragma unroll
     for (int ks = 0; ks < KCH; ++ks) {
-      // A[kv=col][k = ks*16 + half*8 + j] from the swizzled K tile
-      const int byte = (col * HEAD_DIM + ks * 16 + half * 8) * 2 ^ ((col & 7) << 4);
+      // A[kv=col][k = ks*16 + half*8 + j] from the padded K tile
+      const int byte = (col * (HEAD_DIM + 8) + ks * 16 + half * 8) * 2;
       bf16x8_t a = *reinterpret_cast<const bf16x8_t*>(
           reinterpret_cast<const char*>(k_lds) + byte);
       acc_s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qb[ks], acc_s, 0, 0, 0);
