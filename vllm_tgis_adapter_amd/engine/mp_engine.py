@@ -16,8 +16,6 @@ from __future__ import annotations
 
 import asyncio
 import multiprocessing as mp
-import os
-import queue
 import time
 import traceback
 from typing import AsyncIterator, Optional

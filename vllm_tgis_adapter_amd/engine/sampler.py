@@ -6,8 +6,9 @@ extraction, vectorised over the step's sampling rows.  All tensor math runs
 on-device; the per-position logprob dicts the TGIS wire format needs
 (grpc_server.py:701-756) are assembled host-side from one batched top-k.
 
-The fused HIP sampling kernel will take over the penalty/filter/sample path;
-this module stays the orchestration + host assembly layer.
+Greedy and plain temperature-sampling rows take the fused HIP kernel
+(kernels/sampling.hip, exponential-race identity) in one pass; top-k/top-p,
+penalties, processors and logprob extraction run the vectorised torch path.
 """
 
 from __future__ import annotations

@@ -4,9 +4,11 @@ Same attention stack as llama; the MLP is a top-k routed mixture of SwiGLU
 experts.  Experts are TP-sharded on the intermediate dimension (every rank
 holds a slice of every expert), so the only collective per MoE block is the
 same single all-reduce a dense row-parallel MLP needs — the right trade at
-xGMI's per-link bandwidth for the TP=4 baseline config.  The expert compute
-is batched per expert (sort tokens by expert, one GEMM per expert); the
-grouped-GEMM HIP kernel will replace the per-expert loop.
+xGMI's per-link bandwidth for the TP=4 baseline config.  Expert compute runs
+through the grouped-GEMM HIP kernel (kernels/moe_gemm.hip): tokens sorted by
+expert, one fused gate/up+SiLU launch plus one down launch covering every
+expert segment — shape-static, so MoE decode steps hipGraph-capture; a
+per-expert torch loop remains as the fallback for non-128-aligned dims.
 """
 
 from __future__ import annotations
