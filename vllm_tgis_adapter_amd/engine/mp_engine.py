@@ -109,7 +109,7 @@ class AsyncMPEngine:
         self._lora_futs: dict[int, asyncio.Future] = {}
         self._lora_fut_seq = 0
         self._errored_with: Optional[BaseException] = None
-        self._reader_started = False
+        self._reader_loop = None
         from .metrics import EngineMetrics
 
         self._metrics = EngineMetrics(self.model_config.model)
@@ -138,11 +138,16 @@ class AsyncMPEngine:
         return self.model_config
 
     def _ensure_reader(self) -> None:
-        if self._reader_started:
-            return
         loop = asyncio.get_event_loop()
+        if self._reader_loop is loop:
+            return
+        if self._reader_loop is not None:
+            try:
+                self._reader_loop.remove_reader(self._out_parent.fileno())
+            except Exception:
+                pass
         loop.add_reader(self._out_parent.fileno(), self._drain_outputs)
-        self._reader_started = True
+        self._reader_loop = loop
 
     def _drain_outputs(self) -> None:
         try:
