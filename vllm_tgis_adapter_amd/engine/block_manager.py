@@ -10,7 +10,6 @@ allocate/free; ref-counted to allow future prefix sharing.
 from __future__ import annotations
 
 from collections import deque
-from typing import Optional
 
 from .request import Request
 

@@ -18,7 +18,6 @@ sampler (HIP masking kernel planned once the C++ walker lands).
 
 from __future__ import annotations
 
-import json as _json
 from typing import Callable, Optional
 
 from .types import StructuredOutputsParams

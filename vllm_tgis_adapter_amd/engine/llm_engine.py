@@ -15,7 +15,7 @@ from ..parallel import get_tp_rank, init_distributed
 from .block_manager import BlockManager
 from .config import EngineConfig
 from .detokenizer import Detokenizer, StopChecker
-from .request import Request, RequestStatus
+from .request import Request
 from .scheduler import Scheduler
 from .tokenizer import get_tokenizer
 from .types import LoRARequest, RequestOutput, SamplingParams

@@ -9,7 +9,6 @@ decode kernel over the rest without re-gathering.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Optional
 
 import torch
 

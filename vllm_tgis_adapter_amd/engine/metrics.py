@@ -3,7 +3,6 @@ tests/test_http_server.py:32-34 of the reference)."""
 
 from __future__ import annotations
 
-from typing import Optional
 
 from prometheus_client import REGISTRY, Counter, Gauge, Histogram
 

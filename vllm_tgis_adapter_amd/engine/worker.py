@@ -9,7 +9,7 @@ samples.  Non-zero ranks sit in ``worker_loop`` executing broadcast commands.
 from __future__ import annotations
 
 import gc
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 import numpy as np

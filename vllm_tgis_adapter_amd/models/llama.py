@@ -23,7 +23,6 @@ from ..parallel.layers import (
     ParallelLMHead,
     RowParallelLinear,
     VocabParallelEmbedding,
-    _init_weight,
 )
 
 
