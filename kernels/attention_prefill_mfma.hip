@@ -164,17 +164,34 @@ __global__ __launch_bounds__(256, 2) void paged_prefill_mfma_kernel(
       acc_s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qb[ks], acc_s, 0, 0, 0);
     }
 
-    // ---- masked online softmax (stats per q row = this lane's col) -----
+    // ---- online softmax (stats per q row = this lane's col) --------------
+    // Fast path for fully-visible tiles (strictly below the causal diagonal
+    // of every q row of this wave AND inside seq_len — the common case for
+    // long prompts): no per-element mask math.  PMC r1: the masked loops
+    // were ~1/3 of the kernel's VALU issue.
+    const int wave_q_pos_min = seq_len - q_len + tile_base + wave * 32;
+    // wave-uniform: the q_len term implies every lane's q row is valid
+    const bool full_tile =
+        (kv_base + KVT - 1 <= wave_q_pos_min) &&
+        (kv_base + KVT <= seq_len) && (tile_base + wave * 32 + 31 < q_len);
     float s_val[16];
     float local_max = -FLT_MAX;
+    if (full_tile) {
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int kv_local = (r & 3) + 8 * (r >> 2) + 4 * half;
-      const int kv_pos = kv_base + kv_local;
-      float sv = acc_s[r] * scale;
-      if (!q_valid || kv_pos > q_pos || kv_pos >= seq_len) sv = -FLT_MAX;
-      s_val[r] = sv;
-      local_max = fmaxf(local_max, sv);
+      for (int r = 0; r < 16; ++r) {
+        s_val[r] = acc_s[r] * scale;
+        local_max = fmaxf(local_max, s_val[r]);
+      }
+    } else {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kv_local = (r & 3) + 8 * (r >> 2) + 4 * half;
+        const int kv_pos = kv_base + kv_local;
+        float sv = acc_s[r] * scale;
+        if (!q_valid || kv_pos > q_pos || kv_pos >= seq_len) sv = -FLT_MAX;
+        s_val[r] = sv;
+        local_max = fmaxf(local_max, sv);
+      }
     }
     const float tile_max = fmaxf(local_max, __shfl_xor(local_max, 32, 64));
     const float m_new = fmaxf(m_state, tile_max);
@@ -183,10 +200,18 @@ __global__ __launch_bounds__(256, 2) void paged_prefill_mfma_kernel(
 
     float p[16];
     float local_sum = 0.f;
+    if (full_tile) {
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      p[r] = s_val[r] == -FLT_MAX ? 0.f : __expf(s_val[r] - m_new);
-      local_sum += p[r];
+      for (int r = 0; r < 16; ++r) {
+        p[r] = __expf(s_val[r] - m_new);
+        local_sum += p[r];
+      }
+    } else {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        p[r] = s_val[r] == -FLT_MAX ? 0.f : __expf(s_val[r] - m_new);
+        local_sum += p[r];
+      }
     }
     l_state = l_state * rescale + local_sum + __shfl_xor(local_sum, 32, 64);
 
