@@ -40,6 +40,8 @@ def start_server(args) -> subprocess.Popen:
     ]
     if args.num_gpu_blocks:
         cmd += ["--num-gpu-blocks", str(args.num_gpu_blocks)]
+    if args.enable_prefix_caching:
+        cmd.append("--enable-prefix-caching")
     env = dict(os.environ)
     if args.step_timing:
         env["VTA_STEP_TIMING"] = "1"
@@ -186,6 +188,7 @@ def main() -> int:
                     help="randomize stream start times over this many seconds")
     ap.add_argument("--step-timing", action="store_true",
                     help="run the server with VTA_STEP_TIMING=1")
+    ap.add_argument("--enable-prefix-caching", action="store_true")
     ap.add_argument("--server-log", default="serve_bench_server.log")
     args = ap.parse_args()
     args.server_log = open(args.server_log, "w")

@@ -184,6 +184,7 @@ def _from_hf_config(model: str, hf: dict) -> ModelConfig:
 class CacheConfig:
     block_size: int = 16
     gpu_memory_utilization: float = 0.85
+    enable_prefix_caching: bool = False
     num_gpu_blocks: Optional[int] = None  # None => profile at init
     kv_cache_dtype: str = "auto"  # "auto" => model dtype
 

@@ -33,7 +33,10 @@ class LLMEngine:
         self.tokenizer = get_tokenizer(self.model_config)
         self.worker = Worker(config)
         num_blocks = self.worker.init_kv_cache()
-        self.block_manager = BlockManager(num_blocks, config.cache_config.block_size)
+        self.block_manager = BlockManager(
+            num_blocks, config.cache_config.block_size,
+            enable_prefix_caching=config.cache_config.enable_prefix_caching,
+        )
         self.scheduler = Scheduler(config.scheduler_config, self.block_manager)
         self.detokenizer = Detokenizer(self.tokenizer)
         self.stop_checker = StopChecker(self.model_config.max_model_len)
@@ -113,6 +116,7 @@ class LLMEngine:
         result = self.worker.execute(sched)
         for it in sched.items:
             it.request.num_computed_tokens += it.num_new_tokens
+            self.block_manager.register_prefix(it.request)
         if pt is not None:
             _sync()
             t2 = time.perf_counter()

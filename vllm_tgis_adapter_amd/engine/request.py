@@ -69,6 +69,9 @@ class Request:
         # KV bookkeeping
         self.block_ids: list[int] = []
         self.num_computed_tokens = 0  # tokens whose KV is in cache
+        # prefix-caching state (block_manager.match_prefix/register_prefix)
+        self.prefix_key = None
+        self.registered_blocks = 0
 
         # Incremental detokenization state
         self.output_text = ""

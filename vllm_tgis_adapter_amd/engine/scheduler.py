@@ -147,6 +147,8 @@ class Scheduler:
                 return out
         while self.waiting and budget > 0 and len(self.running) < self.config.max_num_seqs:
             req = self.waiting[0]
+            if req.num_computed_tokens == 0 and not req.block_ids:
+                self.block_manager.match_prefix(req)
             chunk, samples = self._schedule_chunk(req, budget)
             if chunk <= 0:
                 break

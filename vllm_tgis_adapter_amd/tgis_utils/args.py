@@ -126,6 +126,8 @@ def make_engine_arg_parser(parser: argparse.ArgumentParser) -> argparse.Argument
     parser.add_argument("--tensor-parallel-size", type=int, default=None)
     parser.add_argument("--max-num-seqs", type=int, default=256)
     parser.add_argument("--max-num-batched-tokens", type=int, default=8192)
+    parser.add_argument("--enable-prefix-caching", action="store_true",
+                        help="share full prompt-prefix KV blocks across requests")
     parser.add_argument("--block-size", type=int, default=16)
     parser.add_argument("--gpu-memory-utilization", type=float, default=0.85)
     parser.add_argument("--num-gpu-blocks", type=int, default=None)
@@ -272,6 +274,7 @@ def engine_config_from_args(args: argparse.Namespace):
             block_size=args.block_size,
             gpu_memory_utilization=args.gpu_memory_utilization,
             num_gpu_blocks=args.num_gpu_blocks,
+            enable_prefix_caching=args.enable_prefix_caching,
         ),
         scheduler_config=SchedulerConfig(
             max_num_seqs=args.max_num_seqs,
