@@ -110,3 +110,35 @@ def test_llama_engine_greedy_determinism_gpu():
     b = _run_engine("llama-1b", "bfloat16", greedy_only=True)
     for rid in a:
         assert a[rid].outputs[0].token_ids == b[rid].outputs[0].token_ids
+
+
+def test_ngram_spec_decode_gpu():
+    """Speculative decoding on the HIP path: repetitive prompt, fewer engine
+    steps than tokens generated, exact token-count contract."""
+    from vllm_tgis_adapter_amd.engine import (
+        EngineConfig, LLMEngine, ModelConfig, SamplingParams,
+    )
+    from vllm_tgis_adapter_amd.engine.config import CacheConfig, SchedulerConfig
+
+    mc = ModelConfig.from_model_arg("llama-1b", dtype="bfloat16")
+    cfg = EngineConfig(
+        model_config=mc,
+        cache_config=CacheConfig(block_size=16, num_gpu_blocks=1024),
+        scheduler_config=SchedulerConfig(max_num_seqs=8, max_num_batched_tokens=2048),
+        device="cuda",
+        speculative_model="ngram",
+        speculative_num_tokens=4,
+    )
+    engine = LLMEngine(cfg)
+    prompt = [7, 8, 9, 10] * 12
+    engine.add_request("sp", None, prompt,
+                       SamplingParams(temperature=0.0, max_tokens=24))
+    steps = 0
+    final = None
+    while engine.has_unfinished():
+        for o in engine.step():
+            if o.finished:
+                final = o
+        steps += 1
+        assert steps < 200
+    assert final is not None and len(final.outputs[0].token_ids) == 24

@@ -214,6 +214,9 @@ class EngineConfig:
     max_loras: int = 8
     max_lora_rank: int = 64
     seed: int = 0
+    # speculative decoding: "ngram" enables prompt-lookup drafts (E17)
+    speculative_model: "str | None" = None
+    speculative_num_tokens: int = 4
     # load deterministic rank-independent synthetic weights (TP equivalence
     # tests) instead of per-shard random init
     load_synthetic_weights: bool = False

@@ -41,7 +41,11 @@ class LLMEngine:
         self.detokenizer = Detokenizer(self.tokenizer)
         self.stop_checker = StopChecker(self.model_config.max_model_len)
         self.eos_token_id = self.tokenizer.eos_token_id
+        from . import spec as _spec
         from .metrics import EngineMetrics
+
+        self.spec_enabled = _spec.is_ngram_spec(config.speculative_model)
+        self.spec_k = config.speculative_num_tokens
 
         self.metrics = EngineMetrics(self.model_config.model)
         # optional per-phase step timing (bench --timing): phase -> seconds
@@ -107,6 +111,14 @@ class LLMEngine:
             _sync = _torch.cuda.synchronize if self.worker.device == "cuda" else (lambda: None)
             _sync()
             t0 = time.perf_counter()
+        if self.spec_enabled:
+            from . import spec as _spec
+
+            for req in self.scheduler.running:
+                req.spec_draft = (
+                    _spec.propose(req, self.spec_k, self.model_config.max_model_len)
+                    if _spec.eligible(req) else []
+                )
         sched = self.scheduler.schedule()
         if sched.is_empty:
             return []
@@ -124,6 +136,30 @@ class LLMEngine:
             pt["build_batch"] += getattr(self.worker, "last_build_time", 0.0)
 
         now = time.time()
+
+        # speculative results: accept the longest matching draft prefix plus
+        # the model's bonus token; roll computed back past rejected draft KV
+        for it, preds in (result.spec_results or []):
+            req = it.request
+            draft = req.spec_draft
+            req.spec_draft = []
+            if req.status.is_finished:
+                continue
+            accepted = 0
+            while accepted < len(draft) and preds[accepted] == draft[accepted]:
+                accepted += 1
+            req.num_computed_tokens -= len(draft) - accepted
+            new_tokens = draft[:accepted] + [preds[accepted]]
+            if req.metrics.first_token_time is None:
+                req.metrics.first_token_time = now
+            req.metrics.last_token_time = now
+            for tok in new_tokens:
+                if req.status.is_finished:
+                    break
+                req.output_token_ids.append(tok)
+                new_text = self.detokenizer.append_token(req, tok)
+                self.stop_checker.check(req, tok, new_text)
+
         sampler_out = result.sampler_output
         for it, token_id, lp in zip(
             self.worker._sampling_items, sampler_out.token_ids, sampler_out.logprobs

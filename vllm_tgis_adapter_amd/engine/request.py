@@ -72,6 +72,8 @@ class Request:
         # prefix-caching state (block_manager.match_prefix/register_prefix)
         self.prefix_key = None
         self.registered_blocks = 0
+        # speculative decoding: drafts pending verification this step
+        self.spec_draft: list[int] = []
 
         # Incremental detokenization state
         self.output_text = ""
@@ -123,6 +125,13 @@ class Request:
     @property
     def all_token_ids(self) -> list[int]:
         return self.prompt_token_ids + self.output_token_ids
+
+    def token_slice(self, start: int, end: int) -> list[int]:
+        """Tokens [start, end) including any pending speculative draft."""
+        toks = self.prompt_token_ids + self.output_token_ids
+        if self.spec_draft and end > len(toks):
+            toks = toks + self.spec_draft
+        return toks[start:end]
 
     def finish(self, status: RequestStatus, stop_reason: object = None) -> None:
         self.status = status

@@ -95,9 +95,15 @@ class Scheduler:
     # ------------------------------------------------------------------
     def _schedule_chunk(self, req: Request, budget: int) -> tuple[int, bool]:
         """(chunk, samples) under the token budget; 0 chunk => skip."""
-        remaining = req.num_tokens - req.num_computed_tokens
+        remaining = req.num_tokens - req.num_computed_tokens + len(req.spec_draft)
         chunk = min(remaining, budget)
         samples = chunk == remaining
+        if req.spec_draft and chunk < remaining:
+            # draft didn't fit the budget; drop it for this step
+            req.spec_draft = []
+            remaining = req.num_tokens - req.num_computed_tokens
+            chunk = min(remaining, budget)
+            samples = chunk == remaining
         return chunk, samples
 
     def schedule(self) -> SchedulerOutput:
@@ -182,6 +188,7 @@ class Scheduler:
         # Recompute-from-scratch preemption: KV is rebuilt on re-admission
         # (already-generated tokens are recomputed like prompt tokens).
         req.num_computed_tokens = 0
+        req.spec_draft = []
         req.status = RequestStatus.PREEMPTED
         self.waiting.appendleft(req)
         out.preempted.append(req)

@@ -29,6 +29,8 @@ from .scheduler import ScheduledItem, SchedulerOutput
 class ExecuteResult:
     # one entry per sampling item (scheduler items with samples=True, in order)
     sampler_output: SamplerOutput
+    # speculative items: (item, model predictions for draft rows + bonus)
+    spec_results: list = None
     # item-aligned prompt logprob additions handled directly on Request objects
 
 
@@ -303,7 +305,7 @@ class Worker:
             req = it.request
             s = req.num_computed_tokens
             e = s + it.num_new_tokens
-            toks = req.all_token_ids[s:e]
+            toks = req.token_slice(s, e)
             token_ids.extend(toks)
             positions.extend(range(s, e))
             lid = req.lora_request.lora_int_id if req.lora_request else 0
@@ -321,12 +323,19 @@ class Worker:
 
         row = 0
         sampling_items: list[ScheduledItem] = []
+        spec_items: list[tuple[ScheduledItem, int, int]] = []  # (item, row0, D)
+        spec_rows: list[int] = []
         # (item, rows-in-extra-space, prompt token indices, carry_out row or None)
         prompt_lp_specs = []
         for it in ordered:
             req = it.request
             nrow = row + it.num_new_tokens
-            if it.samples:
+            if it.samples and req.spec_draft:
+                d = len(req.spec_draft)
+                # logits rows for the real last token + every draft position
+                spec_items.append((it, len(spec_rows), d))
+                spec_rows.extend(range(nrow - 1 - d, nrow))
+            elif it.samples:
                 sample_rows.append(nrow - 1)
                 sampling_items.append(it)
             if (
@@ -347,6 +356,7 @@ class Worker:
                 prompt_lp_specs.append((it, lp_rows, list(range(s + 1, e)), carry_out))
             row = nrow
 
+        logit_rows_all = sample_rows + spec_rows + extra_rows
         batch = dict(
             token_ids=token_ids,
             positions=positions,
@@ -356,14 +366,16 @@ class Worker:
             prefill_tables=prefill_tables,
             decode_seq_lens=decode_seq_lens,
             decode_tables=decode_tables,
-            logit_rows=sample_rows + extra_rows,
+            logit_rows=logit_rows_all,
             num_sample_rows=len(sample_rows),
             lora_ids=lora_ids if any(lora_ids) else None,
         )
         # host-only bookkeeping (not broadcast)
         self._sampling_items = sampling_items
+        self._spec_items = spec_items
+        self._num_spec_rows = len(spec_rows)
         self._prompt_lp_specs = prompt_lp_specs
-        self._extra_row_base = len(sample_rows)
+        self._extra_row_base = len(sample_rows) + len(spec_rows)
         return batch
 
     # ------------------------------------------------------------------
@@ -448,6 +460,15 @@ class Worker:
         else:
             sampler_out = SamplerOutput(token_ids=[], logprobs=[])
 
+        # speculative verification (E17): greedy argmax over the draft rows;
+        # the engine accepts the longest matching prefix + the bonus token
+        spec_results: list[tuple[ScheduledItem, list[int]]] = []
+        if self._spec_items:
+            srows = logits[ns:ns + self._num_spec_rows]
+            preds = torch.argmax(srows, dim=-1).tolist()
+            for it, row0, d in self._spec_items:
+                spec_results.append((it, preds[row0:row0 + d + 1]))
+
         # prompt logprobs (E8): extra logits rows follow the sampling rows,
         # in the order build_batch appended them to extra_rows.
         off = self._extra_row_base
@@ -475,7 +496,7 @@ class Worker:
                 req._plp_carry = logits[off].float().log_softmax(dim=-1)
                 off += 1
 
-        return ExecuteResult(sampler_output=sampler_out)
+        return ExecuteResult(sampler_output=sampler_out, spec_results=spec_results)
 
     # ------------------------------------------------------------------
     def worker_loop(self) -> None:

@@ -231,11 +231,8 @@ def postprocess_tgis_args(args: argparse.Namespace) -> argparse.Namespace:
             )
         args.speculative_model = args.speculator_name
 
-    if args.speculator_n_candidates or args.speculator_max_batch_size:
-        logger.warning(
-            "speculator_n_candidates and speculator_max_batch_size args are not "
-            "yet supported"
-        )
+    if args.speculator_max_batch_size:
+        logger.warning("speculator_max_batch_size is not yet supported")
     if args.max_batch_size is not None:
         logger.warning(
             "max_batch_size is set to %d but will be ignored for now. "
@@ -281,6 +278,8 @@ def engine_config_from_args(args: argparse.Namespace):
             max_num_batched_tokens=args.max_num_batched_tokens,
         ),
         device=args.device,
+        speculative_model=args.speculative_model,
+        speculative_num_tokens=args.speculator_n_candidates or 4,
         tensor_parallel_size=args.tensor_parallel_size or 1,
         enforce_eager=args.enforce_eager,
         enable_lora=args.enable_lora or bool(args.adapter_cache or args.prefix_store_path),
