@@ -194,9 +194,10 @@ class SchedulerConfig:
     max_num_seqs: int = 256
     max_num_batched_tokens: int = 8192
     # hold a trickle of arrivals (<batch) up to this age so prefills batch
-    # into fewer non-graphed mixed steps (see scheduler.schedule)
+    # into fewer non-graphed mixed steps (see scheduler.schedule); the delay
+    # bounds the TTFT cost of the batching
     prefill_admit_batch: int = 4
-    prefill_admit_delay_s: float = 0.1
+    prefill_admit_delay_s: float = 0.025
     # chunked prefill is always on; a prompt longer than the remaining token
     # budget is split across steps (reference tolerates multiple prompt-only
     # outputs: grpc_server.py:369-373)
