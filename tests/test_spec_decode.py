@@ -101,3 +101,22 @@ def test_spec_stop_sequence_mid_draft():
         steps += 1
         assert steps < 200
     assert done is not None
+
+
+def test_spec_plus_prefix_cache_equivalence():
+    """Both features on: outputs still exactly match the plain engine."""
+    mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+    cfg = EngineConfig(
+        model_config=mc,
+        cache_config=CacheConfig(block_size=16, num_gpu_blocks=256,
+                                 enable_prefix_caching=True),
+        scheduler_config=SchedulerConfig(max_num_seqs=8, max_num_batched_tokens=512),
+        seed=0,
+        speculative_model="ngram",
+    )
+    both = LLMEngine(cfg)
+    reqs = [("r1", REPEATY * 3, {}), ("r2", REPEATY * 3, {}),
+            ("r3", REPEATY * 2 + [9, 9], {})]
+    fancy, _ = run_all(both, reqs)
+    plain, _ = run_all(make(False), reqs)
+    assert fancy == plain
