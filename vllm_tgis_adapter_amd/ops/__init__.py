@@ -174,11 +174,13 @@ def paged_attention_prefill(
 
 
 def linear(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None = None) -> torch.Tensor:
-    """F.linear with a custom CDNA4 skinny-GEMM path for decode batches.
+    """F.linear with a custom CDNA4 skinny-GEMM path for small decode batches.
 
     hipBLASLt runs the M<=64 llama decode shapes at 22-43% of the HBM
     roofline (tools/gemm_bench.py); the hand-written MFMA kernel streams the
-    weight matrix once with waves splitting K in-workgroup.
+    weight matrix once with waves splitting K in-workgroup.  Larger M stays
+    on hipBLASLt (the in-tree tile kernel only reached parity there —
+    VTA_GEMM_MAX_M raises the cutover for experiments).
     """
     max_m = int(os.environ.get("VTA_GEMM_MAX_M", "64"))
     if (
