@@ -80,3 +80,25 @@ def test_mp_metrics_mirrored(mp_engine, loop):
         return False
 
     assert loop.run_until_complete(run())
+
+
+def test_mp_engine_death_detection(loop):
+    mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+    cfg = EngineConfig(
+        model_config=mc,
+        cache_config=CacheConfig(block_size=16, num_gpu_blocks=64),
+        scheduler_config=SchedulerConfig(max_num_seqs=4, max_num_batched_tokens=128),
+        seed=0,
+    )
+    eng = AsyncMPEngine(cfg)
+    try:
+        assert eng.is_running
+        eng._proc.kill()
+        eng._proc.join(timeout=10)
+        assert eng.errored
+        with pytest.raises(Exception):
+            eng.generate(prompt={"prompt_token_ids": [1, 2, 3]},
+                         sampling_params=SamplingParams(max_tokens=2),
+                         request_id="dead")
+    finally:
+        eng.shutdown()
