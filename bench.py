@@ -29,7 +29,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=32)
     p.add_argument("--warmup", type=int, default=8)
     p.add_argument("--model", type=str, default="llama-3-8b")
-    p.add_argument("--batch", type=int, default=256)
+    p.add_argument("--batch", type=int, default=512)
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--dtype", type=str, default="bfloat16")
     p.add_argument("--block-size", type=int, default=16)
