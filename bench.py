@@ -52,7 +52,7 @@ def main():
         # CPU dry-run mode (no GPU in the dev container): tiny model
         args.model = "tiny-llama"
         args.dtype = "float32"
-        args.batch = min(args.batch, 8)
+        args.batch = min(args.batch, 64)
         args.prompt_len = min(args.prompt_len, 64)
 
     from vllm_tgis_adapter_amd import ops
