@@ -61,10 +61,22 @@ CTX = LoRAContext()
 _STACK_CACHE: dict = {}
 
 
+def invalidate_adapter(lora_int_id: int) -> None:
+    """Drop cached stacks containing this adapter (removal / reload)."""
+    for key in [k for k in _STACK_CACHE if lora_int_id in k[0]]:
+        del _STACK_CACHE[key]
+
+
 def set_context(token_lora_ids: Optional[torch.Tensor], adapters: dict[int, LoRAAdapter]) -> None:
     CTX.token_lora_ids = token_lora_ids
     CTX.adapters = adapters
-    CTX.slot_ids = tuple(sorted(adapters))
+    new_slot_ids = tuple(sorted(adapters))
+    if new_slot_ids != CTX.slot_ids:
+        # the stacks are keyed by the active-adapter set; evict stale sets so
+        # GPU memory stays bounded by the live combination, not the history
+        for key in [k for k in _STACK_CACHE if k[0] != new_slot_ids]:
+            del _STACK_CACHE[key]
+    CTX.slot_ids = new_slot_ids
     CTX.token_slots = None
     if token_lora_ids is not None and adapters:
         id2slot = {lid: s for s, lid in enumerate(CTX.slot_ids)}
@@ -126,8 +138,7 @@ def apply_lora(
     # GPU: batched BGMV shrink/expand HIP kernels over the mixed batch (E12)
     if (
         CTX.token_slots is not None
-        and x.device.type == "cuda"
-        and ops.has_native()
+        and ops.native_enabled(x)
         and x.is_contiguous()
         and out.is_contiguous()
         and x.shape[1] % 8 == 0

@@ -30,6 +30,7 @@ def eligible(req: Request) -> bool:
     p = req.sampling_params
     return (
         p.temperature == 0.0
+        and not (p.min_tokens and req.num_output_tokens < p.min_tokens)
         and p.logprobs is None
         and p.prompt_logprobs is None
         and not p.logits_processors

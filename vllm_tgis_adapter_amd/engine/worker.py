@@ -151,6 +151,10 @@ class Worker:
 
         if self.tp > 1 and self.rank == 0:
             tp_broadcast_object(("add_lora", lora_path, lora_int_id))
+        if lora_int_id in self.loras:
+            from .lora import invalidate_adapter
+
+            invalidate_adapter(lora_int_id)
         self.loras[lora_int_id] = load_lora_adapter(
             lora_path, lora_int_id,
             device=self.device, dtype=self.model_config.dtype,
