@@ -23,13 +23,53 @@ from typing import AsyncIterator, Optional
 from .async_engine import _STREAM_END, EngineDeadError, _AsyncStream
 from .config import EngineConfig
 from .tokenizer import get_tokenizer
-from .types import LoRARequest, RequestOutput, SamplingParams
+from .types import (
+    CompletionOutput,
+    LoRARequest,
+    RequestMetrics,
+    RequestOutput,
+    SamplingParams,
+)
+
+
+def _enc_output(o: RequestOutput) -> tuple:
+    """RequestOutput -> plain tuple for the pipe (pickle of flat tuples is
+    several times cheaper than pickling nested dataclass instances; at >30k
+    outputs/s the difference is a visible slice of the front-end core)."""
+    co = o.outputs[0]
+    m = o.metrics
+    return (
+        o.request_id, co.text, co.token_ids, co.cumulative_logprob,
+        co.finish_reason, co.stop_reason, o.finished,
+        o.prompt, o.prompt_token_ids, co.logprobs, o.prompt_logprobs,
+        (m.arrival_time, m.first_scheduled_time, m.time_in_queue,
+         m.first_token_time, m.last_token_time) if (o.finished and m) else None,
+    )
+
+
+def _dec_output(t: tuple) -> RequestOutput:
+    (rid, text, token_ids, cum, fr, sr, fin,
+     prompt, ptoks, lps, plps, mt) = t
+    return RequestOutput(
+        request_id=rid,
+        prompt=prompt,
+        prompt_token_ids=ptoks,
+        outputs=[CompletionOutput(0, text, token_ids, cum, lps, fr, sr)],
+        finished=fin,
+        prompt_logprobs=plps,
+        metrics=RequestMetrics(*mt) if mt else None,
+    )
 
 
 def _engine_proc_main(config: EngineConfig, cmd_conn, out_conn) -> None:
     """Child process: build the engine and run the step loop."""
     try:
-        from .llm_engine import LLMEngine
+        import os
+
+        if os.environ.get("VTA_NULL_ENGINE", "0") == "1":
+            from .null_engine import NullEngine as LLMEngine
+        else:
+            from .llm_engine import LLMEngine
 
         engine = LLMEngine(config)
         out_conn.send(("ready", None))
@@ -54,7 +94,7 @@ def _engine_proc_main(config: EngineConfig, cmd_conn, out_conn) -> None:
                         out_conn.send(("request_error", request_id, repr(e)))
                 elif kind == "abort":
                     out = engine.abort_request(cmd[1])
-                    outs = [out] if out is not None else []
+                    outs = [_enc_output(out)] if out is not None else []
                     out_conn.send(("outputs", outs, [cmd[1]]))
                 elif kind == "add_lora":
                     try:
@@ -68,7 +108,9 @@ def _engine_proc_main(config: EngineConfig, cmd_conn, out_conn) -> None:
                 outputs = engine.step()
                 worked = True
                 if outputs:
-                    out_conn.send(("outputs", outputs, None))
+                    out_conn.send(
+                        ("outputs", [_enc_output(o) for o in outputs], None)
+                    )
             now = time.time()
             if now - last_metrics > 1.0:
                 last_metrics = now
@@ -156,11 +198,12 @@ class AsyncMPEngine:
                 kind = msg[0]
                 if kind == "outputs":
                     _, outputs, forced_end = msg
-                    for out in outputs:
-                        stream = self._streams.get(out.request_id)
+                    for enc in outputs:
+                        stream = self._streams.get(enc[0])
                         if stream is None:
                             continue
-                        stream.queue.put_nowait(out)
+                        out = _dec_output(enc)
+                        stream.push(out)
                         if out.finished:
                             stream.finished = True
                             stream.queue.put_nowait(_STREAM_END)
@@ -241,6 +284,10 @@ class AsyncMPEngine:
         self._ensure_reader()
         loop = asyncio.get_event_loop()
         stream = _AsyncStream(request_id, loop)
+        from .types import RequestOutputKind
+
+        stream.delta = sampling_params.output_kind == RequestOutputKind.DELTA
+        stream.peers = self._streams
 
         async def _gen():
             self._streams[request_id] = stream
