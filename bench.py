@@ -1,16 +1,26 @@
 """Flagship serving benchmark (driver contract).
 
-Measures the engine's steady-state decode throughput (output tokens/s) on the
-BASELINE.json headline config — Llama-3-8B, bf16, synthetic prompts with
-random-init weights — plus p50 TTFT (one prompt prefill through the engine),
-reported in the config block.
+Default ``--mode serve`` measures the BASELINE.json headline metric END TO
+END: output tokens/s delivered to gRPC ``GenerateStream`` clients (+ p50
+TTFT), against the real dual-front-end server on the Llama-3-8B bf16 config
+with synthetic prompts / random-init weights.  The timed region is EXACTLY
+``--steps`` engine steps, barrier+``torch.cuda.synchronize()``-bracketed on
+both sides inside the engine process (CLOCK_MONOTONIC timestamps t0/t1
+exposed via the VTA_BENCH ``/bench/window`` endpoint); the reported value is
+the number of tokens the clients RECEIVED over the wire inside [t0, t1)
+divided by t1-t0.  At steady state with a full always-on decode batch this
+is the serving throughput of the whole stack: scheduler + HIP kernels +
+sampler + detokenizer + pipe transport + proto encode + gRPC delivery.
 
-One step = one continuous-batching engine step over a fixed decode batch
-(scheduler + HIP-kernel forward + sampler + detokenizer — the serving hot
-loop).  For --gpus N the model runs TP=N over RCCL (strong scaling: same
-model, same batch).  Rank 0 drives; other ranks run the broadcast worker
-loop with barrier commands so every rank times the same region; the printed
-value uses the MAX elapsed over ranks.
+``--mode engine`` times the bare engine step loop (scheduler + forward +
+sampler + detokenizer, no wire) — the r1 bench, kept for comparison.
+
+For ``--gpus N`` the driver launches this under torchrun, one rank per GPU
+over RCCL (strong scaling: same model TP=N, same batch).  Rank 0 spawns the
+server process (which becomes TP rank 0 of the group via the inherited
+torchrun env) and drives the client load; ranks >0 run the broadcast worker
+loop.  The window barrier spans all ranks, so rank 0's elapsed is the MAX
+over ranks by construction.
 """
 
 from __future__ import annotations
@@ -18,6 +28,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import torch
@@ -28,6 +39,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=32)
     p.add_argument("--warmup", type=int, default=8)
+    p.add_argument("--mode", choices=["serve", "engine"], default="serve")
     p.add_argument("--model", type=str, default="llama-3-8b")
     p.add_argument("--batch", type=int, default=512)
     p.add_argument("--prompt-len", type=int, default=512)
@@ -35,42 +47,67 @@ def parse_args():
     p.add_argument("--block-size", type=int, default=16)
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--ttft-samples", type=int, default=5)
+    p.add_argument("--grpc-port", type=int, default=18033)
+    p.add_argument("--http-port", type=int, default=18080)
     p.add_argument("--timing", action="store_true",
-                   help="print per-phase step timing breakdown to stderr")
+                   help="engine mode: per-phase step timing to stderr")
+    p.add_argument("--server-log", type=str, default="bench_server.log")
     return p.parse_args()
 
 
-def main():
-    args = parse_args()
-    rank = int(os.environ.get("RANK", "0"))
-    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
-    assert world == args.gpus or args.gpus == 1, (world, args.gpus)
-    tp = args.gpus
+def result_json(args, *, value, ms_per_step, tp, extra_config):
+    cfg = {
+        "model": args.model,
+        "global_batch": args.batch,
+        "seq_len": args.prompt_len,
+        "parallelism": f"tp{tp}",
+        "decode_context": args.prompt_len,
+    }
+    cfg.update(extra_config)
+    return {
+        "metric": "grpc_stream_output_tokens_per_s"
+        if args.mode == "serve" else "engine_output_tokens_per_s",
+        "value": round(value, 2),
+        "unit": "tokens/s",
+        "n_gpus": args.gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(ms_per_step, 3),
+        "higher_is_better": True,
+        "scaling": "strong",
+        "vs_baseline": None,
+        "dtype": "bf16" if args.dtype.startswith("b") else args.dtype,
+        "data": "synthetic",
+        "config": cfg,
+    }
 
-    device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
-    if device == "cpu":
-        # CPU dry-run mode (no GPU in the dev container): tiny model
-        args.model = "tiny-llama"
-        args.dtype = "float32"
-        args.batch = min(args.batch, 64)
-        args.prompt_len = min(args.prompt_len, 64)
 
-    from vllm_tgis_adapter_amd import ops
-    from vllm_tgis_adapter_amd.engine import (
-        EngineConfig,
-        LLMEngine,
-        ModelConfig,
-        SamplingParams,
-    )
-    from vllm_tgis_adapter_amd.engine.config import CacheConfig, SchedulerConfig
+# ---------------------------------------------------------------------------
+# serve mode: the BASELINE wire metric
+# ---------------------------------------------------------------------------
+
+def serve_worker_rank(args, device: str) -> None:
+    """Ranks >0 under torchrun: build the TP shard, serve broadcast batches."""
+    from vllm_tgis_adapter_amd.engine.worker import Worker
     from vllm_tgis_adapter_amd.parallel import init_distributed
 
-    if device == "cuda" and not ops.has_native():
-        raise RuntimeError("HIP extension _C not built — run __graft_entry__.build()")
+    cfg = engine_config(args, device, tp=args.gpus)
+    init_distributed(args.gpus, device=device)
+    worker = Worker(cfg)
+    worker.init_kv_cache()
+    try:
+        worker.worker_loop()
+    except RuntimeError:
+        # rank 0's server process exited (bench teardown) — not a failure
+        pass
+
+
+def engine_config(args, device: str, tp: int):
+    from vllm_tgis_adapter_amd.engine import EngineConfig, ModelConfig
+    from vllm_tgis_adapter_amd.engine.config import CacheConfig, SchedulerConfig
 
     mc = ModelConfig.from_model_arg(args.model, dtype=args.dtype)
-    gen_budget = args.warmup + args.steps + 16
-    cfg = EngineConfig(
+    return EngineConfig(
         model_config=mc,
         cache_config=CacheConfig(block_size=args.block_size),
         scheduler_config=SchedulerConfig(
@@ -81,6 +118,177 @@ def main():
         tensor_parallel_size=tp,
         seed=0,
     )
+
+
+async def serve_drive(args, gen_budget: int) -> dict:
+    """Client side: C always-on streams, exactly-K-steps window, TTFT probes."""
+    import grpc
+    import grpc.aio
+
+    from vllm_tgis_adapter_amd.grpc import proto
+    from vllm_tgis_adapter_amd.grpc.stubs import GenerationStub, HealthStub
+
+    import asyncio
+    import urllib.request
+
+    target = f"localhost:{args.grpc_port}"
+
+    deadline = time.time() + 600
+    while True:
+        try:
+            async with grpc.aio.insecure_channel(target) as ch:
+                resp = await HealthStub(ch).Check(
+                    proto.HealthCheckRequest(service=""))
+                if resp.status == 1:
+                    break
+        except Exception:
+            pass
+        if time.time() > deadline:
+            raise TimeoutError("server never became healthy")
+        await asyncio.sleep(2)
+
+    channel = grpc.aio.insecure_channel(
+        target, options=[("grpc.max_concurrent_streams", 4096)])
+    stub = GenerationStub(channel)
+    # ~prompt_len tokens with the byte-level synthetic tokenizer ("ab " = 3)
+    text = "ab " * max(1, args.prompt_len // 3)
+
+    first_token_evt = [asyncio.Event() for _ in range(args.batch)]
+    # per-stream delivery log: (monotonic_ts, delta_tokens)
+    deliveries: list[list] = [[] for _ in range(args.batch)]
+    ttfts: list[float] = []
+
+    async def one_stream(i: int, max_new: int, probe: bool = False):
+        params = proto.Parameters()
+        params.stopping.max_new_tokens = max_new
+        params.stopping.min_new_tokens = max_new
+        req = proto.SingleGenerationRequest(
+            request=proto.GenerationRequest(text=text), params=params)
+        seen = 0
+        t_start = time.monotonic()
+        try:
+            async for msg in stub.GenerateStream(req):
+                now = time.monotonic()
+                if msg.generated_token_count > seen:
+                    if seen == 0:
+                        if probe:
+                            ttfts.append((now - t_start) * 1e3)
+                        else:
+                            first_token_evt[i].set()
+                    if not probe:
+                        deliveries[i].append(
+                            (now, msg.generated_token_count - seen))
+                    seen = msg.generated_token_count
+        except grpc.aio.AioRpcError:
+            pass  # stream cancelled at teardown
+
+    tasks = [asyncio.get_event_loop().create_task(one_stream(i, gen_budget))
+             for i in range(args.batch)]
+    await asyncio.gather(*(e.wait() for e in first_token_evt))
+
+    # every stream is decoding: arm the exactly-K-steps window
+    def _post():
+        body = json.dumps({"warmup": args.warmup, "steps": args.steps}).encode()
+        req = urllib.request.Request(
+            f"http://localhost:{args.http_port}/bench/window", data=body,
+            headers={"Content-Type": "application/json"})
+        return json.loads(urllib.request.urlopen(req, timeout=600).read())
+
+    window = await asyncio.get_event_loop().run_in_executor(None, _post)
+    t0, t1 = window["t0"], window["t1"]
+
+    # p50 TTFT probes while the batch is still decoding (loaded TTFT)
+    for _ in range(args.ttft_samples):
+        await one_stream(0, 2, probe=True)
+
+    # allow in-flight deliveries to land, then stop the load
+    await asyncio.sleep(0.5)
+    for t in tasks:
+        t.cancel()
+    await asyncio.gather(*tasks, return_exceptions=True)
+    await channel.close()
+
+    wire_tokens = sum(
+        d for log in deliveries for (ts, d) in log if t0 <= ts < t1)
+    ttfts.sort()
+    return {
+        "window": window,
+        "wire_tokens": wire_tokens,
+        "p50_ttft_ms": ttfts[len(ttfts) // 2] if ttfts else None,
+    }
+
+
+def run_serve(args, device: str) -> None:
+    import asyncio
+    import subprocess
+
+    rank = int(os.environ.get("RANK", "0"))
+    if args.gpus > 1 and rank != 0:
+        serve_worker_rank(args, device)
+        return
+
+    cmd = [
+        sys.executable, "-m", "vllm_tgis_adapter_amd",
+        "--model-name", args.model, "--dtype", args.dtype,
+        "--device", device,
+        "--max-num-seqs", str(max(args.batch, 8)),
+        "--max-num-batched-tokens", str(max(8192, args.batch * 2)),
+        "--block-size", str(args.block_size),
+        "--grpc-port", str(args.grpc_port), "--port", str(args.http_port),
+    ]
+    if args.gpus > 1:
+        cmd += ["--num-gpus", str(args.gpus)]
+    env = dict(os.environ)
+    env["VTA_BENCH"] = "1"
+    log = open(args.server_log, "w")
+    srv = subprocess.Popen(cmd, stdout=log, stderr=log, env=env)
+
+    # budget so no stream finishes inside warmup+window (+ramp margin)
+    gen_budget = args.warmup + args.steps + 256
+    try:
+        res = asyncio.run(serve_drive(args, gen_budget))
+    finally:
+        srv.terminate()
+        try:
+            srv.wait(timeout=30)
+        except subprocess.TimeoutExpired:
+            srv.kill()
+
+    win = res["window"]
+    elapsed = win["elapsed_s"]
+    value = res["wire_tokens"] / elapsed
+    out = result_json(
+        args,
+        value=value,
+        ms_per_step=elapsed / args.steps * 1e3,
+        tp=args.gpus,
+        extra_config={
+            "p50_ttft_ms": round(res["p50_ttft_ms"], 2)
+            if res["p50_ttft_ms"] is not None else None,
+            "engine_tokens_per_s": round(win["produced"] / elapsed, 1),
+            "transport": "grpc GenerateStream (loopback client)",
+        },
+    )
+    print(json.dumps(out))
+
+
+# ---------------------------------------------------------------------------
+# engine mode: bare step loop (r1 bench)
+# ---------------------------------------------------------------------------
+
+def run_engine(args, device: str) -> None:
+    rank = int(os.environ.get("RANK", "0"))
+    tp = args.gpus
+
+    from vllm_tgis_adapter_amd import ops
+    from vllm_tgis_adapter_amd.engine import LLMEngine, SamplingParams
+    from vllm_tgis_adapter_amd.parallel import init_distributed
+
+    if device == "cuda" and not ops.has_native():
+        raise RuntimeError("HIP extension _C not built — run __graft_entry__.build()")
+
+    gen_budget = args.warmup + args.steps + 16
+    cfg = engine_config(args, device, tp)
 
     if tp > 1:
         init_distributed(tp, device=device)
@@ -93,8 +301,6 @@ def main():
             torch.cuda.synchronize()
 
     if tp > 1 and rank != 0:
-        # worker ranks: build the worker stack and serve broadcast commands,
-        # timing the same barrier-delimited region as rank 0
         from vllm_tgis_adapter_amd.engine.worker import Worker
         from vllm_tgis_adapter_amd.parallel import tp_broadcast_object
 
@@ -114,10 +320,9 @@ def main():
                 barrier_sync()
                 times.append(time.perf_counter())
             elif kind == "elapsed":
-                # contribute this rank's elapsed to the all-reduce MAX
-                # (nccl needs a device tensor)
                 dev = "cuda" if device == "cuda" else "cpu"
-                t = torch.tensor([times[-1] - times[-2]], dtype=torch.float64, device=dev)
+                t = torch.tensor([times[-1] - times[-2]],
+                                 dtype=torch.float64, device=dev)
                 dist.all_reduce(t, op=dist.ReduceOp.MAX)
             elif kind == "stop":
                 return
@@ -131,22 +336,23 @@ def main():
             tp_broadcast_object(("barrier",))
         barrier_sync()
 
+    mc = cfg.model_config
     vocab = mc.vocab_size
     g = torch.Generator().manual_seed(1234)
 
     def synth_prompt(n):
         return torch.randint(4, vocab - 4, (n,), generator=g).tolist()
 
-    # fill the batch; prompts prefill during warmup
     for i in range(args.batch):
         engine.add_request(
             f"bench-{i}", None, synth_prompt(args.prompt_len),
-            SamplingParams(temperature=0.0, max_tokens=gen_budget + args.prompt_len),
+            SamplingParams(temperature=0.0,
+                           max_tokens=gen_budget + args.prompt_len),
         )
 
-    # warm up until every request finished prefilling, then W more steps
     while engine.scheduler.waiting or any(
-        r.num_computed_tokens < r.num_prompt_tokens for r in engine.scheduler.running
+        r.num_computed_tokens < r.num_prompt_tokens
+        for r in engine.scheduler.running
     ):
         engine.step()
     for _ in range(args.warmup):
@@ -160,19 +366,17 @@ def main():
     t0 = time.perf_counter()
     produced = 0
     for _ in range(args.steps):
-        outs = engine.step()
+        engine.step()
         produced += len(engine.worker._sampling_items)
     rank0_barrier()
     t1 = time.perf_counter()
 
     elapsed = t1 - t0
     if args.timing and engine.phase_times:
-        import sys as _sys
-
         pt = dict(engine.phase_times)
         n = max(1, pt.pop("steps"))
         parts = {k: round(v / n * 1e3, 3) for k, v in pt.items()}
-        print(f"[timing] per-step ms over {n} steps: {parts}", file=_sys.stderr)
+        print(f"[timing] per-step ms over {n} steps: {parts}", file=sys.stderr)
         engine.phase_times = None
     if tp > 1:
         dev = "cuda" if device == "cuda" else "cpu"
@@ -184,7 +388,7 @@ def main():
     tokens_per_s = produced / elapsed
     ms_per_step = elapsed / args.steps * 1e3
 
-    # TTFT: fresh single request, time to first sampled token (p50 of samples)
+    # TTFT: fresh single request, time to first sampled token
     ttfts = []
     for i in range(args.ttft_samples):
         rid = f"ttft-{i}"
@@ -203,7 +407,6 @@ def main():
     ttfts.sort()
     p50_ttft_ms = ttfts[len(ttfts) // 2]
 
-    # drain remaining requests cheaply
     for i in range(args.batch):
         engine.abort_request(f"bench-{i}")
     for i in range(args.ttft_samples):
@@ -211,29 +414,30 @@ def main():
     if tp > 1:
         tp_broadcast_object(("stop",))
 
-    result = {
-        "metric": "output_tokens_per_s",
-        "value": round(tokens_per_s, 2),
-        "unit": "tokens/s",
-        "n_gpus": args.gpus,
-        "steps": args.steps,
-        "warmup": args.warmup,
-        "ms_per_step": round(ms_per_step, 3),
-        "higher_is_better": True,
-        "scaling": "strong",
-        "vs_baseline": None,
-        "dtype": "bf16" if args.dtype.startswith("b") else args.dtype,
-        "data": "synthetic",
-        "config": {
-            "model": args.model,
-            "global_batch": args.batch,
-            "seq_len": args.prompt_len,
-            "parallelism": f"tp{tp}",
-            "p50_ttft_ms": round(p50_ttft_ms, 2),
-            "decode_context": args.prompt_len,
-        },
-    }
-    print(json.dumps(result))
+    out = result_json(
+        args, value=tokens_per_s, ms_per_step=ms_per_step, tp=tp,
+        extra_config={"p50_ttft_ms": round(p50_ttft_ms, 2)},
+    )
+    print(json.dumps(out))
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    assert world == args.gpus or args.gpus == 1, (world, args.gpus)
+
+    device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
+    if device == "cpu":
+        # CPU dry-run mode (no GPU in the dev container): tiny model
+        args.model = "tiny-llama"
+        args.dtype = "float32"
+        args.batch = min(args.batch, 64)
+        args.prompt_len = min(args.prompt_len, 64)
+
+    if args.mode == "serve":
+        run_serve(args, device)
+    else:
+        run_engine(args, device)
 
 
 if __name__ == "__main__":

@@ -137,7 +137,12 @@ def run_tp_worker(args: "argparse.Namespace") -> None:
     init_distributed(cfg.tensor_parallel_size, device=cfg.resolve_device())
     worker = Worker(cfg)
     worker.init_kv_cache()
-    worker.worker_loop()
+    try:
+        worker.worker_loop()
+    except RuntimeError as e:
+        # rank 0 went away mid-broadcast: treat as shutdown, not a crash
+        # (a SIGTERM'd server cannot always broadcast the "stop" command)
+        logger.info("TP worker loop ended: %s", e)
 
 
 def main(argv=None) -> None:

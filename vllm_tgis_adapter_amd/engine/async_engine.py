@@ -122,6 +122,7 @@ class AsyncLLMEngine:
         self.engine = LLMEngine(config)
         self.model_config = config.model_config
         self._streams: dict[str, _AsyncStream] = {}
+        self._bench_fut = None  # concurrent.futures.Future | None
         self._cmds: "queue.Queue[tuple]" = queue.Queue()
         self._errored_with: Optional[BaseException] = None
         self._running = True
@@ -163,6 +164,15 @@ class AsyncLLMEngine:
         self._cmds.put(("add_lora", lora_request, fut))
         self._wakeup.set()
         await asyncio.wrap_future(fut)
+
+    async def bench_window(self, warmup_steps: int, timed_steps: int) -> dict:
+        """Arm the engine's exactly-K-steps timing window (driver bench)."""
+        import concurrent.futures
+
+        fut: concurrent.futures.Future = concurrent.futures.Future()
+        self._cmds.put(("bench_window", warmup_steps, timed_steps, fut))
+        self._wakeup.set()
+        return await asyncio.wrap_future(fut)
 
     def generate(
         self,
@@ -244,6 +254,12 @@ class AsyncLLMEngine:
                             items.append((stream, out))
                     if items:
                         items[0][0].loop.call_soon_threadsafe(_dispatch_batch, items)
+                    if self._bench_fut is not None:
+                        res = eng.bench_window_result()
+                        if res is not None:
+                            fut, self._bench_fut = self._bench_fut, None
+                            if not fut.done():
+                                fut.set_result(res)
                 if not worked:
                     self._wakeup.wait(timeout=0.05)
                     self._wakeup.clear()
@@ -272,6 +288,10 @@ class AsyncLLMEngine:
                 fut.set_result(None)
             except BaseException as e:
                 fut.set_exception(e)
+        elif kind == "bench_window":
+            _, warmup_steps, timed_steps, fut = cmd
+            self.engine.arm_bench_window(warmup_steps, timed_steps)
+            self._bench_fut = fut
         elif kind == "abort":
             request_id = cmd[1]
             out = self.engine.abort_request(request_id)

@@ -103,6 +103,54 @@ class LLMEngine:
         return self.scheduler.has_unfinished()
 
     # ------------------------------------------------------------------
+    # Benchmark step-window (driver contract): time EXACTLY `timed_steps`
+    # engine steps, barrier+synchronize-bracketed on both sides, after
+    # `warmup_steps` armed warmup steps.  Timestamps are CLOCK_MONOTONIC
+    # (system-wide on Linux) so a client process can window its own
+    # delivery timestamps against them.
+    def arm_bench_window(self, warmup_steps: int, timed_steps: int) -> None:
+        self._bench_win = {
+            "warmup": int(warmup_steps), "steps": int(timed_steps),
+            "count": 0, "t0": None, "t1": None, "produced": 0, "result": None,
+        }
+
+    def bench_window_result(self) -> Optional[dict]:
+        bw = getattr(self, "_bench_win", None)
+        return bw["result"] if bw else None
+
+    def _bench_barrier(self) -> None:
+        if self.config.tensor_parallel_size > 1:
+            import torch.distributed as dist
+
+            from ..parallel import tp_broadcast_object
+
+            tp_broadcast_object(("barrier",))
+            dist.barrier()
+        if self.worker.device == "cuda":
+            import torch
+
+            torch.cuda.synchronize()
+
+    def _bench_tick(self, produced_this_step: int) -> None:
+        bw = self._bench_win
+        if bw["result"] is not None:
+            return
+        if bw["t0"] is not None:
+            bw["produced"] += produced_this_step
+        bw["count"] += 1
+        if bw["count"] == bw["warmup"]:
+            self._bench_barrier()
+            bw["t0"] = time.monotonic()
+        elif bw["count"] == bw["warmup"] + bw["steps"]:
+            self._bench_barrier()
+            bw["t1"] = time.monotonic()
+            bw["result"] = {
+                "t0": bw["t0"], "t1": bw["t1"],
+                "elapsed_s": bw["t1"] - bw["t0"],
+                "steps": bw["steps"], "produced": bw["produced"],
+            }
+
+    # ------------------------------------------------------------------
     def step(self) -> list[RequestOutput]:
         pt = self.phase_times
         if pt is not None:
@@ -136,6 +184,7 @@ class LLMEngine:
             pt["build_batch"] += getattr(self.worker, "last_build_time", 0.0)
 
         now = time.time()
+        spec_token_count = 0
 
         # speculative results: accept the longest matching draft prefix plus
         # the model's bonus token; roll computed back past rejected draft KV
@@ -150,6 +199,7 @@ class LLMEngine:
                 accepted += 1
             req.num_computed_tokens -= len(draft) - accepted
             new_tokens = draft[:accepted] + [preds[accepted]]
+            spec_token_count += len(new_tokens)
             if req.metrics.first_token_time is None:
                 req.metrics.first_token_time = now
             req.metrics.last_token_time = now
@@ -221,6 +271,8 @@ class LLMEngine:
                       f"{toks} tokens): per-step ms {parts}",
                       file=_sys.stderr, flush=True)
                 pt.clear()
+        if getattr(self, "_bench_win", None) is not None:
+            self._bench_tick(len(sampler_out.token_ids) + spec_token_count)
         self.metrics.num_running.set(len(self.scheduler.running))
         self.metrics.num_waiting.set(len(self.scheduler.waiting))
         self.metrics.kv_usage.set(

@@ -75,6 +75,7 @@ def _engine_proc_main(config: EngineConfig, cmd_conn, out_conn) -> None:
         out_conn.send(("ready", None))
         running = True
         last_metrics = 0.0
+        bench_pending = False
         while running:
             worked = False
             while cmd_conn.poll(0):
@@ -102,6 +103,9 @@ def _engine_proc_main(config: EngineConfig, cmd_conn, out_conn) -> None:
                         out_conn.send(("lora_ok", cmd[2], None))
                     except BaseException as e:
                         out_conn.send(("lora_ok", cmd[2], repr(e)))
+                elif kind == "bench_window":
+                    engine.arm_bench_window(cmd[1], cmd[2])
+                    bench_pending = True
                 elif kind == "stop":
                     running = False
             if running and engine.has_unfinished():
@@ -111,6 +115,11 @@ def _engine_proc_main(config: EngineConfig, cmd_conn, out_conn) -> None:
                     out_conn.send(
                         ("outputs", [_enc_output(o) for o in outputs], None)
                     )
+                if bench_pending:
+                    res = engine.bench_window_result()
+                    if res is not None:
+                        bench_pending = False
+                        out_conn.send(("bench_result", res))
             now = time.time()
             if now - last_metrics > 1.0:
                 last_metrics = now
@@ -148,6 +157,7 @@ class AsyncMPEngine:
         if kind != "ready":
             raise RuntimeError(f"engine process failed to start: {kind}")
         self._streams: dict[str, _AsyncStream] = {}
+        self._bench_fut: Optional[asyncio.Future] = None
         self._lora_futs: dict[int, asyncio.Future] = {}
         self._lora_fut_seq = 0
         self._errored_with: Optional[BaseException] = None
@@ -227,6 +237,11 @@ class AsyncMPEngine:
                             fut.set_result(None)
                         else:
                             fut.set_exception(RuntimeError(err))
+                elif kind == "bench_result":
+                    fut = self._bench_fut
+                    self._bench_fut = None
+                    if fut is not None and not fut.done():
+                        fut.set_result(msg[1])
                 elif kind == "metrics":
                     self._metrics.apply_snapshot(msg[1], self._metrics_prev)
                     self._metrics_prev = msg[1]
@@ -250,6 +265,16 @@ class AsyncMPEngine:
     async def abort(self, request_id: str) -> None:
         if not self.errored:
             self._cmd_parent.send(("abort", request_id))
+
+    async def bench_window(self, warmup_steps: int, timed_steps: int) -> dict:
+        """Arm the engine's exactly-K-steps timing window; resolves with
+        {t0, t1, elapsed_s, steps, produced} once the window closes."""
+        self._ensure_reader()
+        loop = asyncio.get_event_loop()
+        fut: asyncio.Future = loop.create_future()
+        self._bench_fut = fut
+        self._cmd_parent.send(("bench_window", warmup_steps, timed_steps))
+        return await fut
 
     async def add_lora(self, lora_request: LoRARequest) -> None:
         self._ensure_reader()

@@ -517,6 +517,12 @@ class Worker:
                 self.execute_batch(self.recv_packed_batch(cmd[1]))
             elif kind == "add_lora":
                 self.add_lora(cmd[1], cmd[2])
+            elif kind == "barrier":
+                import torch.distributed as dist
+
+                dist.barrier()
+                if self.device == "cuda":
+                    torch.cuda.synchronize()
             elif kind == "stop":
                 return
 

@@ -103,6 +103,19 @@ async def build_http_server(args: "argparse.Namespace", engine: "AsyncLLMEngine"
 
         return Response(content=generate_latest(REGISTRY), media_type="text/plain")
 
+    import os as _os
+
+    if _os.environ.get("VTA_BENCH", "0") == "1":
+        # benchmark-only control surface (bench.py serve mode): arm the
+        # engine's exactly-K-steps window and long-poll for its result
+        @app.post("/bench/window")
+        async def bench_window(request: Request):
+            body = await request.json()
+            res = await engine.bench_window(
+                int(body.get("warmup", 0)), int(body.get("steps", 1))
+            )
+            return res
+
     @app.post("/v1/completions")
     async def completions(request: Request):
         body = await request.json()
