@@ -287,3 +287,41 @@ def test_validation_error_strings_match_reference_exactly():
     }
     for name, text in expected.items():
         assert getattr(E, name).value == text, name
+
+
+def test_guided_json_schema(grpc_client):
+    import json
+
+    req = _gen_req(max_new=60)
+    req.params.decoding.json_schema = (
+        '{"type": "object", "properties": {"name": {"type": "string"},'
+        ' "age": {"type": "integer"}}, "required": ["name", "age"]}'
+    )
+    r = grpc_client.Generate(req, timeout=120).responses[0]
+    assert r.text
+    if r.stop_reason != 1:  # finished before MAX_TOKENS => complete document
+        doc = json.loads(r.text)
+        assert isinstance(doc["name"], str) and isinstance(doc["age"], int)
+    else:  # truncated: still a prefix of a conforming document
+        assert r.text.lstrip().startswith('{"name"')
+
+
+def test_guided_grammar(grpc_client):
+    req = _gen_req(max_new=30)
+    req.params.decoding.grammar = (
+        'root ::= "SELECT " col " from t1"\ncol ::= "a" | "b"\n'
+    )
+    r = grpc_client.Generate(req, timeout=120).responses[0]
+    assert r.text.startswith("SELECT ")
+    assert "SELECT a from t1".startswith(r.text) or \
+        "SELECT b from t1".startswith(r.text)
+
+
+def test_guided_bad_schema_aborts(grpc_client):
+    import grpc as _grpc
+
+    req = _gen_req(max_new=4)
+    req.params.decoding.json_schema = '{"type": "object"}'  # unsupported
+    with pytest.raises(_grpc.RpcError) as e:
+        grpc_client.Generate(req, timeout=30)
+    assert e.value.code() == _grpc.StatusCode.INVALID_ARGUMENT

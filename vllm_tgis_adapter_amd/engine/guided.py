@@ -525,6 +525,22 @@ class _ChoiceAutomaton:
         return state in self.choices
 
 
+def validate_structured_outputs(params: StructuredOutputsParams) -> None:
+    """Compile-check the constraint so malformed patterns/schemas/grammars
+    abort INVALID_ARGUMENT at request conversion (reference behavior:
+    engine-side param validation surfaces at grpc_server.py:606-627)."""
+    if params.regex is not None:
+        RegexAutomaton(params.regex)
+    elif params.json is not None:
+        from .json_schema import schema_to_regex
+
+        RegexAutomaton(schema_to_regex(params.json))
+    elif params.grammar is not None:
+        from .grammar import GrammarAutomaton
+
+        GrammarAutomaton(params.grammar)
+
+
 def build_guided_state(params: StructuredOutputsParams, tokenizer) -> GuidedState:
     eos = tokenizer.eos_token_id
     if params.regex is not None:
@@ -533,9 +549,19 @@ def build_guided_state(params: StructuredOutputsParams, tokenizer) -> GuidedStat
     if params.choice is not None:
         a = _ChoiceAutomaton(list(params.choice))
         return GuidedState(a, a.initial(), tokenizer, eos)
-    if params.json is not None or params.json_object:
+    if params.json is not None:
+        # a schema (str or dict) constrains shape via schema->regex; the
+        # bare json_object flag only demands well-formed JSON
+        from .json_schema import schema_to_regex
+
+        a = RegexAutomaton(schema_to_regex(params.json))
+        return GuidedState(a, a.start, tokenizer, eos)
+    if params.json_object:
         a = JsonPrefixAcceptor()
         return GuidedState(a, a.initial(), tokenizer, eos)
     if params.grammar is not None:
-        raise ValueError("guided grammar is not supported yet")
+        from .grammar import GrammarAutomaton
+
+        a = GrammarAutomaton(params.grammar)
+        return GuidedState(a, a.start, tokenizer, eos)
     raise ValueError("empty structured outputs params")

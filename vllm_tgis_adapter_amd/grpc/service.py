@@ -469,6 +469,12 @@ class TextGenerationService:
         extra = {}
         structured = get_structured_output_params(decoding)
         if structured is not None:
+            from ..engine.guided import validate_structured_outputs
+
+            try:
+                validate_structured_outputs(structured)
+            except ValueError as e:
+                await context.abort(StatusCode.INVALID_ARGUMENT, str(e))
             extra["structured_outputs"] = structured
 
         time_limit_millis = stopping.time_limit_millis
