@@ -84,22 +84,35 @@ class Sampler:
         fused_ok = _ops.native_enabled(logits)
         simple = True
         any_sampling = False
-        for req in requests:
+        ban_rows: list[int] = []  # rows with min_tokens pending: EOS banned
+        ban_eos: list[int] = []
+        for i, req in enumerate(requests):
             p = req.sampling_params
             if (
                 p.repetition_penalty != 1.0
                 or p.logits_processors
                 or req.guided_state is not None
                 or p.logprobs is not None
-                or (p.min_tokens and req.num_output_tokens < p.min_tokens)
             ):
                 simple = False
                 break
+            if (p.min_tokens and req.num_output_tokens < p.min_tokens
+                    and req.eos_token_id is not None):
+                ban_rows.append(i)
+                ban_eos.append(req.eos_token_id)
             if p.temperature != 0.0:
                 any_sampling = True
                 if not (p.top_k <= 0 or p.top_k >= vocab) or p.top_p < 1.0:
                     simple = False
                     break
+        if simple and ban_rows:
+            # min_tokens EOS suppression as ONE batched scatter so the whole
+            # batch stays on the fused path (a per-request python loop here
+            # cost ~11 ms/step at batch 512 — the r2 serving-bench regression)
+            logits[
+                torch.tensor(ban_rows, device=logits.device),
+                torch.tensor(ban_eos, device=logits.device),
+            ] = _NEG_INF
         if simple and fused_ok:
             temps = torch.tensor(
                 [r.sampling_params.temperature for r in requests],
