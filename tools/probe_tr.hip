@@ -5,17 +5,20 @@ __global__ void probe(short* out, short* out2) {
   __shared__ short lds[2048];
   for (int i = threadIdx.x; i < 2048; i += blockDim.x) lds[i] = (short)i;
   __syncthreads();
+  // opaque consumer: keeps the LDS array + its writes alive (the asm below
+  // references LDS only through raw integer offsets)
+  if ((int)out[0] == -32768) out[0] = lds[threadIdx.x];
   // uniform base address experiment
   u32x2 v;
   unsigned addr = 0;  // byte offset into LDS
   asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
-               : "=v"(v) : "v"(addr));
+               : "=v"(v) : "v"(addr) : "memory");
   union { u32x2 u; short s[4]; } c; c.u = v;
   for (int j = 0; j < 4; ++j) out[threadIdx.x * 4 + j] = c.s[j];
   // per-lane address experiment: lane provides addr = lane*8 bytes
   unsigned addr2 = threadIdx.x * 8;
   asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
-               : "=v"(v) : "v"(addr2));
+               : "=v"(v) : "v"(addr2) : "memory");
   c.u = v;
   for (int j = 0; j < 4; ++j) out2[threadIdx.x * 4 + j] = c.s[j];
 }
