@@ -33,6 +33,11 @@ void launch_mfma_probe_32(const unsigned short*, const unsigned short*, float*,
                           hipStream_t);
 void launch_mfma_probe_16(const unsigned short*, const unsigned short*, float*,
                           hipStream_t);
+void launch_paged_prefill_mfma_v2(__hip_bfloat16*, const __hip_bfloat16*,
+                                  const __hip_bfloat16*, const __hip_bfloat16*,
+                                  const int*, const int*, const int*, float,
+                                  int, int, int, int, int, int, int,
+                                  hipStream_t);
 void launch_paged_prefill_mfma(__hip_bfloat16*, const __hip_bfloat16*,
                                const __hip_bfloat16*, const __hip_bfloat16*,
                                const int*, const int*, const int*, float, int,
@@ -225,6 +230,22 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
     const char* e = getenv("VTA_PREFILL_VALU");
     return e && e[0] == '1';
   }();
+  static const bool force_v1 = [] {
+    const char* e = getenv("VTA_PREFILL_V1");
+    return e && e[0] == '1';
+  }();
+  if (!force_valu && !force_v1 &&
+      q.scalar_type() == at::ScalarType::BFloat16 && head_dim == 128) {
+    launch_paged_prefill_mfma_v2(
+        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(k_cache.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(v_cache.data_ptr()),
+        block_tables.data_ptr<int>(), query_start_loc.data_ptr<int>(),
+        seq_lens.data_ptr<int>(), (float)scale, nseq, nheads, kvh, head_dim,
+        block_size, max_blocks, (int)max_query_len, current_stream());
+    return;
+  }
   if (!force_valu && q.scalar_type() == at::ScalarType::BFloat16 &&
       (head_dim == 64 || head_dim == 128)) {
     launch_paged_prefill_mfma(
