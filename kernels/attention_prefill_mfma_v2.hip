@@ -151,8 +151,10 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
     const int buf = (kv_base / KVT2) & 1;
     __syncthreads();  // publish tile (kv_base) to all waves
 
-    const bool wave_active = q_valid ? (kv_base <= wave_max_pos)
-                                     : false;
+    // wave-uniform: MFMAs must run with a full EXEC mask — a per-lane guard
+    // here drops the masked lanes' K fragments from the score MFMA
+    const bool wave_active =
+        (tile_base + wave * 32 < q_len) && (kv_base <= wave_max_pos);
     const char* k_lds = reinterpret_cast<const char*>(k_lds2[buf]);
     const char* v_lds = v_lds2[buf];
 
@@ -203,7 +205,7 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
             const int kv_pos =
                 kv_base + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
             float v = acc_s[kb][r] * sc2;
-            if (kv_pos > q_pos || kv_pos >= seq_len) v = -FLT_MAX;
+            if (!q_valid || kv_pos > q_pos || kv_pos >= seq_len) v = -FLT_MAX;
             s2[kb * 16 + r] = v;
             local_max = fmaxf(local_max, v);
           }
@@ -213,8 +215,9 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
       // defer-max (T13): skip the O/l rescale while the running max holds.
       // The decision is wave-uniform and taken BEFORE any of this tile's P
       // enters O or l, so everything at the old scale rescales exactly once.
-      const bool need_rescale =
-          !__all(tile_max <= m2_state + DEFER_MAX_THR) || m2_state == -FLT_MAX;
+      // wave-uniform (the rescale block shuffles across lanes)
+      const bool need_rescale = !__all(
+          m2_state != -FLT_MAX && tile_max <= m2_state + DEFER_MAX_THR);
       float m2_new = m2_state;
       if (need_rescale) {
         m2_new = fmaxf(m2_state, tile_max);

@@ -329,102 +329,29 @@ class TextGenerationService:
         generated_token_count: int, max_is_token_limit: bool, tokenizer,
         time_limit_reached: bool = False,
     ):
-        stop_reason, stop_sequence = self._convert_reason(
-            output, max_is_token_limit=max_is_token_limit,
-            time_limit_reached=time_limit_reached, tokenizer=tokenizer,
+        from .convert import append_token_details, resolve_stop
+
+        stop = resolve_stop(
+            output, capped=max_is_token_limit,
+            deadline_hit=time_limit_reached, tokenizer=tokenizer,
         )
         response = proto.GenerationResponse(
             text=output.text,
             generated_token_count=generated_token_count,
-            stop_reason=stop_reason,
-            stop_sequence=stop_sequence or "",
+            stop_reason=stop.reason,
+            stop_sequence=stop.sequence or "",
         )
         if resp_options.generated_tokens:
-            self._convert_tokens(
+            append_token_details(
+                response.tokens,
                 to_list(output.token_ids),
                 output.logprobs,
-                include_logprobs=resp_options.token_logprobs,
-                include_ranks=resp_options.token_ranks,
-                top_n_tokens=resp_options.top_n_tokens,
+                want_logprob=resp_options.token_logprobs,
+                want_rank=resp_options.token_ranks,
+                top_n=resp_options.top_n_tokens,
                 tokenizer=tokenizer,
-                token_infos=response.tokens,
             )
         return response
-
-    @staticmethod
-    def _convert_reason(
-        output: "CompletionOutput", *, max_is_token_limit: bool,
-        time_limit_reached: bool, tokenizer,
-    ):
-        finish_reason = output.finish_reason
-        stop_sequence = None
-        SR = proto.StopReasonValue
-        if finish_reason is None:
-            stop_reason = SR.TIME_LIMIT if time_limit_reached else SR.NOT_FINISHED
-        elif finish_reason == "length":
-            stop_reason = SR.TOKEN_LIMIT if max_is_token_limit else SR.MAX_TOKENS
-        elif finish_reason == "stop":
-            stop_reason = SR.STOP_SEQUENCE
-            stop_str_or_tok = output.stop_reason
-            if stop_str_or_tok is None:
-                stop_reason = SR.EOS_TOKEN
-                stop_sequence = getattr(tokenizer, "eos_token", None)
-            elif isinstance(stop_str_or_tok, int):
-                stop_reason = SR.EOS_TOKEN
-                stop_sequence = tokenizer.convert_ids_to_tokens(stop_str_or_tok)
-            elif isinstance(stop_str_or_tok, str):
-                stop_sequence = stop_str_or_tok
-            else:
-                logger.warning("Unexpected stop_reason type: %s", type(stop_str_or_tok))
-        elif finish_reason == "abort":
-            stop_reason = SR.TIME_LIMIT if time_limit_reached else SR.CANCELLED
-        else:
-            logger.warning("Unrecognized finish_reason: %s", finish_reason)
-            stop_reason = SR.CANCELLED
-        return stop_reason, stop_sequence
-
-    @staticmethod
-    def _convert_tokens(
-        token_ids: list[int],
-        logprobs_list,
-        *,
-        include_logprobs: bool,
-        include_ranks: bool,
-        top_n_tokens: int,
-        tokenizer,
-        token_infos,
-        token_start_offset: int = 0,
-    ) -> None:
-        if token_start_offset:
-            token_ids = token_ids[token_start_offset:]
-            if logprobs_list is not None:
-                logprobs_list = logprobs_list[token_start_offset:]
-        token_texts = tokenizer.convert_ids_to_tokens(token_ids)
-        for i, text in enumerate(token_texts):
-            token_info = proto.TokenInfo(text=text)
-            logprobs: "PosLogprobs | None" = logprobs_list[i] if logprobs_list else None
-            if logprobs is None:
-                token_infos.append(token_info)
-                continue
-            if include_logprobs or include_ranks:
-                lp = logprobs[token_ids[i]]
-                if include_logprobs:
-                    token_info.logprob = lp.logprob
-                if include_ranks:
-                    # rank -1 means "dummy" (spec-decode with logprobs off)
-                    rank = lp.rank if lp.rank is not None else 0
-                    token_info.rank = max(0, rank)
-            if top_n_tokens:
-                items = sorted(
-                    logprobs.items(), key=lambda kv: kv[1].logprob, reverse=True
-                )[:top_n_tokens]
-                tt_texts = tokenizer.convert_ids_to_tokens([tid for tid, _ in items])
-                for tt_text, (_, lp) in zip(tt_texts, items):
-                    tt = token_info.top_tokens.add()
-                    tt.text = tt_text
-                    if include_logprobs:
-                        tt.logprob = lp.logprob
-            token_infos.append(token_info)
 
     # ------------------------------------------------------------------
     async def _validate_and_convert_params(self, params, tokenizer, context):
