@@ -99,6 +99,61 @@ def append_token_details(
         dest.append(info)
 
 
+def requested_logprob_count(resp_options, greedy: bool) -> Optional[int]:
+    """How many logprobs/position the engine must track for this request.
+
+    The count covers the requested top-N alternatives plus room for the
+    sampled token itself; under greedy decoding with logprobs the sampled
+    token IS the top-1, so one slot overlaps.  None = no logprobs needed.
+    """
+    base = 1 if (resp_options.token_logprobs or resp_options.token_ranks) else 0
+    n = resp_options.top_n_tokens
+    if n:
+        base += n
+        if greedy and resp_options.token_logprobs:
+            base -= 1
+    return base or None
+
+
+def tokenize_with_caps(
+    prompt: str,
+    sampling_params,
+    *,
+    tokenizer,
+    add_special_tokens: bool,
+    truncate_to: Optional[int],
+    max_model_len: int,
+    default_max_new: int,
+    validate_input_fn,
+) -> tuple[list[int], bool]:
+    """Encode the prompt and settle the request's max_tokens budget.
+
+    Returns (input token ids, capped) where ``capped`` records that the
+    effective max_tokens came from the model context limit rather than the
+    client — downstream a length-finish then reports TOKEN_LIMIT instead
+    of MAX_TOKENS (reference behavior grpc_server.py:787-798).
+    Raises ValueError for over-long input (TGIS error strings).
+    """
+    enc_kwargs: dict = {"add_special_tokens": add_special_tokens}
+    if truncate_to is not None:
+        enc_kwargs["truncation"] = True
+        enc_kwargs["max_length"] = truncate_to
+    ids = tokenizer(prompt, **enc_kwargs).input_ids
+    n_in = len(ids)
+
+    validate_input_fn(sampling_params, n_in, max_model_len)
+
+    room = max_model_len - n_in
+    requested = sampling_params.max_tokens
+    if requested is None:
+        sampling_params.max_tokens = min(default_max_new, room)
+        return ids, True
+    if requested > room:
+        sampling_params.max_tokens = room
+        return ids, True
+    return ids, False
+
+
 def _fill_position(info, token_id, lp_map, want_logprob, want_rank, top_n,
                    tokenizer) -> None:
     entry = lp_map.get(token_id)

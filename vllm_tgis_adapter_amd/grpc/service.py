@@ -307,14 +307,16 @@ class TextGenerationService:
         if result.prompt_token_ids:
             response.input_token_count = len(result.prompt_token_ids)
             if resp_options.input_tokens:
-                self._convert_tokens(
+                from .convert import append_token_details
+
+                append_token_details(
+                    response.input_tokens,
                     result.prompt_token_ids,
                     result.prompt_logprobs,
-                    include_logprobs=resp_options.token_logprobs,
-                    include_ranks=resp_options.token_ranks,
-                    top_n_tokens=resp_options.top_n_tokens,
+                    want_logprob=resp_options.token_logprobs,
+                    want_rank=resp_options.token_ranks,
+                    top_n=resp_options.top_n_tokens,
                     tokenizer=tokenizer,
-                    token_infos=response.input_tokens,
                 )
         if resp_options.input_text and result.prompt:
             response.text = (
@@ -356,44 +358,58 @@ class TextGenerationService:
     # ------------------------------------------------------------------
     async def _validate_and_convert_params(self, params, tokenizer, context):
         """proto Parameters -> engine SamplingParams (+ deadline)."""
+        from .convert import requested_logprob_count
+
         try:
             validate_params(params, self.max_max_new_tokens)
         except ValueError as e:
             await context.abort(StatusCode.INVALID_ARGUMENT, str(e))
 
-        resp_options = params.response
-        sampling = params.sampling
+        greedy = params.method == proto.GREEDY
         stopping = params.stopping
         decoding = params.decoding
-        greedy = params.method == proto.GREEDY
 
-        max_new_tokens = stopping.max_new_tokens if stopping.max_new_tokens > 0 else None
-        min_new_tokens = max(0, stopping.min_new_tokens)
+        kwargs: dict = {
+            "logprobs": requested_logprob_count(params.response, greedy),
+            "max_tokens": stopping.max_new_tokens or None,
+            "min_tokens": max(0, stopping.min_new_tokens),
+            "repetition_penalty": decoding.repetition_penalty or 1.0,
+            "stop": list(stopping.stop_sequences) or None,
+            "skip_special_tokens": self.skip_special_tokens,
+            "include_stop_str_in_output": (
+                stopping.include_stop_sequence
+                if stopping.HasField("include_stop_sequence")
+                else self.default_include_stop_seqs
+            ),
+        }
+        if params.response.input_tokens and not self.disable_prompt_logprobs:
+            kwargs["prompt_logprobs"] = kwargs["logprobs"]
 
-        logprobs = 1 if (resp_options.token_logprobs or resp_options.token_ranks) else 0
-        top_n_tokens = resp_options.top_n_tokens
-        if top_n_tokens:
-            # top_n plus the sampled token; greedy+logprobs overlaps by one
-            logprobs += top_n_tokens
-            if greedy and resp_options.token_logprobs:
-                logprobs -= 1
-        logprobs = with_default(logprobs, None)
+        # sampling method: greedy pins temperature 0; sampling carries the
+        # client's temperature/top-k/top-p/seed through unchanged
+        sampling = params.sampling
+        temp = sampling.temperature if sampling.HasField("temperature") else 1.0
+        if greedy or temp == 0.0:
+            kwargs["temperature"] = 0.0
+        else:
+            kwargs["temperature"] = temp
+            kwargs["top_k"] = sampling.top_k or -1
+            kwargs["top_p"] = sampling.top_p or 1.0
+            if sampling.HasField("seed"):
+                kwargs["seed"] = sampling.seed
 
-        logits_processors = []
+        # per-request logits processors (host-side hooks, E9)
+        procs = []
         if not greedy and 0.0 < sampling.typical_p < 1.0:
-            logits_processors.append(TypicalLogitsWarperWrapper(mass=sampling.typical_p))
+            procs.append(TypicalLogitsWarperWrapper(mass=sampling.typical_p))
         if decoding.HasField("length_penalty"):
-            logits_processors.append(
-                ExpDecayLengthPenaltyWarper(
-                    length_penalty=(
-                        decoding.length_penalty.start_index,
-                        decoding.length_penalty.decay_factor,
-                    ),
-                    eos_token_id=tokenizer.eos_token_id,
-                )
-            )
+            lp = decoding.length_penalty
+            procs.append(ExpDecayLengthPenaltyWarper(
+                length_penalty=(lp.start_index, lp.decay_factor),
+                eos_token_id=tokenizer.eos_token_id,
+            ))
+        kwargs["logits_processors"] = procs
 
-        extra = {}
         structured = get_structured_output_params(decoding)
         if structured is not None:
             from ..engine.guided import validate_structured_outputs
@@ -402,41 +418,14 @@ class TextGenerationService:
                 validate_structured_outputs(structured)
             except ValueError as e:
                 await context.abort(StatusCode.INVALID_ARGUMENT, str(e))
-            extra["structured_outputs"] = structured
+            kwargs["structured_outputs"] = structured
 
-        time_limit_millis = stopping.time_limit_millis
-        deadline = (
-            time.time() + time_limit_millis / 1000.0 if time_limit_millis > 0 else None
-        )
-
-        temperature = sampling.temperature if sampling.HasField("temperature") else 1.0
-        if greedy or temperature == 0.0:
-            extra.update({"temperature": 0.0})
-        else:
-            extra.update({
-                "temperature": temperature,
-                "top_k": with_default(sampling.top_k, -1),
-                "top_p": with_default(sampling.top_p, 1.0),
-                "seed": sampling.seed if sampling.HasField("seed") else None,
-            })
+        deadline = None
+        if stopping.time_limit_millis > 0:
+            deadline = time.time() + stopping.time_limit_millis / 1000.0
 
         try:
-            sampling_params = SamplingParams(
-                logprobs=logprobs,
-                prompt_logprobs=logprobs
-                if not self.disable_prompt_logprobs and resp_options.input_tokens
-                else None,
-                max_tokens=max_new_tokens,
-                min_tokens=min_new_tokens,
-                repetition_penalty=with_default(decoding.repetition_penalty, 1.0),
-                logits_processors=logits_processors,
-                stop=with_default(list(stopping.stop_sequences), None),
-                include_stop_str_in_output=stopping.include_stop_sequence
-                if stopping.HasField("include_stop_sequence")
-                else self.default_include_stop_seqs,
-                skip_special_tokens=self.skip_special_tokens,
-                **extra,
-            )
+            sampling_params = SamplingParams(**kwargs)
         except ValueError as e:
             await context.abort(StatusCode.INVALID_ARGUMENT, str(e))
         return sampling_params, deadline
@@ -454,32 +443,22 @@ class TextGenerationService:
     async def _validate_prompt_and_tokenize(
         self, sampling_params, truncate_input_tokens, prompt, tokenizer, context
     ) -> tuple[list[int], bool]:
+        from .convert import tokenize_with_caps
+
         assert self.config is not None
-        max_model_len = self.config.max_model_len
-
-        tokenizer_kwargs = {"add_special_tokens": ADD_SPECIAL_TOKENS}
-        if truncate_input_tokens is not None:
-            tokenizer_kwargs.update({"truncation": True, "max_length": truncate_input_tokens})
-        input_ids = tokenizer(prompt, **tokenizer_kwargs).input_ids
-        token_num = len(input_ids)
-
         try:
-            validate_input(sampling_params, token_num, max_model_len)
+            return tokenize_with_caps(
+                prompt,
+                sampling_params,
+                tokenizer=tokenizer,
+                add_special_tokens=ADD_SPECIAL_TOKENS,
+                truncate_to=truncate_input_tokens,
+                max_model_len=self.config.max_model_len,
+                default_max_new=self.max_max_new_tokens,
+                validate_input_fn=validate_input,
+            )
         except ValueError as e:
             await context.abort(StatusCode.INVALID_ARGUMENT, str(e))
-
-        max_new_tokens = sampling_params.max_tokens
-        max_is_token_limit = False
-        if max_new_tokens is None:
-            # cap at what paged attention can hold (TGIS default-cap semantics)
-            sampling_params.max_tokens = min(
-                self.max_max_new_tokens, max_model_len - token_num
-            )
-            max_is_token_limit = True
-        elif token_num + max_new_tokens > max_model_len:
-            sampling_params.max_tokens = max_model_len - token_num
-            max_is_token_limit = True
-        return input_ids, max_is_token_limit
 
     # ------------------------------------------------------------------
     @log_rpc_handler_errors
