@@ -304,7 +304,9 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
   }
 
   // ---- epilogue: normalize and store -----------------------------------
-  if (!q_valid) return;
+  // NOTE: every lane stores here — in the PV output the lane indexes a d
+  // column and carries ALL 32 q rows for it; q-row validity is per row
+  // (q_local check below), not per lane.
   const float l_inv_own = l_state > 0.f ? 1.f / l_state : 0.f;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
