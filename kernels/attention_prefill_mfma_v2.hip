@@ -35,6 +35,22 @@ typedef __attribute__((address_space(3))) bf16x4_b* lds_tr_ptr;
 #define LOG2E 1.44269504f
 #define DEFER_MAX_THR 11.5f  // log2-domain threshold (= 8 nats, T13)
 
+// cross-half (lane <-> lane+32) exchange via permlane32_swap: r[0]/r[1] are
+// {partner, own} in one order or the other per half, so max/sum over both
+// covers the pair without an LDS round trip (ds_bpermute) on the critical
+// softmax chain.
+__device__ inline float cross_half_max(float v) {
+  unsigned u = __builtin_bit_cast(unsigned, v);
+  auto r = __builtin_amdgcn_permlane32_swap(u, u, false, false);
+  return fmaxf(__builtin_bit_cast(float, r[0]), __builtin_bit_cast(float, r[1]));
+}
+
+__device__ inline float cross_half_sum(float v) {
+  unsigned u = __builtin_bit_cast(unsigned, v);
+  auto r = __builtin_amdgcn_permlane32_swap(u, u, false, false);
+  return __builtin_bit_cast(float, r[0]) + __builtin_bit_cast(float, r[1]);
+}
+
 template <int HEAD_DIM>
 __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
     __hip_bfloat16* __restrict__ out,            // [total_q, nheads, HD]
@@ -210,7 +226,7 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
             local_max = fmaxf(local_max, v);
           }
       }
-      const float tile_max = fmaxf(local_max, __shfl_xor(local_max, 32, 64));
+      const float tile_max = cross_half_max(local_max);
 
       // defer-max (T13): skip the O/l rescale while the running max holds.
       // The decision is wave-uniform and taken BEFORE any of this tile's P
@@ -238,12 +254,23 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
 
       float p[32];
       float local_sum = 0.f;
+      if (full_tile) {
 #pragma unroll
-      for (int i = 0; i < 32; ++i) {
-        p[i] = s2[i] == -FLT_MAX ? 0.f : exp2f(s2[i] - m2_new);
-        local_sum += p[i];
+        for (int i = 0; i < 32; ++i) {
+          p[i] = __builtin_amdgcn_exp2f(s2[i] - m2_new);
+          local_sum += p[i];
+        }
+      } else {
+        // masked rows carry -FLT_MAX: raw v_exp of (-inf - finite) is 0, so
+        // no select is needed; only guard the NaN case -inf - -inf
+        const float m_safe = m2_new == -FLT_MAX ? 0.f : m2_new;
+#pragma unroll
+        for (int i = 0; i < 32; ++i) {
+          p[i] = __builtin_amdgcn_exp2f(s2[i] - m_safe);
+          local_sum += p[i];
+        }
       }
-      l_state += local_sum + __shfl_xor(local_sum, 32, 64);
+      l_state += cross_half_sum(local_sum);
 
       // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap, T12) -------
       bf16x8_t pa[4];
