@@ -40,14 +40,19 @@ typedef __attribute__((address_space(3))) bf16x4_b* lds_tr_ptr;
 // covers the pair without an LDS round trip (ds_bpermute) on the critical
 // softmax chain.
 __device__ inline float cross_half_max(float v) {
-  unsigned u = __builtin_bit_cast(unsigned, v);
-  auto r = __builtin_amdgcn_permlane32_swap(u, u, false, false);
+  unsigned a = __builtin_bit_cast(unsigned, v);
+  unsigned b = a;
+  asm volatile("" : "+v"(b));  // distinct register: swap(u,u) on one
+                               // register yields partner-only in both outputs
+  auto r = __builtin_amdgcn_permlane32_swap(a, b, false, false);
   return fmaxf(__builtin_bit_cast(float, r[0]), __builtin_bit_cast(float, r[1]));
 }
 
 __device__ inline float cross_half_sum(float v) {
-  unsigned u = __builtin_bit_cast(unsigned, v);
-  auto r = __builtin_amdgcn_permlane32_swap(u, u, false, false);
+  unsigned a = __builtin_bit_cast(unsigned, v);
+  unsigned b = a;
+  asm volatile("" : "+v"(b));
+  auto r = __builtin_amdgcn_permlane32_swap(a, b, false, false);
   return __builtin_bit_cast(float, r[0]) + __builtin_bit_cast(float, r[1]);
 }
 
