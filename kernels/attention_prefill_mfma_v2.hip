@@ -202,22 +202,20 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
     if (next_base < kv_limit) stage_load(next_base);
 
     if (wave_active) {
-      // ---- online softmax in the log2 domain ---------------------------
+      // ---- online softmax; the running max m2_state lives in the RAW
+      // score domain (sc2 > 0 commutes with max), so the P pass is one
+      // fma + exp2 per element: p = exp2(acc*sc2 - m*sc2).
       // fast path: tile strictly below every row's causal diagonal
       const bool full_tile = (kv_base + KVT2 - 1 <= wave_q_pos_min) &&
                              (kv_base + KVT2 <= seq_len) &&
                              (tile_base + wave * 32 + 31 < q_len);
-      float s2[32];
       float local_max = -FLT_MAX;
       if (full_tile) {
 #pragma unroll
         for (int kb = 0; kb < 2; ++kb)
 #pragma unroll
-          for (int r = 0; r < 16; ++r) {
-            const float v = acc_s[kb][r] * sc2;
-            s2[kb * 16 + r] = v;
-            local_max = fmaxf(local_max, v);
-          }
+          for (int r = 0; r < 16; ++r)
+            local_max = fmaxf(local_max, acc_s[kb][r]);
       } else {
 #pragma unroll
         for (int kb = 0; kb < 2; ++kb)
@@ -225,9 +223,9 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
           for (int r = 0; r < 16; ++r) {
             const int kv_pos =
                 kv_base + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
-            float v = acc_s[kb][r] * sc2;
+            float v = acc_s[kb][r];
             if (!q_valid || kv_pos > q_pos || kv_pos >= seq_len) v = -FLT_MAX;
-            s2[kb * 16 + r] = v;
+            acc_s[kb][r] = v;
             local_max = fmaxf(local_max, v);
           }
       }
@@ -236,13 +234,13 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
       // defer-max (T13): skip the O/l rescale while the running max holds.
       // The decision is wave-uniform and taken BEFORE any of this tile's P
       // enters O or l, so everything at the old scale rescales exactly once.
-      // wave-uniform (the rescale block shuffles across lanes)
+      const float thr = DEFER_MAX_THR / sc2;
       const bool need_rescale = !__all(
-          m2_state != -FLT_MAX && tile_max <= m2_state + DEFER_MAX_THR);
+          m2_state != -FLT_MAX && tile_max <= m2_state + thr);
       float m2_new = m2_state;
       if (need_rescale) {
         m2_new = fmaxf(m2_state, tile_max);
-        const float rescale = exp2f(m2_state - m2_new);
+        const float rescale = exp2f((m2_state - m2_new) * sc2);
         l_state *= rescale;
         float f_reg[16];
 #pragma unroll
@@ -257,24 +255,20 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
         m2_state = m2_new;
       }
 
+      // p = exp2(fma(acc, sc2, bias)); masked rows hold -FLT_MAX whose fma
+      // lands at -inf -> exp2 gives 0, no select needed
+      const float bias = (m2_new == -FLT_MAX ? 0.f : -m2_new) * sc2;
       float p[32];
       float local_sum = 0.f;
-      if (full_tile) {
 #pragma unroll
-        for (int i = 0; i < 32; ++i) {
-          p[i] = __builtin_amdgcn_exp2f(s2[i] - m2_new);
-          local_sum += p[i];
-        }
-      } else {
-        // masked rows carry -FLT_MAX: raw v_exp of (-inf - finite) is 0, so
-        // no select is needed; only guard the NaN case -inf - -inf
-        const float m_safe = m2_new == -FLT_MAX ? 0.f : m2_new;
+      for (int kb = 0; kb < 2; ++kb)
 #pragma unroll
-        for (int i = 0; i < 32; ++i) {
-          p[i] = __builtin_amdgcn_exp2f(s2[i] - m_safe);
-          local_sum += p[i];
+        for (int r = 0; r < 16; ++r) {
+          const float v = __builtin_amdgcn_exp2f(
+              __builtin_fmaf(acc_s[kb][r], sc2, bias));
+          p[kb * 16 + r] = v;
+          local_sum += v;
         }
-      }
       l_state += cross_half_sum(local_sum);
 
       // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap, T12) -------
