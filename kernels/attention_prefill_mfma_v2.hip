@@ -35,25 +35,15 @@ typedef __attribute__((address_space(3))) bf16x4_b* lds_tr_ptr;
 #define LOG2E 1.44269504f
 #define DEFER_MAX_THR 11.5f  // log2-domain threshold (= 8 nats, T13)
 
-// cross-half (lane <-> lane+32) exchange via permlane32_swap: r[0]/r[1] are
-// {partner, own} in one order or the other per half, so max/sum over both
-// covers the pair without an LDS round trip (ds_bpermute) on the critical
-// softmax chain.
+// cross-half (lane <-> lane+32) reductions.  permlane32_swap is a LANE-LOCAL
+// register-half exchange (tools/probe_swap.hip), not a cross-lane shuffle,
+// so the partner value must come from a real shuffle.
 __device__ inline float cross_half_max(float v) {
-  unsigned a = __builtin_bit_cast(unsigned, v);
-  unsigned b = a;
-  asm volatile("" : "+v"(b));  // distinct register: swap(u,u) on one
-                               // register yields partner-only in both outputs
-  auto r = __builtin_amdgcn_permlane32_swap(a, b, false, false);
-  return fmaxf(__builtin_bit_cast(float, r[0]), __builtin_bit_cast(float, r[1]));
+  return fmaxf(v, __shfl_xor(v, 32, 64));
 }
 
 __device__ inline float cross_half_sum(float v) {
-  unsigned a = __builtin_bit_cast(unsigned, v);
-  unsigned b = a;
-  asm volatile("" : "+v"(b));
-  auto r = __builtin_amdgcn_permlane32_swap(a, b, false, false);
-  return __builtin_bit_cast(float, r[0]) + __builtin_bit_cast(float, r[1]);
+  return v + __shfl_xor(v, 32, 64);
 }
 
 template <int HEAD_DIM>

@@ -268,6 +268,24 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   });
 }
 
+void launch_gemm_tile(__hip_bfloat16*, const __hip_bfloat16*,
+                      const __hip_bfloat16*, int, int, int, hipStream_t);
+
+void gemm_tile(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
+  const int M = x.size(0);
+  const int K = x.size(1);
+  const int N = w.size(0);
+  TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16 &&
+              w.scalar_type() == at::ScalarType::BFloat16);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
+  TORCH_CHECK(M >= 1 && N % 128 == 0 && K % 64 == 0);
+  TORCH_CHECK(w.size(1) == K && y.size(0) == M && y.size(1) == N);
+  launch_gemm_tile(reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+                   reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                   reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()), M, N,
+                   K, current_stream());
+}
+
 void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
   const int M = x.size(0);
   const int K = x.size(1);
@@ -414,6 +432,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "paged decode attention (CDNA4)");
   m.def("paged_attention_prefill", &paged_attention_prefill,
         "paged causal prefill attention (CDNA4)");
+  m.def("gemm_tile", &gemm_tile,
+        "128x128-tile glds-staged bf16 GEMM for decode batches (CDNA4)");
   m.def("gemm_skinny", &gemm_skinny,
         "skinny decode GEMM y = x @ w^T, M <= 64 (CDNA4 MFMA)");
   m.def("gemm_skinny_gated", &gemm_skinny_gated,

@@ -391,3 +391,18 @@ def test_moe_gemm(t, e, h, inter, topk_counts):
         hr = torch.nn.functional.silu(g) * u
         ref = hr @ w2[ee].float().t()
         assert torch.allclose(y[i].float(), ref, atol=0.5, rtol=3e-2), i
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize(("m", "n", "k"), [
+    (128, 256, 128), (512, 512, 256), (300, 384, 192), (512, 1024, 448),
+])
+def test_gemm_tile(m, n, k):
+    from vllm_tgis_adapter_amd import ops
+
+    torch.manual_seed(11)
+    x = torch.randn(m, k, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(n, k, dtype=torch.bfloat16, device="cuda")
+    got = ops.gemm_tile(x, w)
+    ref = (x.float() @ w.float().t())
+    assert torch.allclose(got.float(), ref, atol=2e-1, rtol=2e-2)

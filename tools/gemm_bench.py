@@ -35,9 +35,11 @@ SHAPES = [
 
 
 def _fn(native):
-    if native:
-        from vllm_tgis_adapter_amd import ops
+    from vllm_tgis_adapter_amd import ops
 
+    if native == "tile":
+        return lambda x, w: ops.gemm_tile(x, w)
+    if native:
         return ops.linear
     return torch.nn.functional.linear
 
@@ -67,21 +69,26 @@ def main():
     p.add_argument("--iters", type=int, default=200)
     p.add_argument("--native", action="store_true",
                    help="route through ops.linear (custom skinny kernel)")
+    p.add_argument("--tile", action="store_true",
+                   help="force the 128x128-tile kernel (gemm_tile)")
     args = p.parse_args()
     assert torch.cuda.is_available()
     results = {}
     for name, n, k in SHAPES:
-        ms = bench_shape(args.m, n, k, args.iters, native=args.native)
+        ms = bench_shape(args.m, n, k, args.iters,
+                         native="tile" if args.tile else args.native)
         bytes_moved = (args.m * k + n * k + args.m * n) * 2
         gbs = bytes_moved / (ms * 1e-3) / 1e9
+        tf = 2 * args.m * n * k / (ms * 1e-3) / 1e12
         results[name] = {
             "ms": round(ms, 4),
             "GB/s": round(gbs, 1),
             "pct_of_8TBs": round(100 * gbs / 8000, 1),
+            "TF": round(tf, 1),
         }
         print(name, results[name], flush=True)
     print(json.dumps({
-        "native": args.native,
+        "native": "tile" if args.tile else args.native,
         "m": args.m,
         "tunableop": os.environ.get("PYTORCH_TUNABLEOP_ENABLED", "0"),
         "results": results,

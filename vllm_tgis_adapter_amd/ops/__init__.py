@@ -183,24 +183,40 @@ def linear(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None = No
     VTA_GEMM_MAX_M raises the cutover for experiments).
     """
     max_m = int(os.environ.get("VTA_GEMM_MAX_M", "64"))
+    tile_max_m = int(os.environ.get("VTA_GEMM_TILE_MAX_M", "0"))
     if (
         bias is None
         and x.dim() == 2
         and _native(x)
         and x.dtype == torch.bfloat16
         and weight.dtype == torch.bfloat16
-        and 1 <= x.shape[0] <= max_m
         and weight.shape[0] % 128 == 0
         and x.shape[1] % 64 == 0
         and x.is_contiguous()
         and weight.is_contiguous()
     ):
-        out = torch.empty(
-            (x.shape[0], weight.shape[0]), dtype=x.dtype, device=x.device
-        )
-        _C.gemm_skinny(out, x, weight)
-        return out
+        m = x.shape[0]
+        if 1 <= m <= max_m:
+            out = torch.empty(
+                (m, weight.shape[0]), dtype=x.dtype, device=x.device
+            )
+            _C.gemm_skinny(out, x, weight)
+            return out
+        if max_m < m <= tile_max_m:
+            out = torch.empty(
+                (m, weight.shape[0]), dtype=x.dtype, device=x.device
+            )
+            _C.gemm_tile(out, x, weight)
+            return out
     return torch.nn.functional.linear(x, weight, bias)
+
+
+def gemm_tile(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """Direct 128x128-tile GEMM path (benchmarks/tests)."""
+    assert _native(x)
+    out = torch.empty((x.shape[0], weight.shape[0]), dtype=x.dtype, device=x.device)
+    _C.gemm_tile(out, x, weight)
+    return out
 
 
 def gated_mlp_up(x: torch.Tensor, w13: torch.Tensor) -> torch.Tensor | None:
