@@ -44,6 +44,8 @@ void launch_paged_prefill_mfma(__hip_bfloat16*, const __hip_bfloat16*,
                                int, int, int, int, int, int, hipStream_t);
 int decode_mfma_num_partitions(int nseq, int kvh, int max_context);
 int gemm_skinny_num_ksplit(int N, int K, int M);
+void launch_gemm_tile(__hip_bfloat16*, const __hip_bfloat16*,
+                      const __hip_bfloat16*, int, int, int, hipStream_t);
 void launch_gemm_skinny(__hip_bfloat16*, float*, const __hip_bfloat16*,
                         const __hip_bfloat16*, int, int, int, int,
                         hipStream_t);
@@ -267,9 +269,6 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
         (int)max_query_len, current_stream());
   });
 }
-
-void launch_gemm_tile(__hip_bfloat16*, const __hip_bfloat16*,
-                      const __hip_bfloat16*, int, int, int, hipStream_t);
 
 void gemm_tile(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
   const int M = x.size(0);

@@ -240,16 +240,56 @@ class Worker:
             )
             if not grouped_ok:
                 return
-        if self.tp > 1 and os.environ.get("VTA_GRAPH_TP", "0") != "1":
+        if self.tp > 1 and os.environ.get("VTA_GRAPH_TP", "1") != "1":
             return
         from .graph_runner import DecodeGraphRunner
 
-        self.graph_runner = DecodeGraphRunner(
-            self.model, self.kv_caches, self.model_config, self.block_size,
-            self.config.scheduler_config.max_num_seqs,
-            self.model_config.max_model_len, self.device,
-        )
-        self.graph_runner.capture()
+        try:
+            self.graph_runner = DecodeGraphRunner(
+                self.model, self.kv_caches, self.model_config, self.block_size,
+                self.config.scheduler_config.max_num_seqs,
+                self.model_config.max_model_len, self.device,
+            )
+            self.graph_runner.capture()
+            if self.tp > 1:
+                # RCCL-inside-graph is the risky capture: verify replay
+                # against an eager forward on identical inputs before trusting
+                # it (all ranks run this concurrently so collectives align)
+                self._validate_graph_replay()
+        except Exception:
+            import traceback
+
+            traceback.print_exc()
+            print("[worker] hipGraph capture failed; decode stays eager",
+                  flush=True)
+            self.graph_runner = None
+
+    @torch.inference_mode()
+    def _validate_graph_replay(self) -> None:
+        gr = self.graph_runner
+        vocab = self.model_config.vocab_size
+        for b in {gr.buckets[0], gr.buckets[-1]}:
+            ids = (torch.arange(b, device=self.device) * 7919 + 13) % vocab
+            pos = torch.zeros(b, dtype=torch.long, device=self.device)
+            slots = torch.full((b,), -1, dtype=torch.long, device=self.device)
+            seq_lens = torch.zeros(b, dtype=torch.int32, device=self.device)
+            bt = torch.zeros((b, gr.max_blocks), dtype=torch.int32,
+                             device=self.device)
+            gr.in_ids[:b].copy_(ids)
+            gr.in_pos[:b].copy_(pos)
+            gr.in_slots[:b].copy_(slots)
+            gr.in_seq_lens[:b].copy_(seq_lens)
+            gr.in_block_tables[:b].zero_()
+            hidden = self.model(gr.in_ids[:b], gr.in_pos[:b], self.kv_caches,
+                                gr._meta(b))
+            eager = self.model.compute_logits(hidden).float()
+            replay = gr.run(ids, pos, slots, seq_lens, bt).float()
+            torch.cuda.synchronize()
+            if not torch.allclose(eager, replay, atol=5e-2, rtol=5e-2):
+                raise RuntimeError(
+                    f"graph replay mismatch at bucket {b}: "
+                    f"max|diff|={float((eager - replay).abs().max())}"
+                )
 
     @torch.inference_mode()
     def _profile_peak_memory(self) -> None:
