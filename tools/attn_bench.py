@@ -61,7 +61,8 @@ def main():
                      "TFLOPs": round(flops / us / 1e6, 1)}
         print(name, out[name], flush=True)
 
-    decode_cases = [("decode 256x576", 256, 576), ("decode 64x512", 64, 512)]
+    decode_cases = [("decode 256x576", 256, 576), ("decode 64x512", 64, 512),
+                ("decode 512x576", 512, 576), ("decode 512x1024", 512, 1024)]
     if args.only == "prefill":
         decode_cases = []
     for name, nseq, ctx in decode_cases:
