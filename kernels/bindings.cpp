@@ -46,6 +46,9 @@ int decode_mfma_num_partitions(int nseq, int kvh, int max_context);
 int gemm_skinny_num_ksplit(int N, int K, int M);
 void launch_gemm_tile(__hip_bfloat16*, const __hip_bfloat16*,
                       const __hip_bfloat16*, int, int, int, hipStream_t);
+void launch_xgmi_allreduce(const unsigned long long*, __hip_bfloat16*,
+                           const __hip_bfloat16*, long, long, int, int,
+                           hipStream_t);
 void launch_gemm_skinny(__hip_bfloat16*, float*, const __hip_bfloat16*,
                         const __hip_bfloat16*, int, int, int, int,
                         hipStream_t);
@@ -270,6 +273,45 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   });
 }
 
+// ---- direct-xGMI all-reduce plumbing (E15) --------------------------------
+// Raw hipMalloc (NOT the torch caching allocator: hipIpcGetMemHandle needs
+// the allocation base).  Returns (device_ptr, ipc_handle_bytes).
+std::tuple<int64_t, py::bytes> xar_alloc(int64_t nbytes) {
+  void* p = nullptr;
+  TORCH_CHECK(hipMalloc(&p, nbytes) == hipSuccess, "xar_alloc failed");
+  TORCH_CHECK(hipMemset(p, 0, nbytes) == hipSuccess);
+  TORCH_CHECK(hipDeviceSynchronize() == hipSuccess);
+  hipIpcMemHandle_t h;
+  TORCH_CHECK(hipIpcGetMemHandle(&h, p) == hipSuccess,
+              "hipIpcGetMemHandle failed");
+  return {reinterpret_cast<int64_t>(p),
+          py::bytes(reinterpret_cast<const char*>(&h), sizeof(h))};
+}
+
+int64_t xar_open(const std::string& handle_bytes) {
+  TORCH_CHECK(handle_bytes.size() == sizeof(hipIpcMemHandle_t));
+  hipIpcMemHandle_t h;
+  memcpy(&h, handle_bytes.data(), sizeof(h));
+  void* p = nullptr;
+  TORCH_CHECK(hipIpcOpenMemHandle(&p, h, hipIpcMemLazyEnablePeerAccess) ==
+                  hipSuccess,
+              "hipIpcOpenMemHandle failed");
+  return reinterpret_cast<int64_t>(p);
+}
+
+void xgmi_allreduce(torch::Tensor tensor, torch::Tensor bufs, int64_t cap,
+                    int64_t rank, int64_t world) {
+  TORCH_CHECK(tensor.scalar_type() == at::ScalarType::BFloat16);
+  TORCH_CHECK(tensor.is_contiguous() && tensor.is_cuda());
+  TORCH_CHECK(bufs.scalar_type() == at::ScalarType::Long && bufs.is_cuda());
+  TORCH_CHECK(tensor.numel() * 2 <= cap, "tensor exceeds staging capacity");
+  launch_xgmi_allreduce(
+      reinterpret_cast<const unsigned long long*>(bufs.data_ptr()),
+      reinterpret_cast<__hip_bfloat16*>(tensor.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(tensor.data_ptr()),
+      tensor.numel(), cap, (int)rank, (int)world, current_stream());
+}
+
 void gemm_tile(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
   const int M = x.size(0);
   const int K = x.size(1);
@@ -431,6 +473,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "paged decode attention (CDNA4)");
   m.def("paged_attention_prefill", &paged_attention_prefill,
         "paged causal prefill attention (CDNA4)");
+  m.def("xar_alloc", &xar_alloc,
+        "raw hipMalloc + IPC handle for the xGMI all-reduce staging buffer");
+  m.def("xar_open", &xar_open, "open a peer's IPC handle -> device pointer");
+  m.def("xgmi_allreduce", &xgmi_allreduce,
+        "in-place one-shot all-reduce over peer-mapped xGMI buffers");
   m.def("gemm_tile", &gemm_tile,
         "128x128-tile glds-staged bf16 GEMM for decode batches (CDNA4)");
   m.def("gemm_skinny", &gemm_skinny,

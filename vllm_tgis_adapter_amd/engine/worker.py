@@ -145,6 +145,11 @@ class Worker:
         self.graph_runner = None  # set by capture_decode_graphs()
         self.loras: dict[int, object] = {}  # lora_int_id -> LoRAAdapter
 
+        if self.tp > 1 and self.device == "cuda":
+            from ..parallel import init_xgmi_allreduce
+
+            init_xgmi_allreduce()  # E15: direct-xGMI AR, RCCL fallback
+
     # ------------------------------------------------------------------
     def add_lora(self, lora_path: str, lora_int_id: int) -> None:
         from .lora import load_lora_adapter
