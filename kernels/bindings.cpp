@@ -49,6 +49,9 @@ void launch_gemm_tile(__hip_bfloat16*, const __hip_bfloat16*,
 void launch_xgmi_allreduce(const unsigned long long*, __hip_bfloat16*,
                            const __hip_bfloat16*, long, long, int, int,
                            hipStream_t);
+void launch_gemm_skinny_q(__hip_bfloat16*, float*, const __hip_bfloat16*,
+                          const unsigned char*, const float*, int, int, int,
+                          int, int, hipStream_t);
 void launch_gemm_skinny(__hip_bfloat16*, float*, const __hip_bfloat16*,
                         const __hip_bfloat16*, int, int, int, int,
                         hipStream_t);
@@ -349,6 +352,34 @@ void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
                      N, K, KS, current_stream());
 }
 
+void gemm_skinny_q(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
+                   torch::Tensor wscale, int64_t qbits) {
+  const int M = x.size(0);
+  const int K = x.size(1);
+  const int N = wq.size(0);
+  TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16);
+  TORCH_CHECK(wq.scalar_type() == at::ScalarType::UInt8 ||
+              wq.scalar_type() == at::ScalarType::Char);
+  TORCH_CHECK(wscale.scalar_type() == at::ScalarType::Float);
+  TORCH_CHECK(x.is_contiguous() && wq.is_contiguous() &&
+              wscale.is_contiguous() && y.is_contiguous());
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 128 == 0 && K % 128 == 0);
+  TORCH_CHECK(qbits == 8 || qbits == 4);
+  TORCH_CHECK(wq.size(1) == (qbits == 8 ? K : K / 2));
+  const int KS = gemm_skinny_num_ksplit(N, K, M);
+  torch::Tensor part;
+  float* pp = nullptr;
+  if (KS > 1) {
+    part = at::empty({KS, M, N}, x.options().dtype(at::ScalarType::Float));
+    pp = part.data_ptr<float>();
+  }
+  launch_gemm_skinny_q(
+      reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), pp,
+      reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+      reinterpret_cast<const unsigned char*>(wq.data_ptr()),
+      wscale.data_ptr<float>(), (int)qbits, M, N, K, KS, current_stream());
+}
+
 void gemm_skinny_gated(torch::Tensor y, torch::Tensor x, torch::Tensor w13) {
   const int M = x.size(0);
   const int K = x.size(1);
@@ -480,6 +511,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "in-place one-shot all-reduce over peer-mapped xGMI buffers");
   m.def("gemm_tile", &gemm_tile,
         "128x128-tile glds-staged bf16 GEMM for decode batches (CDNA4)");
+  m.def("gemm_skinny_q", &gemm_skinny_q,
+        "weight-only-quantized (int8 / int4-g128) skinny decode GEMM");
   m.def("gemm_skinny", &gemm_skinny,
         "skinny decode GEMM y = x @ w^T, M <= 64 (CDNA4 MFMA)");
   m.def("gemm_skinny_gated", &gemm_skinny_gated,

@@ -217,6 +217,9 @@ class EngineConfig:
     seed: int = 0
     # speculative decoding: "ngram" enables prompt-lookup drafts (E17)
     speculative_model: "str | None" = None
+    # weight-only quantization: None | "int8" | "int4" (awq/gptq/squeezellm
+    # map to int4 group-128 RTN — SURVEY.md E18)
+    quantization: "str | None" = None
     speculative_num_tokens: int = 4
     # load deterministic rank-independent synthetic weights (TP equivalence
     # tests) instead of per-shard random init

@@ -138,6 +138,9 @@ class Worker:
 
             self.model.load_weights(synth_llama_weights(self.model_config, config.seed))
 
+        if config.quantization:
+            self._quantize_model(config.quantization)
+
         self.block_size = config.cache_config.block_size
         self.kv_caches: list[tuple[torch.Tensor, torch.Tensor]] = []
         self.num_blocks = 0
@@ -151,6 +154,40 @@ class Worker:
             init_xgmi_allreduce()  # E15: direct-xGMI AR, RCCL fallback
 
     # ------------------------------------------------------------------
+    def _quantize_model(self, method: str) -> None:
+        """Weight-only RTN quantization of the attention/MLP linears (E18).
+
+        --quantize int8 -> per-channel int8; int4 / awq / gptq / squeezellm
+        -> packed int4 with group-128 scales (the 4-bit storage profile of
+        those methods; checkpoint-specific scales are not available offline,
+        so RTN is applied to the loaded weights).  Embedding / lm_head stay
+        bf16 (their accuracy sensitivity; standard practice).
+        """
+        from ..parallel.layers import (
+            ColumnParallelLinear,
+            MergedColumnParallelLinear,
+            RowParallelLinear,
+            quantize_module_,
+        )
+
+        qbits = 8 if method == "int8" else 4
+        kinds = (ColumnParallelLinear, MergedColumnParallelLinear,
+                 RowParallelLinear)
+        count = 0
+        for mod in self.model.modules():
+            if isinstance(mod, kinds) and hasattr(mod, "weight"):
+                # int4 scales group along K (128); int8 has no shape need —
+                # the native-kernel dispatch checks its own constraints and
+                # falls back to dequant+hipBLASLt otherwise
+                if qbits == 4 and mod.weight.shape[1] % 128:
+                    continue
+                quantize_module_(mod, qbits)
+                count += 1
+        if self.device == "cuda":
+            torch.cuda.empty_cache()
+        print(f"[worker] quantized {count} linears to "
+              f"{'int8' if qbits == 8 else 'int4-g128'} ({method})", flush=True)
+
     def add_lora(self, lora_path: str, lora_int_id: int) -> None:
         from .lora import load_lora_adapter
 

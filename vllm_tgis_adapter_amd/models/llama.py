@@ -115,7 +115,7 @@ class LlamaMLP(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         from ..engine.lora import CTX as _lora_ctx
 
-        if not _lora_ctx.active:
+        if not _lora_ctx.active and getattr(self.gate_up, "quant_bits", None) is None:
             h = ops.gated_mlp_up(x, self.gate_up.weight)
             if h is not None:
                 return self.down(h, lora=self._down_lora)

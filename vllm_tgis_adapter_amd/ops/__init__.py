@@ -219,6 +219,69 @@ def gemm_tile(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     return out
 
 
+# ---------------------------------------------------------------------------
+# Weight-only quantization: RTN into int8 (per-out-channel
+# scale) or packed int4 with group-128 scales (the --quantize awq/gptq/
+# squeezellm 4-bit surface; offline RTN since checkpoint-specific scales
+# don't exist in this environment).
+# ---------------------------------------------------------------------------
+
+def quantize_weight(w: torch.Tensor, qbits: int) -> tuple[torch.Tensor, torch.Tensor]:
+    """Returns (wq, scales). int8: wq [N,K] int8, scales [N] f32.
+    int4: wq [N,K/2] uint8 offset-binary nibbles, scales [N,K/128] f32."""
+    assert qbits in (8, 4)
+    wf = w.float()
+    if qbits == 8:
+        scales = wf.abs().amax(dim=1).clamp(min=1e-8) / 127.0
+        q = torch.round(wf / scales.unsqueeze(1)).clamp(-127, 127).to(torch.int8)
+        return q, scales.contiguous()
+    n, k = wf.shape
+    assert k % 128 == 0
+    g = wf.view(n, k // 128, 128)
+    scales = g.abs().amax(dim=2).clamp(min=1e-8) / 7.0
+    q = torch.round(g / scales.unsqueeze(2)).clamp(-8, 7).to(torch.int16) + 8
+    q = q.view(n, k)
+    packed = (q[:, 0::2] | (q[:, 1::2] << 4)).to(torch.uint8)
+    return packed.contiguous(), scales.contiguous()
+
+
+def dequantize_weight(wq: torch.Tensor, scales: torch.Tensor, qbits: int,
+                      dtype: torch.dtype = torch.bfloat16) -> torch.Tensor:
+    if qbits == 8:
+        return (wq.float() * scales.unsqueeze(1)).to(dtype)
+    n = wq.shape[0]
+    lo = (wq & 0xF).to(torch.float32) - 8.0
+    hi = (wq >> 4).to(torch.float32) - 8.0
+    q = torch.stack([lo, hi], dim=2).view(n, -1)  # interleave even/odd k
+    k = q.shape[1]
+    return (q.view(n, k // 128, 128) * scales.unsqueeze(2)).view(n, k).to(dtype)
+
+
+def linear_quant(x: torch.Tensor, wq: torch.Tensor, scales: torch.Tensor,
+                 qbits: int, bias: torch.Tensor | None = None) -> torch.Tensor:
+    """W8A16/W4A16 linear: native dequant-GEMM for decode batches, dequant +
+    hipBLASLt above (prefill is compute-bound; the quant win is the weight
+    stream at small M)."""
+    x2 = x.reshape(-1, x.shape[-1])
+    m = x2.shape[0]
+    n = wq.shape[0]
+    k = x2.shape[1]
+    if (
+        bias is None
+        and _native(x)
+        and x.dtype == torch.bfloat16
+        and 1 <= m <= 64
+        and n % 128 == 0
+        and k % 128 == 0
+        and x2.is_contiguous()
+    ):
+        out = torch.empty((m, n), dtype=x.dtype, device=x.device)
+        _C.gemm_skinny_q(out, x2, wq, scales, qbits)
+        return out.view(*x.shape[:-1], n)
+    w = dequantize_weight(wq, scales, qbits, x.dtype)
+    return torch.nn.functional.linear(x, w, bias)
+
+
 def gated_mlp_up(x: torch.Tensor, w13: torch.Tensor) -> torch.Tensor | None:
     """Fused silu(x@Wg^T) * (x@Wu^T) for merged [gate; up] weights.
 

@@ -16,6 +16,28 @@ from .. import ops
 from . import divide, get_tp_rank, get_tp_world_size, tp_all_gather, tp_all_reduce
 
 
+def quantize_module_(mod: nn.Module, qbits: int) -> None:
+    """Replace a linear's bf16 weight with RTN-quantized storage (E18).
+
+    Works on Column/Merged/Row parallel linears; forward then dispatches to
+    ops.linear_quant (native dequant skinny GEMM at decode batch sizes).
+    """
+    w = mod.weight.data
+    wq, scales = ops.quantize_weight(w, qbits)
+    del mod.weight
+    mod.register_buffer("weight_q", wq)
+    mod.register_buffer("weight_scale", scales)
+    mod.quant_bits = qbits
+
+
+def _linear_fwd(mod: nn.Module, x: torch.Tensor) -> torch.Tensor:
+    qbits = getattr(mod, "quant_bits", None)
+    if qbits is not None:
+        return ops.linear_quant(x, mod.weight_q, mod.weight_scale, qbits,
+                                mod.bias)
+    return ops.linear(x, mod.weight, mod.bias)
+
+
 def _init_weight(shape, dtype, std: float = 0.006) -> nn.Parameter:
     # Random-init path for synthetic benchmarking (BASELINE measures on
     # random-init weights); real checkpoints overwrite via load_weights.
@@ -38,7 +60,7 @@ class ColumnParallelLinear(nn.Module):
         self.bias = _init_weight((self.out_per_rank,), dtype, std=0.0) if bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return ops.linear(x, self.weight, self.bias)
+        return _linear_fwd(self, x)
 
     def load_full_weight(self, w: torch.Tensor) -> None:
         r = get_tp_rank()
@@ -65,7 +87,7 @@ class MergedColumnParallelLinear(nn.Module):
         self.bias = _init_weight((total,), dtype, std=0.0) if bias else None
 
     def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
-        y = ops.linear(x, self.weight, self.bias)
+        y = _linear_fwd(self, x)
         if lora is not None:
             from ..engine.lora import apply_lora
 
@@ -95,7 +117,11 @@ class RowParallelLinear(nn.Module):
         self.bias = _init_weight((out_size,), dtype, std=0.0) if bias else None
 
     def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
-        y = ops.linear(x, self.weight)
+        qbits = getattr(self, "quant_bits", None)
+        if qbits is not None:
+            y = ops.linear_quant(x, self.weight_q, self.weight_scale, qbits)
+        else:
+            y = ops.linear(x, self.weight)
         if lora is not None:
             # delta added BEFORE the all-reduce: A is input-sharded, so the
             # per-rank partial deltas sum to the full LoRA delta

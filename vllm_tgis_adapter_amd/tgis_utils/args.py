@@ -162,8 +162,10 @@ def add_tgis_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
     parser.add_argument("--max-concurrent-requests", type=int)
     parser.add_argument("--dtype-str", type=str, help="deprecated, use dtype")
     parser.add_argument("--quantize", type=str,
-                        choices=["awq", "gptq", "squeezellm", None],
-                        help="weight quantization method")
+                        choices=["awq", "gptq", "squeezellm", "int8", "int4",
+                                 None],
+                        help="weight quantization method (4-bit surfaces "
+                             "apply RTN int4-g128; also accepts int8/int4)")
     parser.add_argument("--num-gpus", type=int)
     parser.add_argument("--num-shard", type=int)
     parser.add_argument("--output-special-tokens", type=_bool_from_string, default=False)
@@ -285,5 +287,6 @@ def engine_config_from_args(args: argparse.Namespace):
         enable_lora=args.enable_lora or bool(args.adapter_cache or args.prefix_store_path),
         max_loras=args.max_loras,
         max_lora_rank=args.max_lora_rank,
+        quantization=args.quantization,
         seed=args.seed,
     )
