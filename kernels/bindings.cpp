@@ -358,7 +358,7 @@ void gemm_skinny_q(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
   const int K = x.size(1);
   const int N = wq.size(0);
   TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16);
-  TORCH_CHECK(wq.scalar_type() == at::ScalarType::UInt8 ||
+  TORCH_CHECK(wq.scalar_type() == at::ScalarType::Byte ||
               wq.scalar_type() == at::ScalarType::Char);
   TORCH_CHECK(wscale.scalar_type() == at::ScalarType::Float);
   TORCH_CHECK(x.is_contiguous() && wq.is_contiguous() &&

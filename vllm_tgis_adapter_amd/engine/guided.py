@@ -456,6 +456,43 @@ def _vocab_strings(tokenizer) -> list[str]:
     return _VOCAB_CACHE[key]
 
 
+class _TrieNode:
+    __slots__ = ("children", "token_ids")
+
+    def __init__(self):
+        self.children: dict[str, "_TrieNode"] = {}
+        self.token_ids: list[int] = []
+
+
+_TRIE_CACHE: dict[int, _TrieNode] = {}
+
+
+def _vocab_trie(tokenizer) -> _TrieNode:
+    """Character trie over the vocab strings (built once per tokenizer).
+
+    The allowed-set walk then costs the size of the ALIVE prefix subtree
+    instead of vocab x token-length character steps — for selective
+    automaton states (the usual case) that is orders of magnitude less
+    work, and shared prefixes are walked once either way.
+    """
+    key = id(tokenizer)
+    root = _TRIE_CACHE.get(key)
+    if root is None:
+        root = _TrieNode()
+        for tid, s in enumerate(_vocab_strings(tokenizer)):
+            if not s:
+                continue
+            node = root
+            for ch in s:
+                nxt = node.children.get(ch)
+                if nxt is None:
+                    nxt = node.children[ch] = _TrieNode()
+                node = nxt
+            node.token_ids.append(tid)
+        _TRIE_CACHE[key] = root
+    return root
+
+
 class GuidedState:
     """Automaton + current state + tokenizer-level filtering."""
 
@@ -479,12 +516,23 @@ class GuidedState:
             cached = None
         if cached is not None:
             return cached
-        allowed = []
-        for tid, s in enumerate(self.vocab):
-            if not s:
-                continue
-            if self.automaton.walk(self.state, s) is not None:
-                allowed.append(tid)
+        # walk the vocab TRIE carrying automaton state: dead prefixes prune
+        # their whole subtree, shared prefixes are stepped once
+        allowed: list[int] = []
+        step = getattr(self.automaton, "step", None)
+        if step is not None:
+            stack = [(_vocab_trie(self.tokenizer), self.state)]
+            while stack:
+                node, st = stack.pop()
+                allowed.extend(node.token_ids)
+                for ch, child in node.children.items():
+                    nst = step(st, ch)
+                    if nst is not None:
+                        stack.append((child, nst))
+        else:  # choice automaton: prefix matching, no char-step API
+            for tid, s in enumerate(self.vocab):
+                if s and self.automaton.walk(self.state, s) is not None:
+                    allowed.append(tid)
         if self.automaton.is_accepting(self.state):
             allowed.append(self.eos_token_id)
         if not allowed:
@@ -541,10 +589,30 @@ def validate_structured_outputs(params: StructuredOutputsParams) -> None:
         GrammarAutomaton(params.grammar)
 
 
+# Compiled automatons (and their per-state allowed caches) are shared across
+# requests using the same pattern — serving workloads repeat schemas/regexes,
+# so steady state pays one dict lookup per step instead of a recompile and a
+# cold vocab walk per request.  Bounded FIFO (compiled automatons are small;
+# the allowed caches dominate).
+_AUTOMATON_CACHE: dict = {}
+_AUTOMATON_CACHE_MAX = 128
+
+
+def _cached_automaton(key, build):
+    hit = _AUTOMATON_CACHE.get(key)
+    if hit is None:
+        hit = build()
+        if len(_AUTOMATON_CACHE) >= _AUTOMATON_CACHE_MAX:
+            _AUTOMATON_CACHE.pop(next(iter(_AUTOMATON_CACHE)))
+        _AUTOMATON_CACHE[key] = hit
+    return hit
+
+
 def build_guided_state(params: StructuredOutputsParams, tokenizer) -> GuidedState:
     eos = tokenizer.eos_token_id
     if params.regex is not None:
-        a = RegexAutomaton(params.regex)
+        a = _cached_automaton(("re", params.regex),
+                              lambda: RegexAutomaton(params.regex))
         return GuidedState(a, a.start, tokenizer, eos)
     if params.choice is not None:
         a = _ChoiceAutomaton(list(params.choice))
@@ -554,14 +622,18 @@ def build_guided_state(params: StructuredOutputsParams, tokenizer) -> GuidedStat
         # bare json_object flag only demands well-formed JSON
         from .json_schema import schema_to_regex
 
-        a = RegexAutomaton(schema_to_regex(params.json))
+        key = params.json if isinstance(params.json, str) else repr(params.json)
+        a = _cached_automaton(
+            ("schema", key),
+            lambda: RegexAutomaton(schema_to_regex(params.json)))
         return GuidedState(a, a.start, tokenizer, eos)
     if params.json_object:
-        a = JsonPrefixAcceptor()
+        a = _cached_automaton(("json",), JsonPrefixAcceptor)
         return GuidedState(a, a.initial(), tokenizer, eos)
     if params.grammar is not None:
         from .grammar import GrammarAutomaton
 
-        a = GrammarAutomaton(params.grammar)
+        a = _cached_automaton(("ebnf", params.grammar),
+                              lambda: GrammarAutomaton(params.grammar))
         return GuidedState(a, a.start, tokenizer, eos)
     raise ValueError("empty structured outputs params")
