@@ -219,3 +219,64 @@ def test_tp2_engine_matches_tp1():
     res = _run_spawn(_engine_tp_worker, 2, timeout=600)
     tp2 = res[0]
     assert tp2 == ref, (tp2, ref)
+
+
+def _ep_moe_worker(rank, world, port, q):
+    try:
+        _setup_dist(rank, world, port)
+        torch.manual_seed(0)
+        from vllm_tgis_adapter_amd.engine.config import ModelConfig
+        from vllm_tgis_adapter_amd.models.mixtral import MoEBlock
+
+        cfg = ModelConfig.from_model_arg("tiny-mixtral", dtype="float32")
+        cfg.expert_parallel = True
+        moe = MoEBlock(cfg)
+        assert moe.ep, "EP mode should be active at world>1"
+
+        # deterministic full weights, identical on every rank
+        g = torch.Generator().manual_seed(7)
+        E, I, H = cfg.num_experts, cfg.intermediate_size, cfg.hidden_size
+        gate = torch.randn(E, H, generator=g) * 0.2
+        w13 = torch.randn(E, 2 * I, H, generator=g) * 0.2
+        w2 = torch.randn(E, H, I, generator=g) * 0.2
+        x = torch.randn(6, H, generator=g)
+
+        moe.gate.data.copy_(gate)
+        epr = moe.experts_per_rank
+        moe.w13.data.copy_(w13[moe.expert0:moe.expert0 + epr])
+        moe.w2.data.copy_(w2[moe.expert0:moe.expert0 + epr])
+
+        out = moe(x)
+
+        # dense single-rank reference
+        import torch.nn.functional as F
+
+        from vllm_tgis_adapter_amd import ops
+
+        logits = x @ gate.t()
+        weights, ids = ops.topk_softmax(logits, cfg.num_experts_per_tok)
+        ref = torch.zeros_like(x)
+        for t in range(x.shape[0]):
+            for j in range(cfg.num_experts_per_tok):
+                e = int(ids[t, j])
+                h = ops.silu_and_mul(x[t:t + 1] @ w13[e].t())
+                ref[t] += float(weights[t, j]) * (h @ w2[e].t())[0]
+        ok = torch.allclose(out, ref, atol=1e-4, rtol=1e-4)
+        q.put((rank, {"ok": bool(ok),
+                      "maxdiff": float((out - ref).abs().max())}))
+    except Exception:
+        import traceback
+
+        q.put((rank, "ERROR " + traceback.format_exc()))
+    finally:
+        import torch.distributed as dist
+
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_ep_moe_matches_dense():
+    """Expert-parallel MoE (all-to-all dispatch) == dense reference (E16)."""
+    results = _run_spawn(_ep_moe_worker, 2)
+    for rank, payload in results.items():
+        assert payload["ok"], (rank, payload)

@@ -96,6 +96,9 @@ class ModelConfig:
     # MoE
     num_experts: int = 0
     num_experts_per_tok: int = 0
+    # MoE layout: False = TP-sharded experts (default; graph-capturable),
+    # True = expert-parallel with token all-to-all
+    expert_parallel: bool = False
     # Where to load weights from; None => random init (synthetic bench mode)
     weights_path: Optional[str] = None
     # HF config passthrough for tokenizer etc.

@@ -268,6 +268,8 @@ class Worker:
         if os.environ.get("VTA_FORCE_REFERENCE", "0") == "1":
             return  # torch-reference ops sync to host; not capturable
         mc = self.model_config
+        if getattr(mc, "expert_parallel", False) and self.tp > 1:
+            return  # EP all-to-all host-syncs on split sizes: eager only
         if getattr(mc, "num_experts", 0):
             # the grouped-GEMM MoE path (argsort/bincount/index_add — all
             # shape-static) is capturable; the per-expert fallback loop uses

@@ -31,11 +31,76 @@ class MoEBlock(nn.Module):
         self.num_experts = cfg.num_experts
         self.top_k = cfg.num_experts_per_tok
         self.hidden = cfg.hidden_size
-        self.inter = divide(cfg.intermediate_size, tp)
+        # expert-parallel (E16): whole experts distributed across ranks with
+        # token all-to-all; otherwise every rank holds an intermediate slice
+        # of every expert (TP-sharded, graph-capturable)
+        self.ep = bool(cfg.expert_parallel) and tp > 1 and cfg.num_experts % tp == 0
+        if self.ep:
+            from ..parallel import get_tp_rank
+
+            self.tp = tp
+            self.experts_per_rank = cfg.num_experts // tp
+            self.expert0 = get_tp_rank() * self.experts_per_rank
+            self.inter = cfg.intermediate_size  # full intermediate
+            n_local = self.experts_per_rank
+        else:
+            self.inter = divide(cfg.intermediate_size, tp)
+            n_local = cfg.num_experts
         self.gate = _init_weight((cfg.num_experts, cfg.hidden_size), cfg.dtype, std=0.02)
-        # w13: [E, 2*I/tp, H] fused gate+up; w2: [E, H, I/tp]
-        self.w13 = _init_weight((cfg.num_experts, 2 * self.inter, cfg.hidden_size), cfg.dtype)
-        self.w2 = _init_weight((cfg.num_experts, cfg.hidden_size, self.inter), cfg.dtype)
+        # w13: [E_local, 2*I_local, H] fused gate+up; w2: [E_local, H, I_local]
+        self.w13 = _init_weight((n_local, 2 * self.inter, cfg.hidden_size), cfg.dtype)
+        self.w2 = _init_weight((n_local, cfg.hidden_size, self.inter), cfg.dtype)
+
+    def _forward_ep(self, x: torch.Tensor, flat_ids, flat_w, token_idx):
+        """Dispatch token copies to their experts' owner ranks (all-to-all),
+        compute locally at full intermediate width, return and combine.
+        No output all-reduce: each token's k contributions come home and sum
+        locally.  Variable split sizes host-sync, so EP runs eager."""
+        import torch.distributed as dist
+
+        epr = self.experts_per_rank
+        owner = torch.div(flat_ids, epr, rounding_mode="floor")
+        send_order = torch.argsort(owner, stable=True)
+        send_counts = torch.bincount(owner, minlength=self.tp)
+        recv_counts = torch.empty_like(send_counts)
+        dist.all_to_all_single(recv_counts, send_counts)
+        send_sizes = send_counts.tolist()
+        recv_sizes = recv_counts.tolist()
+
+        x_send = x[token_idx[send_order]].contiguous()
+        ids_send = flat_ids[send_order].contiguous()
+        x_recv = x.new_empty((sum(recv_sizes), self.hidden))
+        ids_recv = ids_send.new_empty(sum(recv_sizes))
+        dist.all_to_all_single(x_recv, x_send, recv_sizes, send_sizes)
+        dist.all_to_all_single(ids_recv, ids_send, recv_sizes, send_sizes)
+
+        local_e = ids_recv - self.expert0
+        order2 = torch.argsort(local_e, stable=True)
+        xs = x_recv[order2].contiguous()
+        if ops.moe_gemm_usable(xs, self.hidden, self.inter):
+            counts = torch.zeros(epr, dtype=torch.int32, device=x.device)
+            counts.scatter_add_(
+                0, local_e, torch.ones_like(local_e, dtype=torch.int32))
+            seg = torch.zeros(epr + 1, dtype=torch.int32, device=x.device)
+            seg[1:] = counts.cumsum(0).to(torch.int32)
+            h = ops.moe_gemm(xs, self.w13, seg, gated=True)
+            ye_sorted = ops.moe_gemm(h, self.w2, seg, gated=False)
+        else:
+            ye_sorted = torch.empty_like(xs)
+            sorted_e = local_e[order2]
+            for e in range(epr):
+                sel = (sorted_e == e).nonzero(as_tuple=True)[0]
+                h = ops.silu_and_mul(F.linear(xs[sel], self.w13[e]))
+                ye_sorted[sel] = F.linear(h, self.w2[e])
+        ye = torch.empty_like(ye_sorted)
+        ye[order2] = ye_sorted
+
+        y_home = x.new_empty((sum(send_sizes), self.hidden))
+        dist.all_to_all_single(y_home, ye.contiguous(), send_sizes, recv_sizes)
+        out = torch.zeros_like(x)
+        out.index_add_(0, token_idx[send_order],
+                       y_home * flat_w[send_order, None])
+        return out
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:  # [T, H]
         t = x.shape[0]
@@ -45,6 +110,8 @@ class MoEBlock(nn.Module):
         flat_ids = ids.reshape(-1)                        # [T*k]
         flat_w = weights.reshape(-1).to(x.dtype)
         token_idx = torch.arange(t, device=x.device).repeat_interleave(self.top_k)
+        if self.ep:
+            return self._forward_ep(x, flat_ids, flat_w, token_idx)
         if ops.moe_gemm_usable(x, self.hidden, self.inter):
             # grouped-GEMM path (E16): sort token-expert pairs by expert and
             # run ONE gated launch + ONE down launch over all segments
@@ -149,15 +216,27 @@ class MixtralForCausalLM(nn.Module):
             moe = layer.moe
             moe.gate.data.copy_(
                 weights[p + "block_sparse_moe.gate.weight"].to(moe.gate.dtype))
-            for e in range(moe.num_experts):
-                ep = f"{p}block_sparse_moe.experts.{e}."
-                w1 = weights[ep + "w1.weight"]  # gate [I, H]
-                w3 = weights[ep + "w3.weight"]  # up   [I, H]
-                w2 = weights[ep + "w2.weight"]  # down [H, I]
-                sl = slice(r * moe.inter, (r + 1) * moe.inter)
-                moe.w13.data[e, :moe.inter].copy_(w1[sl].to(moe.w13.dtype))
-                moe.w13.data[e, moe.inter:].copy_(w3[sl].to(moe.w13.dtype))
-                moe.w2.data[e].copy_(w2[:, sl].to(moe.w2.dtype))
+            if moe.ep:
+                # expert-parallel: this rank holds whole experts
+                for le in range(moe.experts_per_rank):
+                    e = moe.expert0 + le
+                    ep = f"{p}block_sparse_moe.experts.{e}."
+                    moe.w13.data[le, :moe.inter].copy_(
+                        weights[ep + "w1.weight"].to(moe.w13.dtype))
+                    moe.w13.data[le, moe.inter:].copy_(
+                        weights[ep + "w3.weight"].to(moe.w13.dtype))
+                    moe.w2.data[le].copy_(
+                        weights[ep + "w2.weight"].to(moe.w2.dtype))
+            else:
+                for e in range(moe.num_experts):
+                    ep = f"{p}block_sparse_moe.experts.{e}."
+                    w1 = weights[ep + "w1.weight"]  # gate [I, H]
+                    w3 = weights[ep + "w3.weight"]  # up   [I, H]
+                    w2 = weights[ep + "w2.weight"]  # down [H, I]
+                    sl = slice(r * moe.inter, (r + 1) * moe.inter)
+                    moe.w13.data[e, :moe.inter].copy_(w1[sl].to(moe.w13.dtype))
+                    moe.w13.data[e, moe.inter:].copy_(w3[sl].to(moe.w13.dtype))
+                    moe.w2.data[e].copy_(w2[:, sl].to(moe.w2.dtype))
             layer.input_norm.weight.data.copy_(
                 weights[p + "input_layernorm.weight"].to(self.cfg.dtype))
             layer.post_norm.weight.data.copy_(

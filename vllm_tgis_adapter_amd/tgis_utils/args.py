@@ -166,6 +166,9 @@ def add_tgis_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
                                  None],
                         help="weight quantization method (4-bit surfaces "
                              "apply RTN int4-g128; also accepts int8/int4)")
+    parser.add_argument("--expert-parallel", action="store_true",
+                        help="MoE: distribute whole experts across ranks with "
+                             "token all-to-all instead of TP-sharding them")
     parser.add_argument("--num-gpus", type=int)
     parser.add_argument("--num-shard", type=int)
     parser.add_argument("--output-special-tokens", type=_bool_from_string, default=False)
@@ -267,6 +270,8 @@ def engine_config_from_args(args: argparse.Namespace):
     model_config = ModelConfig.from_model_arg(
         args.model, dtype=args.dtype or "auto", max_model_len=args.max_model_len
     )
+    if getattr(args, "expert_parallel", False):
+        model_config.expert_parallel = True
     return EngineConfig(
         model_config=model_config,
         cache_config=CacheConfig(
