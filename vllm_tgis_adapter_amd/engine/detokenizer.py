@@ -19,6 +19,36 @@ class Detokenizer:
         self._special = set(tokenizer.all_special_tokens)
         # id -> (token, token-or-"" when skipping specials); lazily filled
         self._tok_cache: dict[int, tuple[str, str]] = {}
+        # id -> decoded piece (skip-specials variant); lazily filled
+        self._piece_cache: dict[int, tuple[str, str]] = {}
+        self._ctx_free = self._probe_context_free()
+
+    def _probe_context_free(self) -> bool:
+        """True when per-token decode concatenates to the joint decode, so
+        the steady-state hot path can append cached pieces (one dict lookup)
+        instead of two windowed convert_tokens_to_string calls per token."""
+        try:
+            n = min(len(self.tokenizer), 4096)
+            ids = [i for i in range(2, n, max(1, n // 64))][:48]
+            toks = self.tokenizer.convert_ids_to_tokens(ids)
+            pieces = [self.tokenizer.convert_tokens_to_string([t]) for t in toks]
+            for i in range(0, len(ids) - 1, 2):
+                joint = self.tokenizer.convert_tokens_to_string(
+                    [toks[i], toks[i + 1]])
+                if joint != pieces[i] + pieces[i + 1]:
+                    return False
+            return True
+        except Exception:
+            return False
+
+    def _piece(self, token_id: int) -> tuple[str, str]:
+        hit = self._piece_cache.get(token_id)
+        if hit is None:
+            tok = self.tokenizer.convert_ids_to_tokens([token_id])[0]
+            s = self.tokenizer.convert_tokens_to_string([tok])
+            hit = (s, "" if tok in self._special else s)
+            self._piece_cache[token_id] = hit
+        return hit
 
     def _convert(self, ids: list[int], skip_special: bool) -> list[str]:
         toks = self.tokenizer.convert_ids_to_tokens(ids)
@@ -37,6 +67,21 @@ class Detokenizer:
     def append_token(self, req: Request, token_id: int) -> str:
         """Incrementally decode one new token; returns the new text fragment."""
         skip = req.sampling_params.skip_special_tokens
+
+        # fast path: context-free decoder, no pending partial merge -> the
+        # cached piece IS the new text (the windowed double-decode below
+        # cost ~2 Rust calls per token per request, a visible slice of the
+        # per-step postprocess at batch 512)
+        if self._ctx_free and req.read_offset == len(req.prev_token_texts):
+            piece = self._piece(token_id)[1 if skip else 0]
+            if "�" not in piece:
+                req.prev_token_texts.append(self._one_token(token_id, skip))
+                req.prefix_offset = req.read_offset
+                req.read_offset = len(req.prev_token_texts)
+                if piece:
+                    req.output_text += piece
+                return piece
+
         new_tok = self._one_token(token_id, skip)
         req.prev_token_texts.append(new_tok)
 
