@@ -172,10 +172,6 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
     f32x16_t acc_s[2];
     if (wave_active) {
       // ---- S^T = K . Q^T per 32-kv block -------------------------------
-      // setprio: this wave's MFMA stream preempts the sibling wave's loads
-      // on the CU scheduler (waves here are at independent phases, the
-      // regime where it pays — guide T11)
-      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int kb = 0; kb < 2; ++kb) {
         acc_s[kb] = f32x16_t{};
@@ -190,8 +186,6 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
         }
       }
     }
-
-    if (wave_active) __builtin_amdgcn_s_setprio(0);
 
     // ---- async stage: issue next tile's HBM loads now ------------------
     const int next_base = kv_base + KVT2;
@@ -296,7 +290,6 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
 
       // ---- PV via hardware transpose reads, per d0 block ---------------
       // quarter q of the wave covers (d0 parity = q&1, kv +8 per q>>1)
-      __builtin_amdgcn_s_setprio(1);
       const int quart = lane >> 4;
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
@@ -320,7 +313,6 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
               pa[c], u.v, acc_o[dt], 0, 0, 0);
         }
       }
-      __builtin_amdgcn_s_setprio(0);
     }
 
     // ---- write next tile into the other buffer -------------------------
