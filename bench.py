@@ -45,6 +45,10 @@ def parse_args():
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--dtype", type=str, default="bfloat16")
     p.add_argument("--block-size", type=int, default=16)
+    p.add_argument("--kv-cache-dtype", type=str, default="auto",
+                   choices=["auto", "fp8"],
+                   help="KV cache storage (fp8 is an ALTERNATE, non-headline "
+                        "config; the BASELINE metric is bf16)")
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--ttft-samples", type=int, default=5)
     p.add_argument("--grpc-port", type=int, default=18033)
@@ -109,7 +113,8 @@ def engine_config(args, device: str, tp: int):
     mc = ModelConfig.from_model_arg(args.model, dtype=args.dtype)
     return EngineConfig(
         model_config=mc,
-        cache_config=CacheConfig(block_size=args.block_size),
+        cache_config=CacheConfig(block_size=args.block_size,
+                                 kv_cache_dtype=args.kv_cache_dtype),
         scheduler_config=SchedulerConfig(
             max_num_seqs=max(args.batch, 8),
             max_num_batched_tokens=max(8192, args.batch * 2),
@@ -234,6 +239,7 @@ def run_serve(args, device: str) -> None:
         "--max-num-seqs", str(max(args.batch, 8)),
         "--max-num-batched-tokens", str(max(8192, args.batch * 2)),
         "--block-size", str(args.block_size),
+        "--kv-cache-dtype", args.kv_cache_dtype,
         "--grpc-port", str(args.grpc_port), "--port", str(args.http_port),
     ]
     if args.gpus > 1:

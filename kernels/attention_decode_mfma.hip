@@ -24,16 +24,30 @@
 typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 typedef __attribute__((ext_vector_type(16))) float f32x16_t;
 
+// 8 consecutive KV elements as bf16: direct for bf16 caches, hardware
+// cvt_pk_f32_fp8 for e4m3 caches (fp8 KV halves the decode HBM stream)
+template <typename CT>
+DEVINLINE bf16x8_t load_kv8(const CT* p);
+template <>
+DEVINLINE bf16x8_t load_kv8<__hip_bfloat16>(const __hip_bfloat16* p) {
+  return *reinterpret_cast<const bf16x8_t*>(p);
+}
+template <>
+DEVINLINE bf16x8_t load_kv8<unsigned char>(const unsigned char* p) {
+  return __builtin_bit_cast(
+      bf16x8_t, fp8x8_to_bf16x8(*reinterpret_cast<const u32x2_vec_t*>(p)));
+}
+
 #define KVT 32     // kv slots per mfma tile
 #define NWAVES 4   // waves per workgroup
 
-template <int HEAD_DIM>
+template <int HEAD_DIM, typename CT>
 __global__ __launch_bounds__(256, 2) void paged_decode_mfma_kernel(
     float* __restrict__ part_acc,   // [nseq, nheads, P, HEAD_DIM] f32
     float* __restrict__ part_ml,    // [nseq, nheads, P, 2] f32 (m, l)
     const __hip_bfloat16* __restrict__ q,        // [nseq, nheads, HD]
-    const __hip_bfloat16* __restrict__ k_cache,  // [nb, bs, kvh, HD]
-    const __hip_bfloat16* __restrict__ v_cache,
+    const CT* __restrict__ k_cache,  // [nb, bs, kvh, HD] bf16 | e4m3
+    const CT* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [nseq, max_blocks]
     const int* __restrict__ seq_lens,      // [nseq]
     const float scale,
@@ -105,12 +119,12 @@ __global__ __launch_bounds__(256, 2) void paged_decode_mfma_kernel(
       const int pos = kv_base + col;
       if (pos < seq_len) {
         const int block = btable[pos / block_size];
-        const __hip_bfloat16* k_row =
+        const CT* k_row =
             k_cache + ((long)block * block_size + pos % block_size) * kv_row_stride +
             (long)kv_head * HEAD_DIM;
 #pragma unroll
         for (int ks = 0; ks < KCH; ++ks)
-          ka[ks] = *reinterpret_cast<const bf16x8_t*>(k_row + ks * 16 + half * 8);
+          ka[ks] = load_kv8<CT>(k_row + ks * 16 + half * 8);
       } else {
 #pragma unroll
         for (int ks = 0; ks < KCH; ++ks) ka[ks] = bf16x8_t{};
@@ -132,7 +146,7 @@ __global__ __launch_bounds__(256, 2) void paged_decode_mfma_kernel(
           const long row = ((long)block * block_size + pos % block_size) *
                                kv_row_stride +
                            (long)kv_head * HEAD_DIM + d8;
-          vv[pass] = *reinterpret_cast<const bf16x8_t*>(v_cache + row);
+          vv[pass] = load_kv8<CT>(v_cache + row);
         }
       }
 #pragma unroll
@@ -282,13 +296,13 @@ int decode_mfma_num_partitions(int nseq, int kvh, int max_context) {
   return npart;
 }
 
-void launch_paged_decode_mfma(__hip_bfloat16* out, float* part_acc,
-                              float* part_ml, const __hip_bfloat16* q,
-                              const __hip_bfloat16* kc, const __hip_bfloat16* vc,
-                              const int* bt, const int* sl, float scale,
-                              int nseq, int nheads, int kvh, int head_dim,
-                              int block_size, int max_blocks, int npart,
-                              hipStream_t stream) {
+template <typename CT>
+static void launch_decode_ct(__hip_bfloat16* out, float* part_acc,
+                             float* part_ml, const __hip_bfloat16* q,
+                             const CT* kc, const CT* vc, const int* bt,
+                             const int* sl, float scale, int nseq, int nheads,
+                             int kvh, int head_dim, int block_size,
+                             int max_blocks, int npart, hipStream_t stream) {
   const int group = nheads / kvh;
   dim3 grid(nseq, kvh, npart);
   dim3 block(256);
@@ -296,20 +310,43 @@ void launch_paged_decode_mfma(__hip_bfloat16* out, float* part_acc,
   const int P = npart * NWAVES;
   switch (head_dim) {
     case 64:
-      hipLaunchKernelGGL(paged_decode_mfma_kernel<64>, grid, block, 0, stream,
-                         part_acc, part_ml, q, kc, vc, bt, sl, scale, nheads,
-                         kvh, group, block_size, max_blocks);
+      hipLaunchKernelGGL((paged_decode_mfma_kernel<64, CT>), grid, block, 0,
+                         stream, part_acc, part_ml, q, kc, vc, bt, sl, scale,
+                         nheads, kvh, group, block_size, max_blocks);
       hipLaunchKernelGGL(decode_merge_kernel<64>, mgrid, dim3(64), 0, stream,
                          out, part_acc, part_ml, nheads, P);
       break;
     case 128:
-      hipLaunchKernelGGL(paged_decode_mfma_kernel<128>, grid, block, 0, stream,
-                         part_acc, part_ml, q, kc, vc, bt, sl, scale, nheads,
-                         kvh, group, block_size, max_blocks);
+      hipLaunchKernelGGL((paged_decode_mfma_kernel<128, CT>), grid, block, 0,
+                         stream, part_acc, part_ml, q, kc, vc, bt, sl, scale,
+                         nheads, kvh, group, block_size, max_blocks);
       hipLaunchKernelGGL(decode_merge_kernel<128>, mgrid, dim3(128), 0, stream,
                          out, part_acc, part_ml, nheads, P);
       break;
     default:
       abort();
   }
+}
+
+void launch_paged_decode_mfma(__hip_bfloat16* out, float* part_acc,
+                              float* part_ml, const __hip_bfloat16* q,
+                              const __hip_bfloat16* kc, const __hip_bfloat16* vc,
+                              const int* bt, const int* sl, float scale,
+                              int nseq, int nheads, int kvh, int head_dim,
+                              int block_size, int max_blocks, int npart,
+                              hipStream_t stream) {
+  launch_decode_ct<__hip_bfloat16>(out, part_acc, part_ml, q, kc, vc, bt, sl,
+                                   scale, nseq, nheads, kvh, head_dim,
+                                   block_size, max_blocks, npart, stream);
+}
+
+void launch_paged_decode_mfma_fp8(
+    __hip_bfloat16* out, float* part_acc, float* part_ml,
+    const __hip_bfloat16* q, const unsigned char* kc, const unsigned char* vc,
+    const int* bt, const int* sl, float scale, int nseq, int nheads, int kvh,
+    int head_dim, int block_size, int max_blocks, int npart,
+    hipStream_t stream) {
+  launch_decode_ct<unsigned char>(out, part_acc, part_ml, q, kc, vc, bt, sl,
+                                  scale, nseq, nheads, kvh, head_dim,
+                                  block_size, max_blocks, npart, stream);
 }

@@ -26,6 +26,18 @@
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 typedef __attribute__((ext_vector_type(16))) float f32x16_t;
+
+template <typename CT>
+DEVINLINE bf16x8_t load_kv8_v2(const CT* p);
+template <>
+DEVINLINE bf16x8_t load_kv8_v2<__hip_bfloat16>(const __hip_bfloat16* p) {
+  return *reinterpret_cast<const bf16x8_t*>(p);
+}
+template <>
+DEVINLINE bf16x8_t load_kv8_v2<unsigned char>(const unsigned char* p) {
+  return __builtin_bit_cast(
+      bf16x8_t, fp8x8_to_bf16x8(*reinterpret_cast<const u32x2_vec_t*>(p)));
+}
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4_b;
 typedef __attribute__((address_space(3))) bf16x4_b* lds_tr_ptr;
 
@@ -46,12 +58,12 @@ __device__ inline float cross_half_sum(float v) {
   return v + __shfl_xor(v, 32, 64);
 }
 
-template <int HEAD_DIM>
+template <int HEAD_DIM, typename CT>
 __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
     __hip_bfloat16* __restrict__ out,            // [total_q, nheads, HD]
     const __hip_bfloat16* __restrict__ q,        // [total_q, nheads, HD]
-    const __hip_bfloat16* __restrict__ k_cache,  // [nb, bs, kvh, HD]
-    const __hip_bfloat16* __restrict__ v_cache,
+    const CT* __restrict__ k_cache,  // [nb, bs, kvh, HD] bf16 | e4m3
+    const CT* __restrict__ v_cache,
     const int* __restrict__ block_tables,     // [nseq, max_blocks]
     const int* __restrict__ query_start_loc,  // [nseq+1]
     const int* __restrict__ seq_lens,         // [nseq]
@@ -135,8 +147,8 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
         const long row = ((long)block * block_size + pos % block_size) *
                              kv_row_stride +
                          (long)kv_head * HEAD_DIM + st_d8;
-        st_k[pass] = *reinterpret_cast<const bf16x8_t*>(k_cache + row);
-        st_v[pass] = *reinterpret_cast<const bf16x8_t*>(v_cache + row);
+        st_k[pass] = load_kv8_v2<CT>(k_cache + row);
+        st_v[pass] = load_kv8_v2<CT>(v_cache + row);
       }
     }
   };
@@ -338,21 +350,43 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
   }
 }
 
-void launch_paged_prefill_mfma_v2(
-    __hip_bfloat16* out, const __hip_bfloat16* q, const __hip_bfloat16* kc,
-    const __hip_bfloat16* vc, const int* bt, const int* qsl, const int* sl,
-    float scale, int nseq, int nheads, int kvh, int head_dim, int block_size,
-    int max_blocks, int max_query_len, hipStream_t stream) {
+template <typename CT>
+static void launch_prefill_v2_ct(
+    __hip_bfloat16* out, const __hip_bfloat16* q, const CT* kc, const CT* vc,
+    const int* bt, const int* qsl, const int* sl, float scale, int nseq,
+    int nheads, int kvh, int head_dim, int block_size, int max_blocks,
+    int max_query_len, hipStream_t stream) {
   const int qtiles = (max_query_len + QTILE2 - 1) / QTILE2;
   dim3 grid(nseq, nheads, qtiles);
   dim3 block(512);
   switch (head_dim) {
     case 128:
-      hipLaunchKernelGGL(paged_prefill_mfma_v2_kernel<128>, grid, block, 0,
-                         stream, out, q, kc, vc, bt, qsl, sl, scale, nheads,
-                         kvh, block_size, max_blocks);
+      hipLaunchKernelGGL((paged_prefill_mfma_v2_kernel<128, CT>), grid, block,
+                         0, stream, out, q, kc, vc, bt, qsl, sl, scale,
+                         nheads, kvh, block_size, max_blocks);
       break;
     default:
       abort();
   }
+}
+
+void launch_paged_prefill_mfma_v2(
+    __hip_bfloat16* out, const __hip_bfloat16* q, const __hip_bfloat16* kc,
+    const __hip_bfloat16* vc, const int* bt, const int* qsl, const int* sl,
+    float scale, int nseq, int nheads, int kvh, int head_dim, int block_size,
+    int max_blocks, int max_query_len, hipStream_t stream) {
+  launch_prefill_v2_ct<__hip_bfloat16>(out, q, kc, vc, bt, qsl, sl, scale,
+                                       nseq, nheads, kvh, head_dim,
+                                       block_size, max_blocks, max_query_len,
+                                       stream);
+}
+
+void launch_paged_prefill_mfma_v2_fp8(
+    __hip_bfloat16* out, const __hip_bfloat16* q, const unsigned char* kc,
+    const unsigned char* vc, const int* bt, const int* qsl, const int* sl,
+    float scale, int nseq, int nheads, int kvh, int head_dim, int block_size,
+    int max_blocks, int max_query_len, hipStream_t stream) {
+  launch_prefill_v2_ct<unsigned char>(out, q, kc, vc, bt, qsl, sl, scale,
+                                      nseq, nheads, kvh, head_dim, block_size,
+                                      max_blocks, max_query_len, stream);
 }

@@ -49,6 +49,20 @@ void launch_gemm_tile(__hip_bfloat16*, const __hip_bfloat16*,
 void launch_xgmi_allreduce(const unsigned long long*, __hip_bfloat16*,
                            const __hip_bfloat16*, long, long, int, int,
                            hipStream_t);
+void launch_paged_decode_mfma_fp8(__hip_bfloat16*, float*, float*,
+                                  const __hip_bfloat16*, const unsigned char*,
+                                  const unsigned char*, const int*,
+                                  const int*, float, int, int, int, int, int,
+                                  int, int, hipStream_t);
+void launch_paged_prefill_mfma_v2_fp8(__hip_bfloat16*, const __hip_bfloat16*,
+                                      const unsigned char*,
+                                      const unsigned char*, const int*,
+                                      const int*, const int*, float, int, int,
+                                      int, int, int, int, int, hipStream_t);
+void launch_reshape_and_cache_fp8(const __hip_bfloat16*,
+                                  const __hip_bfloat16*, unsigned char*,
+                                  unsigned char*, const long*, int, int,
+                                  hipStream_t);
 void launch_gemm_skinny_q(__hip_bfloat16*, float*, const __hip_bfloat16*,
                           const unsigned char*, const float*, int, int, int,
                           int, int, hipStream_t);
@@ -169,6 +183,18 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
   const int tokens = k.size(0);
   const int row_elems = k_cache.size(2) * k_cache.size(3);
   TORCH_CHECK(slot_mapping.scalar_type() == at::ScalarType::Long);
+  if (k_cache.scalar_type() == at::ScalarType::Byte) {
+    TORCH_CHECK(k.scalar_type() == at::ScalarType::BFloat16,
+                "fp8 KV cache takes bf16 K/V inputs");
+    TORCH_CHECK(row_elems % 8 == 0);
+    launch_reshape_and_cache_fp8(
+        reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+        reinterpret_cast<unsigned char*>(k_cache.data_ptr()),
+        reinterpret_cast<unsigned char*>(v_cache.data_ptr()),
+        slot_mapping.data_ptr<long>(), tokens, row_elems, current_stream());
+    return;
+  }
   DISPATCH_FLOATING(k.scalar_type(), {
     launch_reshape_and_cache<scalar_t>(
         cptr<scalar_t>(k), cptr<scalar_t>(v), ptr<scalar_t>(k_cache),
@@ -194,6 +220,28 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     const char* e = getenv("VTA_DECODE_VALU");
     return e && e[0] == '1';
   }();
+  const bool kv_fp8 = k_cache.scalar_type() == at::ScalarType::Byte;
+  if (kv_fp8) {
+    TORCH_CHECK(q.scalar_type() == at::ScalarType::BFloat16 &&
+                    (head_dim == 64 || head_dim == 128) && group >= 1 &&
+                    group <= 8,
+                "fp8 KV cache requires the bf16 MFMA decode path");
+    const int npart = decode_mfma_num_partitions(nseq, kvh, max_blocks * block_size);
+    const int P = npart * 4;
+    auto opts = q.options().dtype(at::ScalarType::Float);
+    auto part_acc = at::empty({nseq, nheads, P, head_dim}, opts);
+    auto part_ml = at::empty({nseq, nheads, P, 2}, opts);
+    launch_paged_decode_mfma_fp8(
+        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+        part_acc.data_ptr<float>(), part_ml.data_ptr<float>(),
+        reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+        reinterpret_cast<const unsigned char*>(k_cache.data_ptr()),
+        reinterpret_cast<const unsigned char*>(v_cache.data_ptr()),
+        block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), (float)scale,
+        nseq, nheads, kvh, head_dim, block_size, max_blocks, npart,
+        current_stream());
+    return;
+  }
   if (!force_valu && q.scalar_type() == at::ScalarType::BFloat16 &&
       (head_dim == 64 || head_dim == 128) && group >= 1 && group <= 8 &&
       nheads == kvh * group && nseq > 0) {
@@ -242,6 +290,19 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
     const char* e = getenv("VTA_PREFILL_V1");
     return e && e[0] == '1';
   }();
+  if (k_cache.scalar_type() == at::ScalarType::Byte) {
+    TORCH_CHECK(q.scalar_type() == at::ScalarType::BFloat16 && head_dim == 128,
+                "fp8 KV cache requires the bf16 v2 prefill path (hd=128)");
+    launch_paged_prefill_mfma_v2_fp8(
+        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+        reinterpret_cast<const unsigned char*>(k_cache.data_ptr()),
+        reinterpret_cast<const unsigned char*>(v_cache.data_ptr()),
+        block_tables.data_ptr<int>(), query_start_loc.data_ptr<int>(),
+        seq_lens.data_ptr<int>(), (float)scale, nseq, nheads, kvh, head_dim,
+        block_size, max_blocks, (int)max_query_len, current_stream());
+    return;
+  }
   if (!force_valu && !force_v1 &&
       q.scalar_type() == at::ScalarType::BFloat16 && head_dim == 128) {
     launch_paged_prefill_mfma_v2(

@@ -48,6 +48,52 @@ DEVINLINE void store16(T* p, const Vec16<T>& v) {
 }
 
 // ---------------------------------------------------------------------------
+// fp8 (OCP e4m3fn) KV-cache conversions — gfx950 hardware converts; matches
+// torch.float8_e4m3fn bit-for-bit (verified in tests/test_ops_gpu.py)
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(2))) float cvt_f32x2_t;
+typedef __attribute__((ext_vector_type(2))) unsigned int u32x2_vec_t;
+typedef __attribute__((ext_vector_type(8))) short bf16x8_vec_t;
+
+// 8 packed e4m3 bytes (as 2 dwords) -> 8 bf16
+DEVINLINE bf16x8_vec_t fp8x8_to_bf16x8(u32x2_vec_t u) {
+  cvt_f32x2_t f[4];
+  f[0] = __builtin_amdgcn_cvt_pk_f32_fp8(u[0], false);
+  f[1] = __builtin_amdgcn_cvt_pk_f32_fp8(u[0], true);
+  f[2] = __builtin_amdgcn_cvt_pk_f32_fp8(u[1], false);
+  f[3] = __builtin_amdgcn_cvt_pk_f32_fp8(u[1], true);
+  union { unsigned int w[4]; bf16x8_vec_t v; } out;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    unsigned r;
+    asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2"
+                 : "=v"(r) : "v"(f[i][0]), "v"(f[i][1]));
+    out.w[i] = r;
+  }
+  return out.v;
+}
+
+// 8 bf16 -> 8 packed e4m3 bytes (2 dwords)
+DEVINLINE u32x2_vec_t bf16x8_to_fp8x8(bf16x8_vec_t v) {
+  union { bf16x8_vec_t v; short s[8]; } in;
+  in.v = v;
+  float f[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+    f[i] = __bfloat162float(__builtin_bit_cast(__hip_bfloat16, in.s[i]));
+  u32x2_vec_t out{};
+  int w0 = 0, w1 = 0;
+  w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[0], f[1], w0, false);
+  w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[2], f[3], w0, true);
+  w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[4], f[5], w1, false);
+  w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[6], f[7], w1, true);
+  out[0] = (unsigned)w0;
+  out[1] = (unsigned)w1;
+  return out;
+}
+
+// ---------------------------------------------------------------------------
 // reductions
 // ---------------------------------------------------------------------------
 

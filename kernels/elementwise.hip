@@ -119,6 +119,39 @@ void launch_reshape_and_cache(const T* k, const T* v, T* kc, T* vc,
                      0, s, k, v, kc, vc, slots, row_elems);
 }
 
+// fp8 KV cache (SURVEY E5 + E18 adjacency): bf16 K/V quantized to OCP e4m3
+// at cache-write time; halves the decode-attention HBM stream and doubles
+// KV capacity in the 288 GB pool.
+__global__ void reshape_and_cache_fp8_kernel(
+    const __hip_bfloat16* __restrict__ k,  // [T, kvh * hd]
+    const __hip_bfloat16* __restrict__ v,
+    unsigned char* __restrict__ k_cache,   // [nb * bs, kvh * hd] e4m3
+    unsigned char* __restrict__ v_cache,
+    const long* __restrict__ slots,
+    const int row_elems) {
+  const int token = blockIdx.x;
+  const long slot = slots[token];
+  if (slot < 0) return;
+  const __hip_bfloat16* k_src = k + (long)token * row_elems;
+  const __hip_bfloat16* v_src = v + (long)token * row_elems;
+  unsigned char* k_dst = k_cache + slot * row_elems;
+  unsigned char* v_dst = v_cache + slot * row_elems;
+  for (int i = threadIdx.x * 8; i < row_elems; i += blockDim.x * 8) {
+    *reinterpret_cast<u32x2_vec_t*>(k_dst + i) = bf16x8_to_fp8x8(
+        *reinterpret_cast<const bf16x8_vec_t*>(k_src + i));
+    *reinterpret_cast<u32x2_vec_t*>(v_dst + i) = bf16x8_to_fp8x8(
+        *reinterpret_cast<const bf16x8_vec_t*>(v_src + i));
+  }
+}
+
+void launch_reshape_and_cache_fp8(const __hip_bfloat16* k,
+                                  const __hip_bfloat16* v, unsigned char* kc,
+                                  unsigned char* vc, const long* slots,
+                                  int tokens, int row_elems, hipStream_t s) {
+  hipLaunchKernelGGL(reshape_and_cache_fp8_kernel, dim3(tokens), dim3(128), 0,
+                     s, k, v, kc, vc, slots, row_elems);
+}
+
 #define INSTANTIATE(T)                                                         \
   template void launch_silu_and_mul<T>(T*, const T*, int, int, hipStream_t);   \
   template void launch_rotary_embedding<T>(const long*, T*, T*, const float*,  \

@@ -118,6 +118,12 @@ def reshape_and_cache(
     if _native(k):
         _C.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
         return
+    if k_cache.dtype == torch.uint8:  # fp8 KV: quantize via torch e4m3
+        reference.reshape_and_cache(
+            k.to(torch.float8_e4m3fn).view(torch.uint8),
+            v.to(torch.float8_e4m3fn).view(torch.uint8),
+            k_cache, v_cache, slot_mapping)
+        return
     reference.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
 
 
@@ -140,6 +146,9 @@ def paged_attention_decode(
         return out
     n = q.shape[0]
     qsl = torch.arange(n + 1, dtype=torch.int32)
+    if k_cache.dtype == torch.uint8:  # fp8 KV: dequantize for the reference
+        k_cache = k_cache.view(torch.float8_e4m3fn).to(q.dtype)
+        v_cache = v_cache.view(torch.float8_e4m3fn).to(q.dtype)
     r = reference.paged_attention(
         q, k_cache, v_cache, block_tables.cpu(), qsl, seq_lens.cpu(), scale
     )
@@ -167,6 +176,9 @@ def paged_attention_prefill(
             scale, max_query_len
         )
         return out
+    if k_cache.dtype == torch.uint8:  # fp8 KV: dequantize for the reference
+        k_cache = k_cache.view(torch.float8_e4m3fn).to(q.dtype)
+        v_cache = v_cache.view(torch.float8_e4m3fn).to(q.dtype)
     return reference.paged_attention(
         q, k_cache, v_cache, block_tables.cpu(), query_start_loc.cpu(),
         seq_lens.cpu(), scale

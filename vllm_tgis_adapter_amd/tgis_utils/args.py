@@ -131,6 +131,9 @@ def make_engine_arg_parser(parser: argparse.ArgumentParser) -> argparse.Argument
     parser.add_argument("--block-size", type=int, default=16)
     parser.add_argument("--gpu-memory-utilization", type=float, default=0.85)
     parser.add_argument("--num-gpu-blocks", type=int, default=None)
+    parser.add_argument("--kv-cache-dtype", type=str, default="auto",
+                        choices=["auto", "fp8", "fp8_e4m3"],
+                        help="KV cache storage dtype (fp8 = OCP e4m3)")
     parser.add_argument("--enforce-eager", action="store_true",
                         help="disable hipGraph decode capture")
     parser.add_argument("--device", type=str, default="auto")
@@ -279,6 +282,8 @@ def engine_config_from_args(args: argparse.Namespace):
             gpu_memory_utilization=args.gpu_memory_utilization,
             num_gpu_blocks=args.num_gpu_blocks,
             enable_prefix_caching=args.enable_prefix_caching,
+            kv_cache_dtype=("fp8" if args.kv_cache_dtype.startswith("fp8")
+                            else "auto"),
         ),
         scheduler_config=SchedulerConfig(
             max_num_seqs=args.max_num_seqs,
