@@ -67,15 +67,26 @@ def invalidate_adapter(lora_int_id: int) -> None:
         del _STACK_CACHE[key]
 
 
+_RECENT_SETS: list = []  # LRU of active-adapter sets whose stacks stay cached
+_MAX_CACHED_SETS = 4
+
+
 def set_context(token_lora_ids: Optional[torch.Tensor], adapters: dict[int, LoRAAdapter]) -> None:
     CTX.token_lora_ids = token_lora_ids
     CTX.adapters = adapters
     new_slot_ids = tuple(sorted(adapters))
-    if new_slot_ids != CTX.slot_ids:
-        # the stacks are keyed by the active-adapter set; evict stale sets so
-        # GPU memory stays bounded by the live combination, not the history
-        for key in [k for k in _STACK_CACHE if k[0] != new_slot_ids]:
-            del _STACK_CACHE[key]
+    if new_slot_ids != CTX.slot_ids and adapters:
+        # keep stacks for a small LRU of adapter sets (batches often
+        # alternate between a few combinations); evict the rest so GPU
+        # memory stays bounded
+        if new_slot_ids in _RECENT_SETS:
+            _RECENT_SETS.remove(new_slot_ids)
+        _RECENT_SETS.append(new_slot_ids)
+        if len(_RECENT_SETS) > _MAX_CACHED_SETS:
+            _RECENT_SETS.pop(0)
+            keep = set(_RECENT_SETS)
+            for key in [k for k in _STACK_CACHE if k[0] not in keep]:
+                del _STACK_CACHE[key]
     CTX.slot_ids = new_slot_ids
     CTX.token_slots = None
     if token_lora_ids is not None and adapters:
