@@ -1,0 +1,94 @@
+"""Pipelined engine stepping == synchronous stepping, wire-exactly.
+
+The pipelined path overlaps host postprocessing with the next step's GPU
+work; stop-string finishes are decided one step late but always before the
+next token is appended, so text/token-count results must match the sync
+path exactly.
+"""
+
+import os
+
+import pytest
+import torch
+
+from vllm_tgis_adapter_amd.engine import LLMEngine, SamplingParams
+from vllm_tgis_adapter_amd.engine.config import (
+    CacheConfig, EngineConfig, ModelConfig, SchedulerConfig,
+)
+
+
+def _run(pipeline: bool, stops, max_tokens=24, n_req=6):
+    os.environ["VTA_PIPELINE"] = "1" if pipeline else "0"
+    os.environ["VTA_PIPELINE_MIN"] = "1"
+    try:
+        mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+        eng = LLMEngine(EngineConfig(
+            model_config=mc, cache_config=CacheConfig(block_size=16),
+            scheduler_config=SchedulerConfig(max_num_seqs=8,
+                                             max_num_batched_tokens=512),
+            device="cpu", seed=0,
+        ))
+        outs = {}
+        for i in range(n_req):
+            eng.add_request(
+                f"r{i}", None, [20 + i, 30 + i, 40 + i],
+                SamplingParams(temperature=0.0, max_tokens=max_tokens,
+                               stop=list(stops) if stops else None),
+            )
+        finals = {}
+        steps = 0
+        while eng.has_unfinished() and steps < 400:
+            for out in eng.step():
+                if out.finished:
+                    finals[out.request_id] = out
+            steps += 1
+        assert not eng.has_unfinished(), "engine did not drain"
+        for rid, out in finals.items():
+            o = out.outputs[0]
+            outs[rid] = (o.text, len(o.token_ids), o.finish_reason, o.stop_reason)
+        return outs
+    finally:
+        os.environ.pop("VTA_PIPELINE", None)
+        os.environ.pop("VTA_PIPELINE_MIN", None)
+
+
+@pytest.mark.parametrize("stops", [None, ["zq"], ["e", "th"]])
+def test_pipelined_matches_sync(stops):
+    sync = _run(False, stops)
+    pipe = _run(True, stops)
+    assert set(sync) == set(pipe)
+    # FINAL_ONLY-equivalent comparison: same finish reasons; cumulative
+    # output identical (DELTA streams concatenate to the same text)
+    for rid in sync:
+        assert pipe[rid][1:] == sync[rid][1:], (rid, sync[rid], pipe[rid])
+
+
+def test_pipelined_cumulative_text_matches():
+    """Compare the full visible text (CUMULATIVE outputs) with stop strings."""
+    def run(pipeline):
+        os.environ["VTA_PIPELINE"] = "1" if pipeline else "0"
+        os.environ["VTA_PIPELINE_MIN"] = "1"
+        try:
+            mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+            eng = LLMEngine(EngineConfig(
+                model_config=mc, cache_config=CacheConfig(block_size=16),
+                scheduler_config=SchedulerConfig(max_num_seqs=4,
+                                                 max_num_batched_tokens=512),
+                device="cpu", seed=0,
+            ))
+            eng.add_request("a", None, [7, 8, 9],
+                            SamplingParams(temperature=0.0, max_tokens=20,
+                                           stop=["qq", "ab"]))
+            last = None
+            steps = 0
+            while eng.has_unfinished() and steps < 200:
+                for out in eng.step():
+                    last = out
+                steps += 1
+            return (last.outputs[0].text, list(last.outputs[0].token_ids),
+                    last.outputs[0].finish_reason, last.outputs[0].stop_reason)
+        finally:
+            os.environ.pop("VTA_PIPELINE", None)
+            os.environ.pop("VTA_PIPELINE_MIN", None)
+
+    assert run(True) == run(False)

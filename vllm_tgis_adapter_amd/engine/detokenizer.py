@@ -111,6 +111,16 @@ class StopChecker:
         self.max_model_len = max_model_len
 
     def check(self, req: Request, token_id: int, new_text: str) -> None:
+        if self.check_cheap(req, token_id):
+            return
+        self.check_text(req, new_text)
+
+    def check_cheap(self, req: Request, token_id: int) -> bool:
+        """Detokenization-free finish conditions (EOS; length limits for
+        requests without stop strings — those keep reference precedence by
+        deciding after the string scan).  Safe to run before detok, so the
+        pipelined step applies it at token-append time while stop-string
+        scanning rides the deferred phase."""
         p = req.sampling_params
 
         # EOS (suppressed below min_tokens by the sampler; double-check here)
@@ -121,7 +131,19 @@ class StopChecker:
         ):
             # eos token excluded from output text by skip_special_tokens
             req.finish(RequestStatus.FINISHED_STOPPED, stop_reason=None)
-            return
+            return True
+        if not p.stop:
+            if p.max_tokens is not None and req.num_output_tokens >= p.max_tokens:
+                req.finish(RequestStatus.FINISHED_LENGTH)
+                return True
+            if req.num_tokens >= self.max_model_len:
+                req.finish(RequestStatus.FINISHED_LENGTH)
+                return True
+        return False
+
+    def check_text(self, req: Request, new_text: str) -> None:
+        """Stop-string scan (+ length limits for stop-string requests)."""
+        p = req.sampling_params
 
         # stop strings
         if p.stop and req.num_output_tokens >= p.min_tokens:

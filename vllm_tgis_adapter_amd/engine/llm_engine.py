@@ -48,6 +48,14 @@ class LLMEngine:
         self.spec_k = config.speculative_num_tokens
 
         self.metrics = EngineMetrics(self.model_config.model)
+        # pipelined stepping (overlap host postprocess with the next step's
+        # GPU work): deferred (items, [(item, token_id, logprobs)]) of the
+        # previous step, drained while the current step runs on-device
+        self._deferred = None
+        import os as _os2
+
+        self._pipeline_enabled = _os2.environ.get("VTA_PIPELINE", "1") == "1"
+        self._pipeline_min = int(_os2.environ.get("VTA_PIPELINE_MIN", "64"))
         # optional per-phase step timing (bench --timing): phase -> seconds
         self.phase_times: Optional[dict] = None
         import os as _os
@@ -100,7 +108,7 @@ class LLMEngine:
         return req.make_output()
 
     def has_unfinished(self) -> bool:
-        return self.scheduler.has_unfinished()
+        return self.scheduler.has_unfinished() or self._deferred is not None
 
     # ------------------------------------------------------------------
     # Benchmark step-window (driver contract): time EXACTLY `timed_steps`
@@ -152,6 +160,136 @@ class LLMEngine:
 
     # ------------------------------------------------------------------
     def step(self) -> list[RequestOutput]:
+        if (
+            self._pipeline_enabled
+            and not self.spec_enabled
+            and (self._deferred is not None
+                 or len(self.scheduler.running) >= self._pipeline_min)
+        ):
+            return self._step_pipelined()
+        return self._step_sync()
+
+    # ------------------------------------------------------------------
+    def _step_pipelined(self) -> list[RequestOutput]:
+        """Overlapped step: launch this step's GPU work, drain the PREVIOUS
+        step's host postprocessing (detok, stop strings, output building)
+        while it runs, then sync this step's sampled tokens.
+
+        Wire behavior is identical except that outputs are delivered one
+        engine step later and stop-STRING finishes are decided one step
+        later — always before the next token would be appended, so counts
+        and text are exact (EOS and no-stop-length finishes stay immediate
+        via StopChecker.check_cheap).  Engages only at >= VTA_PIPELINE_MIN
+        running requests; spec decode and empty batches use the sync path.
+        """
+        pt = self.phase_times
+        if pt is not None:
+            t0 = time.perf_counter()
+        sched = self.scheduler.schedule()
+        if sched.is_empty:
+            return self._drain_deferred()
+        pending = self.worker.execute_begin(sched)
+        for it in sched.items:
+            it.request.num_computed_tokens += it.num_new_tokens
+            self.block_manager.register_prefix(it.request)
+        if pt is not None:
+            t1 = time.perf_counter()
+            pt["schedule+launch"] += t1 - t0
+
+        outputs = self._drain_deferred()
+        if pt is not None:
+            t2 = time.perf_counter()
+            pt["drain(overlapped)"] += t2 - t1
+
+        result = pending.finish()
+        now = time.time()
+        sampler_out = result.sampler_output
+        pairs = []
+        for it, token_id, lp in zip(
+            self.worker._sampling_items, sampler_out.token_ids,
+            sampler_out.logprobs,
+        ):
+            req = it.request
+            if req.status.is_finished:  # aborted / late-stop before append
+                continue
+            if req.metrics.first_token_time is None:
+                req.metrics.first_token_time = now
+            req.metrics.last_token_time = now
+            req.output_token_ids.append(token_id)
+            if req.logprobs is not None:
+                req.logprobs.append(lp)
+                if lp and token_id in lp:
+                    req.cumulative_logprob += lp[token_id].logprob
+            if req.guided_state is not None:
+                req.guided_state.advance(token_id)
+            self.stop_checker.check_cheap(req, token_id)
+            pairs.append((it, token_id))
+        self._deferred = (sched.items, pairs)
+        if pt is not None:
+            t3 = time.perf_counter()
+            pt["sync+append"] += t3 - t2
+            pt["steps"] += 1
+            pt["tokens"] += len(sampler_out.token_ids)
+            if self._timing_autoprint and pt["steps"] >= 128:
+                import sys as _sys
+
+                n = pt.pop("steps")
+                toks = pt.pop("tokens", 0)
+                pt.pop("prefill_steps", 0)
+                parts = {k: round(v / n * 1e3, 2) for k, v in pt.items()}
+                print(f"[step-timing/pipelined] {n} steps ({toks} tokens): "
+                      f"per-step ms {parts}", file=_sys.stderr, flush=True)
+                pt.clear()
+        if getattr(self, "_bench_win", None) is not None:
+            self._bench_tick(len(sampler_out.token_ids))
+        self.metrics.num_running.set(len(self.scheduler.running))
+        self.metrics.num_waiting.set(len(self.scheduler.waiting))
+        self.metrics.kv_usage.set(
+            1.0 - self.block_manager.num_free_blocks / max(1, self.block_manager.num_blocks)
+        )
+        return outputs
+
+    def _drain_deferred(self) -> list[RequestOutput]:
+        d = self._deferred
+        self._deferred = None
+        if d is None:
+            return []
+        items, pairs = d
+        for it, token_id in pairs:
+            req = it.request
+            if req.status.is_finished:
+                # aborted since; detok state no longer matters
+                continue
+            new_text = self.detokenizer.append_token(req, token_id)
+            self.stop_checker.check_text(req, new_text)
+
+        outputs: list[RequestOutput] = []
+        seen = set()
+        for it in items:
+            req = it.request
+            if id(req) in seen:
+                continue
+            seen.add(id(req))
+            out = req.make_output()
+            if out is not None:
+                outputs.append(out)
+            if req.status.is_finished:
+                self.scheduler.finish_request(req)
+                m = req.metrics
+                self.metrics.request_success.inc()
+                self.metrics.prompt_tokens.inc(req.num_prompt_tokens)
+                self.metrics.generation_tokens.inc(req.num_output_tokens)
+                if m.first_token_time and m.first_scheduled_time:
+                    self.metrics.ttft.observe(m.first_token_time - m.arrival_time)
+                    if req.num_output_tokens > 1 and m.last_token_time:
+                        self.metrics.time_per_output_token.observe(
+                            (m.last_token_time - m.first_token_time)
+                            / (req.num_output_tokens - 1)
+                        )
+        return outputs
+
+    # ------------------------------------------------------------------
+    def _step_sync(self) -> list[RequestOutput]:
         pt = self.phase_times
         if pt is not None:
             import torch as _torch

@@ -64,24 +64,20 @@ class Sampler:
         return req.generator
 
     @torch.inference_mode()
-    def sample(self, logits: torch.Tensor, requests: list[Request]) -> SamplerOutput:
-        """logits: [N, vocab] raw lm-head outputs for the N sampling rows.
+    def try_launch_fused(
+        self, logits: torch.Tensor, requests: list[Request]
+    ) -> Optional[torch.Tensor]:
+        """Launch the fused one-pass sampler WITHOUT host sync.
 
-        The all-greedy no-extras batch (the serving steady state) stays in
-        the lm-head dtype end to end — one argmax pass, no [N, vocab] f32
-        materialisation (2x 131 MB per step at batch 256 / 128k vocab).
+        Returns the device token-id tensor (caller syncs via .tolist()) when
+        every row is greedy or plain temperature sampling — no top-k/top-p,
+        penalties, processors, guided masks or logprob requests — else None.
+        The pipelined engine step uses this to overlap host postprocessing
+        with the next step's GPU work.
         """
-        n, vocab = logits.shape
-        assert n == len(requests)
-
-        # Fused path (E7): rows that are greedy or plain temperature sampling
-        # (no top-k/top-p filter, penalties, processors, guided masks or
-        # logprob requests) sample in ONE kernel pass via the
-        # exponential-race identity; per-request seeded noise keeps seeded
-        # sampling deterministic.
         from .. import ops as _ops
 
-        fused_ok = _ops.native_enabled(logits)
+        n, vocab = logits.shape
         simple = True
         any_sampling = False
         ban_rows: list[int] = []  # rows with min_tokens pending: EOS banned
@@ -105,7 +101,9 @@ class Sampler:
                 if not (p.top_k <= 0 or p.top_k >= vocab) or p.top_p < 1.0:
                     simple = False
                     break
-        if simple and ban_rows:
+        if not simple:
+            return None
+        if ban_rows:
             # min_tokens EOS suppression as ONE batched scatter so the whole
             # batch stays on the fused path (a per-request python loop here
             # cost ~11 ms/step at batch 512 — the r2 serving-bench regression)
@@ -113,27 +111,56 @@ class Sampler:
                 torch.tensor(ban_rows, device=logits.device),
                 torch.tensor(ban_eos, device=logits.device),
             ] = _NEG_INF
-        if simple and fused_ok:
-            temps = torch.tensor(
-                [r.sampling_params.temperature for r in requests],
-                dtype=torch.float32, device=logits.device,
+        if not _ops.native_enabled(logits):
+            # CPU / force-reference: plain argmax-or-race, still deferred-sync
+            if not any_sampling:
+                return torch.argmax(logits, dim=-1)
+            probs = torch.softmax(
+                logits.float() / torch.tensor(
+                    [max(r.sampling_params.temperature, 1e-6) for r in requests]
+                ).unsqueeze(1),
+                dim=-1,
             )
-            noise = None
-            if any_sampling:
-                noise = torch.empty((n, vocab), dtype=torch.float32, device=logits.device)
-                for i, req in enumerate(requests):
-                    if req.sampling_params.temperature != 0.0:
-                        g = self._generator_for(req)
-                        if g is None:
-                            noise[i].exponential_()
-                        else:
-                            noise[i].exponential_(generator=g)
-            out = torch.empty(n, dtype=torch.long, device=logits.device)
-            _ops.sample_argmax(out, logits, temps, noise)
-            return SamplerOutput(token_ids=out.tolist(), logprobs=[None] * n)
-        if simple and not any_sampling:
-            sampled_cpu = torch.argmax(logits, dim=-1).tolist()
-            return SamplerOutput(token_ids=sampled_cpu, logprobs=[None] * n)
+            q = torch.empty_like(probs)
+            for i, req in enumerate(requests):
+                g = self._generator_for(req)
+                if g is None:
+                    q[i].exponential_()
+                else:
+                    q[i].exponential_(generator=g)
+            return torch.argmax(probs / q, dim=-1)
+        temps = torch.tensor(
+            [r.sampling_params.temperature for r in requests],
+            dtype=torch.float32, device=logits.device,
+        )
+        noise = None
+        if any_sampling:
+            noise = torch.empty((n, vocab), dtype=torch.float32, device=logits.device)
+            for i, req in enumerate(requests):
+                if req.sampling_params.temperature != 0.0:
+                    g = self._generator_for(req)
+                    if g is None:
+                        noise[i].exponential_()
+                    else:
+                        noise[i].exponential_(generator=g)
+        out = torch.empty(n, dtype=torch.long, device=logits.device)
+        _ops.sample_argmax(out, logits, temps, noise)
+        return out
+
+    @torch.inference_mode()
+    def sample(self, logits: torch.Tensor, requests: list[Request]) -> SamplerOutput:
+        """logits: [N, vocab] raw lm-head outputs for the N sampling rows.
+
+        The all-greedy no-extras batch (the serving steady state) stays in
+        the lm-head dtype end to end — one argmax pass, no [N, vocab] f32
+        materialisation (2x 131 MB per step at batch 256 / 128k vocab).
+        """
+        n, vocab = logits.shape
+        assert n == len(requests)
+
+        fused = self.try_launch_fused(logits, requests)
+        if fused is not None:
+            return SamplerOutput(token_ids=fused.tolist(), logprobs=[None] * n)
 
         logits = logits.float()
 

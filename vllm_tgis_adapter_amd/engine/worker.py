@@ -34,6 +34,35 @@ class ExecuteResult:
     # item-aligned prompt logprob additions handled directly on Request objects
 
 
+class _DoneStep:
+    """Already-synchronous step result (slow sampling paths)."""
+
+    __slots__ = ("_result",)
+
+    def __init__(self, result: ExecuteResult):
+        self._result = result
+
+    def finish(self) -> ExecuteResult:
+        return self._result
+
+
+class _PendingStep:
+    """Fused-sampler step in flight; finish() syncs the sampled token ids."""
+
+    __slots__ = ("_out",)
+
+    def __init__(self, out: torch.Tensor):
+        self._out = out
+
+    def finish(self) -> ExecuteResult:
+        ids = self._out.tolist()
+        return ExecuteResult(
+            sampler_output=SamplerOutput(token_ids=ids,
+                                         logprobs=[None] * len(ids)),
+            spec_results=[],
+        )
+
+
 def _pad_block_tables(tables, device) -> torch.Tensor:
     if isinstance(tables, np.ndarray):  # already padded (TP broadcast path)
         if tables.size == 0:
@@ -529,8 +558,16 @@ class Worker:
         return self.model.compute_logits(hidden[rows])
 
     # ------------------------------------------------------------------
-    def execute(self, sched: SchedulerOutput) -> ExecuteResult:
-        """Rank-0 entry point for one engine step."""
+    def execute_begin(self, sched: SchedulerOutput):
+        """Launch one step WITHOUT host sync where possible.
+
+        Returns an object with ``.finish() -> ExecuteResult``.  When the whole
+        sampling batch takes the fused kernel, finish() only does the final
+        device->host copy — the pipelined engine step calls it AFTER doing
+        the previous step's host postprocessing, overlapping that work with
+        this step's GPU execution.  Slow-path batches (logprobs, penalties,
+        spec rows, prompt logprobs) complete synchronously inside this call.
+        """
         import time as _time
 
         _tb = _time.perf_counter()
@@ -546,6 +583,19 @@ class Worker:
             dist.broadcast(t, src=0)
         logits = self.execute_batch(batch)
 
+        ns = batch["num_sample_rows"]
+        if ns and not self._spec_items and not self._prompt_lp_specs:
+            sampling_reqs = [it.request for it in self._sampling_items]
+            fused = self.sampler.try_launch_fused(logits[:ns], sampling_reqs)
+            if fused is not None:
+                return _PendingStep(fused)
+        return _DoneStep(self._execute_finish(batch, logits))
+
+    def execute(self, sched: SchedulerOutput) -> ExecuteResult:
+        """Rank-0 entry point for one engine step (synchronous)."""
+        return self.execute_begin(sched).finish()
+
+    def _execute_finish(self, batch: dict, logits) -> ExecuteResult:
         ns = batch["num_sample_rows"]
         sampling_reqs = [it.request for it in self._sampling_items]
         if ns:
