@@ -63,6 +63,22 @@ class _PendingStep:
         )
 
 
+def _pad_np_tables(tables) -> np.ndarray:
+    """Pad a (possibly ragged) block-table list to a dense int32 array."""
+    if isinstance(tables, np.ndarray):
+        if tables.size == 0:
+            return np.zeros((0, 0), dtype=np.int32)
+        return np.ascontiguousarray(tables, dtype=np.int32)
+    if not tables:
+        return np.zeros((0, 0), dtype=np.int32)
+    maxb = max(1, max(len(t) for t in tables))
+    arr = np.zeros((len(tables), maxb), dtype=np.int32)
+    for i, t in enumerate(tables):
+        if t:
+            arr[i, : len(t)] = t
+    return arr
+
+
 def _pad_block_tables(tables, device) -> torch.Tensor:
     if isinstance(tables, np.ndarray):  # already padded (TP broadcast path)
         if tables.size == 0:
@@ -176,6 +192,13 @@ class Worker:
         self.sampler = Sampler(self.device)
         self.graph_runner = None  # set by capture_decode_graphs()
         self.loras: dict[int, object] = {}  # lora_int_id -> LoRAAdapter
+
+        # pinned H2D staging, double-buffered by step parity: .to(cuda,
+        # non_blocking=True) from pageable numpy is silently SYNCHRONOUS
+        # (it serializes with the stream, stalling the pipelined step's
+        # launch phase behind the previous step's GPU work)
+        self._pin: dict[str, torch.Tensor] = {}
+        self._pin_flip = 0
 
         if self.tp > 1 and self.device == "cuda":
             from ..parallel import init_xgmi_allreduce
@@ -502,11 +525,27 @@ class Worker:
 
     # ------------------------------------------------------------------
     @torch.inference_mode()
+    def _h2d(self, name: str, arr: np.ndarray, dtype: torch.dtype) -> torch.Tensor:
+        """numpy -> device via a persistent pinned staging buffer (true async)."""
+        if self.device != "cuda":
+            return torch.from_numpy(np.ascontiguousarray(arr)).to(dtype)
+        need = int(arr.size)
+        key = f"{name}.{self._pin_flip}"
+        buf = self._pin.get(key)
+        if buf is None or buf.numel() < need:
+            buf = torch.empty(max(need, 256), dtype=dtype, pin_memory=True)
+            self._pin[key] = buf
+        staging = buf[:need]
+        staging.copy_(torch.from_numpy(np.ascontiguousarray(arr)).view(-1).to(dtype))
+        return staging.to(self.device, non_blocking=True).view(*arr.shape)
+
+    @torch.inference_mode()
     def execute_batch(self, batch: dict) -> Optional[torch.Tensor]:
         """All ranks: run the forward; returns logits rows on every rank."""
         dev = self.device
-        ids = torch.from_numpy(np.asarray(batch["token_ids"], dtype=np.int64)).to(dev, non_blocking=True)
-        pos = torch.from_numpy(np.asarray(batch["positions"], dtype=np.int64)).to(dev, non_blocking=True)
+        self._pin_flip ^= 1
+        ids = self._h2d("ids", np.asarray(batch["token_ids"], dtype=np.int64), torch.int64)
+        pos = self._h2d("pos", np.asarray(batch["positions"], dtype=np.int64), torch.int64)
         np_seqs = len(batch["prefill_seq_lens"])
 
         # hipGraph replay path: pure-decode batch, every row samples, no LoRA
@@ -519,9 +558,9 @@ class Worker:
             and len(batch["logit_rows"]) == n
             and self.graph_runner.bucket_for(n) is not None
         ):
-            slots = torch.from_numpy(np.asarray(batch["slot_mapping"], dtype=np.int64)).to(dev, non_blocking=True)
-            seq_lens = torch.from_numpy(np.asarray(batch["decode_seq_lens"], dtype=np.int32)).to(dev, non_blocking=True)
-            bt = _pad_block_tables(batch["decode_tables"], dev)
+            slots = self._h2d("slots", np.asarray(batch["slot_mapping"], dtype=np.int64), torch.int64)
+            seq_lens = self._h2d("dsl", np.asarray(batch["decode_seq_lens"], dtype=np.int32), torch.int32)
+            bt = self._h2d("dbt", _pad_np_tables(batch["decode_tables"]), torch.int32)
             return self.graph_runner.run(ids, pos, slots, seq_lens, bt)
 
         npt = batch["qsl"][-1] if np_seqs else 0
@@ -529,15 +568,15 @@ class Worker:
             num_prefill_seqs=np_seqs,
             num_prefill_tokens=npt,
             num_decode_seqs=len(batch["decode_seq_lens"]),
-            slot_mapping=torch.from_numpy(np.asarray(batch["slot_mapping"], dtype=np.int64)).to(dev, non_blocking=True),
-            prefill_query_start_loc=torch.from_numpy(np.asarray(batch["qsl"], dtype=np.int32)).to(dev, non_blocking=True),
-            prefill_seq_lens=torch.from_numpy(np.asarray(batch["prefill_seq_lens"], dtype=np.int32)).to(dev, non_blocking=True),
-            prefill_block_tables=_pad_block_tables(batch["prefill_tables"], dev),
+            slot_mapping=self._h2d("slots", np.asarray(batch["slot_mapping"], dtype=np.int64), torch.int64),
+            prefill_query_start_loc=self._h2d("qsl", np.asarray(batch["qsl"], dtype=np.int32), torch.int32),
+            prefill_seq_lens=self._h2d("psl", np.asarray(batch["prefill_seq_lens"], dtype=np.int32), torch.int32),
+            prefill_block_tables=self._h2d("pbt", _pad_np_tables(batch["prefill_tables"]), torch.int32),
             max_prefill_query_len=max(
                 (b - a for a, b in zip(batch["qsl"], batch["qsl"][1:])), default=0),
             max_prefill_seq_len=max(batch["prefill_seq_lens"], default=0),
-            decode_seq_lens=torch.from_numpy(np.asarray(batch["decode_seq_lens"], dtype=np.int32)).to(dev, non_blocking=True),
-            decode_block_tables=_pad_block_tables(batch["decode_tables"], dev),
+            decode_seq_lens=self._h2d("dsl", np.asarray(batch["decode_seq_lens"], dtype=np.int32), torch.int32),
+            decode_block_tables=self._h2d("dbt", _pad_np_tables(batch["decode_tables"]), torch.int32),
             max_decode_seq_len=max(batch["decode_seq_lens"], default=0),
         )
         from . import lora as lora_rt
@@ -558,6 +597,7 @@ class Worker:
         return self.model.compute_logits(hidden[rows])
 
     # ------------------------------------------------------------------
+    @torch.inference_mode()
     def execute_begin(self, sched: SchedulerOutput):
         """Launch one step WITHOUT host sync where possible.
 
