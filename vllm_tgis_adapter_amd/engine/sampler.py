@@ -53,6 +53,23 @@ class Sampler:
     def __init__(self, device: str, max_logprobs: int = 11):
         self.device = device
         self.max_logprobs = max_logprobs
+        # pinned staging (double-buffered): pageable host->device uploads are
+        # silently synchronous and would stall the pipelined step's launch
+        self._pin: dict = {}
+        self._pin_flip = 0
+
+    def _h2d_list(self, name: str, data: list, dtype: torch.dtype) -> torch.Tensor:
+        if self.device != "cuda":
+            return torch.tensor(data, dtype=dtype)
+        n = len(data)
+        key = f"{name}.{self._pin_flip}"
+        buf = self._pin.get(key)
+        if buf is None or buf.numel() < n:
+            buf = torch.empty(max(n, 256), dtype=dtype, pin_memory=True)
+            self._pin[key] = buf
+        staging = buf[:n]
+        staging.copy_(torch.tensor(data, dtype=dtype))
+        return staging.to(self.device, non_blocking=True)
 
     def _generator_for(self, req: Request) -> Optional[torch.Generator]:
         if req.sampling_params.seed is None:
@@ -103,13 +120,14 @@ class Sampler:
                     break
         if not simple:
             return None
+        self._pin_flip ^= 1
         if ban_rows:
             # min_tokens EOS suppression as ONE batched scatter so the whole
             # batch stays on the fused path (a per-request python loop here
             # cost ~11 ms/step at batch 512 — the r2 serving-bench regression)
             logits[
-                torch.tensor(ban_rows, device=logits.device),
-                torch.tensor(ban_eos, device=logits.device),
+                self._h2d_list("ban_r", ban_rows, torch.long),
+                self._h2d_list("ban_e", ban_eos, torch.long),
             ] = _NEG_INF
         if not _ops.native_enabled(logits):
             # CPU / force-reference: plain argmax-or-race, still deferred-sync
@@ -129,9 +147,9 @@ class Sampler:
                 else:
                     q[i].exponential_(generator=g)
             return torch.argmax(probs / q, dim=-1)
-        temps = torch.tensor(
-            [r.sampling_params.temperature for r in requests],
-            dtype=torch.float32, device=logits.device,
+        temps = self._h2d_list(
+            "temps", [r.sampling_params.temperature for r in requests],
+            torch.float32,
         )
         noise = None
         if any_sampling:
