@@ -621,13 +621,36 @@ class Worker:
             t = torch.from_numpy(payload).to(
                 self.device if self.device == "cuda" else "cpu")
             dist.broadcast(t, src=0)
+        import os as _os
+        import time as _time
+
+        _dbg = _os.environ.get("VTA_LAUNCH_TIMING", "0") == "1"
+        if _dbg:
+            _t1 = _time.perf_counter()
         logits = self.execute_batch(batch)
+        if _dbg:
+            _t2 = _time.perf_counter()
 
         ns = batch["num_sample_rows"]
         if ns and not self._spec_items and not self._prompt_lp_specs:
             sampling_reqs = [it.request for it in self._sampling_items]
             fused = self.sampler.try_launch_fused(logits[:ns], sampling_reqs)
             if fused is not None:
+                if _dbg:
+                    _t3 = _time.perf_counter()
+                    d = self.__dict__.setdefault("_lt", [0.0, 0.0, 0.0, 0])
+                    d[0] += _t1 - _tb
+                    d[1] += _t2 - _t1
+                    d[2] += _t3 - _t2
+                    d[3] += 1
+                    if d[3] >= 128:
+                        import sys as _sys
+
+                        print(f"[launch-timing] per-step ms build={d[0]/d[3]*1e3:.2f} "
+                              f"forward-launch={d[1]/d[3]*1e3:.2f} "
+                              f"sampler-launch={d[2]/d[3]*1e3:.2f}",
+                              file=_sys.stderr, flush=True)
+                        self._lt = [0.0, 0.0, 0.0, 0]
                 return _PendingStep(fused)
         return _DoneStep(self._execute_finish(batch, logits))
 
