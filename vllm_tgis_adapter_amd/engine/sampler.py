@@ -120,6 +120,12 @@ class Sampler:
                     break
         if not simple:
             return None
+        import os as _os
+        import time as _time
+
+        _dbg = _os.environ.get("VTA_LAUNCH_TIMING", "0") == "1"
+        if _dbg:
+            _ta = _time.perf_counter()
         self._pin_flip ^= 1
         if ban_rows:
             # min_tokens EOS suppression as ONE batched scatter so the whole
@@ -161,8 +167,26 @@ class Sampler:
                         noise[i].exponential_()
                     else:
                         noise[i].exponential_(generator=g)
+        if _dbg:
+            _tb2 = _time.perf_counter()
         out = torch.empty(n, dtype=torch.long, device=logits.device)
+        if _dbg:
+            _tc = _time.perf_counter()
         _ops.sample_argmax(out, logits, temps, noise)
+        if _dbg:
+            _td = _time.perf_counter()
+            d = self.__dict__.setdefault("_lt2", [0.0, 0.0, 0.0, 0])
+            d[0] += _tb2 - _ta
+            d[1] += _tc - _tb2
+            d[2] += _td - _tc
+            d[3] += 1
+            if d[3] >= 128:
+                import sys as _sys
+
+                print(f"[sampler-timing] per-step ms ban+temps={d[0]/d[3]*1e3:.2f} "
+                      f"empty={d[1]/d[3]*1e3:.2f} kernel-launch={d[2]/d[3]*1e3:.2f}",
+                      file=_sys.stderr, flush=True)
+                self._lt2 = [0.0, 0.0, 0.0, 0]
         return out
 
     @torch.inference_mode()
