@@ -128,13 +128,14 @@ class Sampler:
             _ta = _time.perf_counter()
         self._pin_flip ^= 1
         if ban_rows:
-            # min_tokens EOS suppression as ONE batched scatter so the whole
-            # batch stays on the fused path (a per-request python loop here
-            # cost ~11 ms/step at batch 512 — the r2 serving-bench regression)
-            logits[
-                self._h2d_list("ban_r", ban_rows, torch.long),
-                self._h2d_list("ban_e", ban_eos, torch.long),
-            ] = _NEG_INF
+            # min_tokens EOS suppression as ONE batched index_fill_ so the
+            # whole batch stays on the fused path.  NOTE: advanced-index
+            # assignment (index_put_) host-synchronizes on ROCm — it cost a
+            # full hidden GPU-step stall in the pipelined launch phase;
+            # flat index_fill_ stays asynchronous.
+            flat = [r * vocab + e for r, e in zip(ban_rows, ban_eos)]
+            logits.view(-1).index_fill_(
+                0, self._h2d_list("ban", flat, torch.long), _NEG_INF)
         if not _ops.native_enabled(logits):
             # CPU / force-reference: plain argmax-or-race, still deferred-sync
             if not any_sampling:
