@@ -125,7 +125,7 @@ def engine_config(args, device: str, tp: int):
     )
 
 
-async def serve_drive(args, gen_budget: int) -> dict:
+async def serve_drive(args, gen_budget: int, stabilize_s: float) -> dict:
     """Client side: C always-on streams, exactly-K-steps window, TTFT probes."""
     import grpc
     import grpc.aio
@@ -187,9 +187,25 @@ async def serve_drive(args, gen_budget: int) -> dict:
         except grpc.aio.AioRpcError:
             pass  # stream cancelled at teardown
 
-    tasks = [asyncio.get_event_loop().create_task(one_stream(i, gen_budget))
+    errors: list = []
+
+    async def guarded(i):
+        try:
+            await one_stream(i, gen_budget)
+        except Exception as e:  # surface instead of hanging the first-token wait
+            errors.append(repr(e))
+        finally:
+            first_token_evt[i].set()
+
+    tasks = [asyncio.get_event_loop().create_task(guarded(i))
              for i in range(args.batch)]
     await asyncio.gather(*(e.wait() for e in first_token_evt))
+    if errors:
+        raise RuntimeError(f"streams failed: {errors[0]} (+{len(errors)-1} more)")
+    # let the delivery pipeline reach steady state: the prefill-ramp backlog
+    # must drain BEFORE t0, otherwise lag shrinking across the window counts
+    # pre-window production as in-window delivery (overstates the value)
+    await asyncio.sleep(stabilize_s)
 
     # every stream is decoding: arm the exactly-K-steps window
     def _post():
@@ -249,10 +265,17 @@ def run_serve(args, device: str) -> None:
     log = open(args.server_log, "w")
     srv = subprocess.Popen(cmd, stdout=log, stderr=log, env=env)
 
-    # budget so no stream finishes inside warmup+window (+ramp margin)
-    gen_budget = args.warmup + args.steps + 256
+    # budget so no stream finishes inside ramp+stabilization+warmup+window,
+    # capped by the model context (prompt + min_new must fit max_model_len)
+    mc = engine_config(args, device, tp=args.gpus).model_config
+    stabilize_s = 3.0 if device == "cuda" else 0.5
+    room = mc.max_model_len - args.prompt_len - 8
+    gen_budget = min(args.warmup + args.steps + 512, room)
+    if gen_budget < args.warmup + args.steps + 8:
+        raise SystemExit(
+            f"context too small for the step window: room={room}")
     try:
-        res = asyncio.run(serve_drive(args, gen_budget))
+        res = asyncio.run(serve_drive(args, gen_budget, stabilize_s))
     finally:
         srv.terminate()
         try:
