@@ -142,3 +142,40 @@ def test_ngram_spec_decode_gpu():
         steps += 1
         assert steps < 200
     assert final is not None and len(final.outputs[0].token_ids) == 24
+
+
+def test_draft_model_spec_gpu_exact():
+    """Draft-model speculation on GPU: greedy output must equal non-spec."""
+    import torch
+
+    from vllm_tgis_adapter_amd.engine import LLMEngine, SamplingParams
+    from vllm_tgis_adapter_amd.engine.config import (
+        CacheConfig, EngineConfig, ModelConfig, SchedulerConfig,
+    )
+
+    def run(spec):
+        mc = ModelConfig.from_model_arg("tiny-llama", dtype="bfloat16")
+        eng = LLMEngine(EngineConfig(
+            model_config=mc,
+            cache_config=CacheConfig(block_size=16, num_gpu_blocks=128),
+            scheduler_config=SchedulerConfig(max_num_seqs=4,
+                                             max_num_batched_tokens=256),
+            device="cuda", seed=3, speculative_model=spec,
+            speculative_num_tokens=4,
+        ))
+        eng.add_request("a", None, [15, 16, 17],
+                        SamplingParams(temperature=0.0, max_tokens=16))
+        toks = None
+        steps = 0
+        while eng.has_unfinished() and steps < 100:
+            for out in eng.step():
+                if out.finished:
+                    toks = list(out.outputs[0].token_ids)
+            steps += 1
+        eng.shutdown()
+        return toks, steps
+
+    base, base_steps = run(None)
+    spec, spec_steps = run("tiny-llama")
+    assert spec == base
+    assert spec_steps < base_steps  # identical draft: full acceptance

@@ -120,3 +120,48 @@ def test_spec_plus_prefix_cache_equivalence():
     fancy, _ = run_all(both, reqs)
     plain, _ = run_all(make(False), reqs)
     assert fancy == plain
+
+
+def _run_engine(spec_model, prompt, max_tokens=24, seed=0):
+    from vllm_tgis_adapter_amd.engine import LLMEngine, SamplingParams
+    from vllm_tgis_adapter_amd.engine.config import (
+        CacheConfig, EngineConfig, ModelConfig, SchedulerConfig,
+    )
+
+    mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+    eng = LLMEngine(EngineConfig(
+        model_config=mc, cache_config=CacheConfig(block_size=16),
+        scheduler_config=SchedulerConfig(max_num_seqs=4,
+                                         max_num_batched_tokens=512),
+        device="cpu", seed=seed, speculative_model=spec_model,
+        speculative_num_tokens=4,
+    ))
+    eng.add_request("a", None, list(prompt),
+                    SamplingParams(temperature=0.0, max_tokens=max_tokens))
+    steps = 0
+    toks = None
+    while eng.has_unfinished() and steps < 200:
+        for out in eng.step():
+            if out.finished:
+                toks = list(out.outputs[0].token_ids)
+        steps += 1
+    assert toks is not None
+    return toks, steps
+
+
+def test_draft_model_spec_exact_and_faster():
+    """Draft model == target model: every draft accepted; greedy output must
+    equal the non-speculative run exactly and take fewer engine steps."""
+    base, base_steps = _run_engine(None, [11, 12, 13, 14])
+    spec, spec_steps = _run_engine("tiny-llama", [11, 12, 13, 14])
+    assert spec == base
+    assert spec_steps < base_steps
+
+
+def test_draft_model_spec_exact_with_divergent_draft():
+    """Draft with different weights (engine seed differs -> draft synthetic
+    weights differ from target's): acceptance is partial but verification
+    keeps greedy output EXACT."""
+    base, _ = _run_engine(None, [3, 4, 5], seed=7)
+    spec, _ = _run_engine("tiny-llama", [3, 4, 5], seed=7)
+    assert spec == base

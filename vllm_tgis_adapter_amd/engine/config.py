@@ -48,6 +48,12 @@ _PRESETS: dict[str, dict] = {
         head_dim=64, rope_theta=0.0, rms_norm_eps=1e-5,
         max_model_len=2048, tie_word_embeddings=True,
     ),
+    "llama-1b-draft": dict(  # draft model for llama-3 speculation (E17)
+        architecture="llama", vocab_size=128256, hidden_size=2048,
+        intermediate_size=8192, num_layers=16, num_heads=32, num_kv_heads=8,
+        head_dim=64, rope_theta=500000.0, rms_norm_eps=1e-5,
+        max_model_len=8192, tie_word_embeddings=True,
+    ),
     "tiny-llama": dict(  # CPU protocol tests
         architecture="llama", vocab_size=2048, hidden_size=64,
         intermediate_size=128, num_layers=2, num_heads=4, num_kv_heads=2,

@@ -44,7 +44,12 @@ class LLMEngine:
         from . import spec as _spec
         from .metrics import EngineMetrics
 
-        self.spec_enabled = _spec.is_ngram_spec(config.speculative_model)
+        from .draft import is_draft_model_spec
+
+        self.spec_draft_model = is_draft_model_spec(config.speculative_model)
+        self.spec_enabled = (
+            _spec.is_ngram_spec(config.speculative_model) or self.spec_draft_model
+        )
         self.spec_k = config.speculative_num_tokens
 
         self.metrics = EngineMetrics(self.model_config.model)
@@ -300,11 +305,20 @@ class LLMEngine:
         if self.spec_enabled:
             from . import spec as _spec
 
-            for req in self.scheduler.running:
-                req.spec_draft = (
-                    _spec.propose(req, self.spec_k, self.model_config.max_model_len)
-                    if _spec.eligible(req) else []
-                )
+            if self.spec_draft_model:
+                eligible = []
+                for req in self.scheduler.running:
+                    req.spec_draft = []
+                    if _spec.eligible(req):
+                        eligible.append(req)
+                if eligible:
+                    self.worker.draft.propose(eligible, self.spec_k)
+            else:
+                for req in self.scheduler.running:
+                    req.spec_draft = (
+                        _spec.propose(req, self.spec_k, self.model_config.max_model_len)
+                        if _spec.eligible(req) else []
+                    )
         sched = self.scheduler.schedule()
         if sched.is_empty:
             return []
@@ -338,6 +352,10 @@ class LLMEngine:
             req.num_computed_tokens -= len(draft) - accepted
             new_tokens = draft[:accepted] + [preds[accepted]]
             spec_token_count += len(new_tokens)
+            if self.spec_draft_model:
+                # draft KV beyond the accepted prefix diverged: roll back
+                req.draft_computed = min(req.draft_computed,
+                                         req._draft_base + accepted)
             if req.metrics.first_token_time is None:
                 req.metrics.first_token_time = now
             req.metrics.last_token_time = now

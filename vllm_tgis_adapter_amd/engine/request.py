@@ -74,6 +74,9 @@ class Request:
         self.registered_blocks = 0
         # speculative decoding: drafts pending verification this step
         self.spec_draft: list[int] = []
+        # draft-model speculation: tokens the draft model has ingested
+        self.draft_computed = 0
+        self._draft_base = 0
 
         # Incremental detokenization state
         self.output_text = ""

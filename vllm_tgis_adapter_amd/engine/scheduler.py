@@ -189,6 +189,7 @@ class Scheduler:
         # (already-generated tokens are recomputed like prompt tokens).
         req.num_computed_tokens = 0
         req.spec_draft = []
+        req.draft_computed = 0  # draft-model KV freed with the blocks
         req.status = RequestStatus.PREEMPTED
         self.waiting.appendleft(req)
         out.preempted.append(req)

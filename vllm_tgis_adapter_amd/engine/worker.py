@@ -193,6 +193,17 @@ class Worker:
         self.graph_runner = None  # set by capture_decode_graphs()
         self.loras: dict[int, object] = {}  # lora_int_id -> LoRAAdapter
 
+        # draft-MODEL speculation (E17): rank 0 holds the whole draft model
+        # (proposals happen before the step batch is broadcast)
+        self.draft = None
+        from .draft import DraftModel, is_draft_model_spec
+
+        if is_draft_model_spec(config.speculative_model) and self.rank == 0:
+            self.draft = DraftModel(
+                config.speculative_model.strip("[]"), self.model_config,
+                self.device, self.block_size,
+            )
+
         # pinned H2D staging, double-buffered by step parity: .to(cuda,
         # non_blocking=True) from pageable numpy is silently SYNCHRONOUS
         # (it serializes with the stream, stalling the pipelined step's
@@ -275,6 +286,9 @@ class Worker:
         self.kv_dtype = kv_dtype
         elt = torch.tensor([], dtype=kv_dtype).element_size()
         block_bytes = 2 * mc.num_layers * self.block_size * kv_heads_local * mc.head_dim * elt
+        if self.draft is not None:
+            # the draft's mirrored cache shares block ids: budget its bytes
+            block_bytes += self.draft.cache_bytes_per_block()
 
         if cfg.cache_config.num_gpu_blocks is not None:
             self.num_blocks = cfg.cache_config.num_gpu_blocks
@@ -309,6 +323,8 @@ class Worker:
             )
             for _ in range(mc.num_layers)
         ]
+        if self.draft is not None:
+            self.draft.alloc_cache(self.num_blocks)
         self._maybe_capture_graphs()
         return self.num_blocks
 
