@@ -58,8 +58,8 @@ __device__ inline float cross_half_sum(float v) {
   return v + __shfl_xor(v, 32, 64);
 }
 
-template <int HEAD_DIM, typename CT>
-__global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
+template <int HEAD_DIM, typename CT, int NW>
+__global__ __launch_bounds__(NW * 64, 512 / (NW * 64)) void paged_prefill_mfma_v2_kernel(
     __hip_bfloat16* __restrict__ out,            // [total_q, nheads, HD]
     const __hip_bfloat16* __restrict__ q,        // [total_q, nheads, HD]
     const CT* __restrict__ k_cache,  // [nb, bs, kvh, HD] bf16 | e4m3
@@ -75,6 +75,10 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
   static_assert(HEAD_DIM == 128, "v2 is specialised for HD=128");
   constexpr int KCH = HEAD_DIM / 16;  // QK^T k-chunks (8)
   constexpr int DT = HEAD_DIM / 32;   // 32-col output tiles (4)
+  constexpr int QT = NW * 32;         // q rows per workgroup
+  constexpr int NTHREADS = NW * 64;
+  // staging: each thread covers (s, d8) slots; pass count scales with size
+  constexpr int ST_PASS = (KVT2 * (HEAD_DIM / 8)) / NTHREADS;
 
   const int seq = blockIdx.x;
   const int head = blockIdx.y;
@@ -84,7 +88,7 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
   const int q_start = query_start_loc[seq];
   const int q_len = query_start_loc[seq + 1] - q_start;
   const int seq_len = seq_lens[seq];
-  const int tile_base = q_tile * QTILE2;
+  const int tile_base = q_tile * QT;
   if (tile_base >= q_len) return;
 
   const int tid = threadIdx.x;
@@ -122,7 +126,7 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
 #pragma unroll
   for (int dt = 0; dt < DT; ++dt) acc_o[dt] = f32x16_t{};
 
-  const int wg_rows = min(QTILE2, q_len - tile_base);
+  const int wg_rows = min(QT, q_len - tile_base);
   const int wg_max_pos = seq_len - q_len + tile_base + wg_rows - 1;
   const int kv_limit = min(seq_len, wg_max_pos + 1);
   // last kv position any of this wave's rows may attend to
@@ -131,14 +135,16 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
       seq_len - q_len + tile_base + wave * 32 + max(wave_rows - 1, 0);
   const int wave_q_pos_min = seq_len - q_len + tile_base + wave * 32;
 
-  // ---- staging map: thread covers (s = tid%32 + pass*32, d8 = (tid/32)*8)
-  const int st_s = tid & 31;
-  const int st_d8 = (tid >> 5) * 8;
-  bf16x8_t st_k[2], st_v[2];
+  // ---- staging map: thread t covers slot (t % ROWS_PER_PASS, t / RPP * 8)
+  // with ROWS_PER_PASS = NTHREADS/16 rows per pass (16 8-elem slots per row)
+  constexpr int RPP = NTHREADS / 16;
+  const int st_s = tid % RPP;
+  const int st_d8 = (tid / RPP) * 8;
+  bf16x8_t st_k[ST_PASS], st_v[ST_PASS];
   auto stage_load = [&](int kv_base) {
 #pragma unroll
-    for (int pass = 0; pass < 2; ++pass) {
-      const int s = st_s + pass * 32;
+    for (int pass = 0; pass < ST_PASS; ++pass) {
+      const int s = st_s + pass * RPP;
       const int pos = kv_base + s;
       st_k[pass] = bf16x8_t{};
       st_v[pass] = bf16x8_t{};
@@ -154,8 +160,8 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
   };
   auto stage_write = [&](int buf) {
 #pragma unroll
-    for (int pass = 0; pass < 2; ++pass) {
-      const int s = st_s + pass * 32;
+    for (int pass = 0; pass < ST_PASS; ++pass) {
+      const int s = st_s + pass * RPP;
       const int k_byte = (s * (HEAD_DIM * 2) + st_d8 * 2) ^ ((s & 15) << 4);
       *reinterpret_cast<bf16x8_t*>(
           reinterpret_cast<char*>(k_lds2[buf]) + k_byte) = st_k[pass];
@@ -350,23 +356,32 @@ __global__ __launch_bounds__(512, 1) void paged_prefill_mfma_v2_kernel(
   }
 }
 
+#include <cstdlib>
+
 template <typename CT>
 static void launch_prefill_v2_ct(
     __hip_bfloat16* out, const __hip_bfloat16* q, const CT* kc, const CT* vc,
     const int* bt, const int* qsl, const int* sl, float scale, int nseq,
     int nheads, int kvh, int head_dim, int block_size, int max_blocks,
     int max_query_len, hipStream_t stream) {
-  const int qtiles = (max_query_len + QTILE2 - 1) / QTILE2;
-  dim3 grid(nseq, nheads, qtiles);
-  dim3 block(512);
-  switch (head_dim) {
-    case 128:
-      hipLaunchKernelGGL((paged_prefill_mfma_v2_kernel<128, CT>), grid, block,
-                         0, stream, out, q, kc, vc, bt, qsl, sl, scale,
-                         nheads, kvh, block_size, max_blocks);
-      break;
-    default:
-      abort();
+  if (head_dim != 128) abort();
+  static const int nw = [] {
+    const char* e = getenv("VTA_PREFILL_WAVES");
+    return (e && e[0] == '8') ? 8 : 4;  // 4-wave/128-row default: halves
+                                        // the causal-idle skew per WG
+  }();
+  if (nw == 8) {
+    const int qtiles = (max_query_len + 255) / 256;
+    hipLaunchKernelGGL((paged_prefill_mfma_v2_kernel<128, CT, 8>),
+                       dim3(nseq, nheads, qtiles), dim3(512), 0, stream, out,
+                       q, kc, vc, bt, qsl, sl, scale, nheads, kvh, block_size,
+                       max_blocks);
+  } else {
+    const int qtiles = (max_query_len + 127) / 128;
+    hipLaunchKernelGGL((paged_prefill_mfma_v2_kernel<128, CT, 4>),
+                       dim3(nseq, nheads, qtiles), dim3(256), 0, stream, out,
+                       q, kc, vc, bt, qsl, sl, scale, nheads, kvh, block_size,
+                       max_blocks);
   }
 }
 
