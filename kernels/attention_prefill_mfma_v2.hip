@@ -367,8 +367,9 @@ static void launch_prefill_v2_ct(
   if (head_dim != 128) abort();
   static const int nw = [] {
     const char* e = getenv("VTA_PREFILL_WAVES");
-    return (e && e[0] == '8') ? 8 : 4;  // 4-wave/128-row default: halves
-                                        // the causal-idle skew per WG
+    // measured: 8-wave/256-row wins (317/560 TF vs 302/498 for 4-wave —
+    // the doubled per-CU staging traffic outweighs the causal-skew saving)
+    return (e && e[0] == '4') ? 4 : 8;
   }();
   if (nw == 8) {
     const int qtiles = (max_query_len + 255) / 256;
