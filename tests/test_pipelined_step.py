@@ -92,3 +92,32 @@ def test_pipelined_cumulative_text_matches():
             os.environ.pop("VTA_PIPELINE_MIN", None)
 
     assert run(True) == run(False)
+
+
+def test_pipelined_finish_accounting_exact():
+    """generation_tokens must count each request exactly once (the deferred
+    finish path used to double-schedule finished requests)."""
+    os.environ["VTA_PIPELINE"] = "1"
+    os.environ["VTA_PIPELINE_MIN"] = "1"
+    try:
+        mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+        eng = LLMEngine(EngineConfig(
+            model_config=mc, cache_config=CacheConfig(block_size=16),
+            scheduler_config=SchedulerConfig(max_num_seqs=8,
+                                             max_num_batched_tokens=512),
+            device="cpu", seed=0,
+        ))
+        n_req, m = 6, 9
+        for i in range(n_req):
+            eng.add_request(f"r{i}", None, [30 + i, 40 + i],
+                            SamplingParams(temperature=0.0, max_tokens=m))
+        steps = 0
+        while eng.has_unfinished() and steps < 200:
+            eng.step()
+            steps += 1
+        snap = eng.metrics.snapshot()
+        assert snap["generation_tokens"] == n_req * m, snap
+        assert snap["request_success"] == n_req, snap
+    finally:
+        os.environ.pop("VTA_PIPELINE", None)
+        os.environ.pop("VTA_PIPELINE_MIN", None)

@@ -228,6 +228,11 @@ class LLMEngine:
             if req.guided_state is not None:
                 req.guided_state.advance(token_id)
             self.stop_checker.check_cheap(req, token_id)
+            if req.status.is_finished and req in self.scheduler.running:
+                # de-schedule NOW: deferring finish_request one step let the
+                # scheduler hand this request another slot and the finish
+                # accounting run twice (metrics double-counted)
+                self.scheduler.running.remove(req)
             pairs.append((it, token_id))
         self._deferred = (sched.items, pairs)
         if pt is not None:
@@ -279,19 +284,27 @@ class LLMEngine:
             if out is not None:
                 outputs.append(out)
             if req.status.is_finished:
-                self.scheduler.finish_request(req)
-                m = req.metrics
-                self.metrics.request_success.inc()
-                self.metrics.prompt_tokens.inc(req.num_prompt_tokens)
-                self.metrics.generation_tokens.inc(req.num_output_tokens)
-                if m.first_token_time and m.first_scheduled_time:
-                    self.metrics.ttft.observe(m.first_token_time - m.arrival_time)
-                    if req.num_output_tokens > 1 and m.last_token_time:
-                        self.metrics.time_per_output_token.observe(
-                            (m.last_token_time - m.first_token_time)
-                            / (req.num_output_tokens - 1)
-                        )
+                self._account_finish(req)
         return outputs
+
+    def _account_finish(self, req: Request) -> None:
+        # Idempotent finish bookkeeping: a stop-string finish decided in a
+        # drain can see the request again in the next step's deferred list.
+        if getattr(req, "_finish_accounted", False):
+            return
+        req._finish_accounted = True
+        self.scheduler.finish_request(req)
+        m = req.metrics
+        self.metrics.request_success.inc()
+        self.metrics.prompt_tokens.inc(req.num_prompt_tokens)
+        self.metrics.generation_tokens.inc(req.num_output_tokens)
+        if m.first_token_time and m.first_scheduled_time:
+            self.metrics.ttft.observe(m.first_token_time - m.arrival_time)
+            if req.num_output_tokens > 1 and m.last_token_time:
+                self.metrics.time_per_output_token.observe(
+                    (m.last_token_time - m.first_token_time)
+                    / (req.num_output_tokens - 1)
+                )
 
     # ------------------------------------------------------------------
     def _step_sync(self) -> list[RequestOutput]:
@@ -397,18 +410,7 @@ class LLMEngine:
             if out is not None:
                 outputs.append(out)
             if req.status.is_finished:
-                self.scheduler.finish_request(req)
-                m = req.metrics
-                self.metrics.request_success.inc()
-                self.metrics.prompt_tokens.inc(req.num_prompt_tokens)
-                self.metrics.generation_tokens.inc(req.num_output_tokens)
-                if m.first_token_time and m.first_scheduled_time:
-                    self.metrics.ttft.observe(m.first_token_time - m.arrival_time)
-                    if req.num_output_tokens > 1 and m.last_token_time:
-                        self.metrics.time_per_output_token.observe(
-                            (m.last_token_time - m.first_token_time)
-                            / (req.num_output_tokens - 1)
-                        )
+                self._account_finish(req)
         if pt is not None:
             t3 = time.perf_counter()
             pt["postprocess"] += t3 - t2
