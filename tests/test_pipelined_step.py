@@ -107,6 +107,7 @@ def test_pipelined_finish_accounting_exact():
                                              max_num_batched_tokens=512),
             device="cpu", seed=0,
         ))
+        base = eng.metrics.snapshot()  # prometheus counters are process-global
         n_req, m = 6, 9
         for i in range(n_req):
             eng.add_request(f"r{i}", None, [30 + i, 40 + i],
@@ -116,8 +117,8 @@ def test_pipelined_finish_accounting_exact():
             eng.step()
             steps += 1
         snap = eng.metrics.snapshot()
-        assert snap["generation_tokens"] == n_req * m, snap
-        assert snap["request_success"] == n_req, snap
+        assert snap["generation_tokens"] - base["generation_tokens"] == n_req * m
+        assert snap["request_success"] - base["request_success"] == n_req
     finally:
         os.environ.pop("VTA_PIPELINE", None)
         os.environ.pop("VTA_PIPELINE_MIN", None)
