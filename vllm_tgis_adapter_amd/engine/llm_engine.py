@@ -17,6 +17,7 @@ from .config import EngineConfig
 from .detokenizer import Detokenizer, StopChecker
 from .request import Request
 from .scheduler import Scheduler
+from . import roctx
 from .tokenizer import get_tokenizer
 from .types import LoRARequest, RequestOutput, SamplingParams
 from .worker import Worker
@@ -165,14 +166,15 @@ class LLMEngine:
 
     # ------------------------------------------------------------------
     def step(self) -> list[RequestOutput]:
-        if (
-            self._pipeline_enabled
-            and not self.spec_enabled
-            and (self._deferred is not None
-                 or len(self.scheduler.running) >= self._pipeline_min)
-        ):
-            return self._step_pipelined()
-        return self._step_sync()
+        with roctx.trace_range("engine.step"):
+            if (
+                self._pipeline_enabled
+                and not self.spec_enabled
+                and (self._deferred is not None
+                     or len(self.scheduler.running) >= self._pipeline_min)
+            ):
+                return self._step_pipelined()
+            return self._step_sync()
 
     # ------------------------------------------------------------------
     def _step_pipelined(self) -> list[RequestOutput]:

@@ -18,6 +18,7 @@ import torch
 from .. import ops
 from ..models import get_model
 from ..parallel import get_tp_rank, get_tp_world_size, tp_broadcast_object
+from . import roctx
 from .config import EngineConfig
 from .metadata import AttnMetadata
 from .request import Request
@@ -643,14 +644,16 @@ class Worker:
         _dbg = _os.environ.get("VTA_LAUNCH_TIMING", "0") == "1"
         if _dbg:
             _t1 = _time.perf_counter()
-        logits = self.execute_batch(batch)
+        with roctx.trace_range("worker.forward"):
+            logits = self.execute_batch(batch)
         if _dbg:
             _t2 = _time.perf_counter()
 
         ns = batch["num_sample_rows"]
         if ns and not self._spec_items and not self._prompt_lp_specs:
             sampling_reqs = [it.request for it in self._sampling_items]
-            fused = self.sampler.try_launch_fused(logits[:ns], sampling_reqs)
+            with roctx.trace_range("worker.sample"):
+                fused = self.sampler.try_launch_fused(logits[:ns], sampling_reqs)
             if fused is not None:
                 if _dbg:
                     _t3 = _time.perf_counter()
