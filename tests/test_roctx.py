@@ -25,7 +25,7 @@ def test_enabled_via_env_subprocess():
         "import ctypes, sys\n"
         "from vllm_tgis_adapter_amd.engine import roctx\n"
         "try:\n"
-        "    ctypes.CDLL('libroctx64.so')\n"
+        "    ctypes.CDLL('librocprofiler-sdk-roctx.so')\n"
         "    have_lib = True\n"
         "except OSError:\n"
         "    have_lib = False\n"

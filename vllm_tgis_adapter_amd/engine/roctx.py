@@ -19,7 +19,12 @@ from contextlib import contextmanager
 
 _lib = None
 if os.environ.get("VTA_ROCTX", "0") == "1":
-    for _name in ("libroctx64.so", "libroctx64.so.4"):
+    # rocprofv3 is a rocprofiler-sdk tool: it intercepts the SDK marker
+    # library, NOT the legacy roctracer libroctx64 (whose ranges it records
+    # as zero regions — verified empirically). Prefer the SDK lib.
+    for _name in ("librocprofiler-sdk-roctx.so",
+                  "librocprofiler-sdk-roctx.so.1",
+                  "libroctx64.so", "libroctx64.so.4"):
         try:
             _lib = ctypes.CDLL(_name)
             break
