@@ -45,6 +45,8 @@ def parse_args():
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--dtype", type=str, default="bfloat16")
     p.add_argument("--block-size", type=int, default=16)
+    p.add_argument("--enforce-eager", action="store_true",
+                   help="disable hipGraph decode capture (fallback-path check)")
     p.add_argument("--kv-cache-dtype", type=str, default="auto",
                    choices=["auto", "fp8"],
                    help="KV cache storage (fp8 is an ALTERNATE, non-headline "
@@ -121,6 +123,7 @@ def engine_config(args, device: str, tp: int):
         ),
         device=device,
         tensor_parallel_size=tp,
+        enforce_eager=args.enforce_eager,
         seed=0,
     )
 
@@ -258,6 +261,8 @@ def run_serve(args, device: str) -> None:
         "--kv-cache-dtype", args.kv_cache_dtype,
         "--grpc-port", str(args.grpc_port), "--port", str(args.http_port),
     ]
+    if args.enforce_eager:
+        cmd += ["--enforce-eager"]
     if args.gpus > 1:
         cmd += ["--num-gpus", str(args.gpus)]
     env = dict(os.environ)
