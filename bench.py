@@ -45,6 +45,8 @@ def parse_args():
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--dtype", type=str, default="bfloat16")
     p.add_argument("--block-size", type=int, default=16)
+    p.add_argument("--top-n", type=int, default=0,
+                   help="request top-N token details (logprob path) per stream")
     p.add_argument("--enforce-eager", action="store_true",
                    help="disable hipGraph decode capture (fallback-path check)")
     p.add_argument("--kv-cache-dtype", type=str, default="auto",
@@ -170,6 +172,10 @@ async def serve_drive(args, gen_budget: int, stabilize_s: float) -> dict:
         params = proto.Parameters()
         params.stopping.max_new_tokens = max_new
         params.stopping.min_new_tokens = max_new
+        if args.top_n:
+            params.response.generated_tokens = True
+            params.response.token_logprobs = True
+            params.response.top_n_tokens = args.top_n
         req = proto.SingleGenerationRequest(
             request=proto.GenerationRequest(text=text), params=params)
         seen = 0

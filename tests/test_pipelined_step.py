@@ -122,3 +122,93 @@ def test_pipelined_finish_accounting_exact():
     finally:
         os.environ.pop("VTA_PIPELINE", None)
         os.environ.pop("VTA_PIPELINE_MIN", None)
+
+
+def _run_logprobs(pipeline: bool, n_req=4, max_tokens=12, logprobs=5):
+    """Engine loop with top-N logprob requests; returns per-request
+    (token_ids, [per-pos sorted (id, rank) logprob summaries])."""
+    os.environ["VTA_PIPELINE"] = "1" if pipeline else "0"
+    os.environ["VTA_PIPELINE_MIN"] = "1"
+    try:
+        mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+        eng = LLMEngine(EngineConfig(
+            model_config=mc, cache_config=CacheConfig(block_size=16),
+            scheduler_config=SchedulerConfig(max_num_seqs=8,
+                                             max_num_batched_tokens=512),
+            device="cpu", seed=0,
+        ))
+        for i in range(n_req):
+            eng.add_request(
+                f"r{i}", None, [20 + i, 30 + i, 40 + i],
+                SamplingParams(temperature=0.0, max_tokens=max_tokens,
+                               logprobs=logprobs,
+                               min_tokens=3 if i % 2 else 0),
+            )
+        finals = {}
+        steps = 0
+        while eng.has_unfinished() and steps < 400:
+            for out in eng.step():
+                if out.finished:
+                    finals[out.request_id] = out
+            steps += 1
+        res = {}
+        for rid, out in finals.items():
+            o = out.outputs[0]
+            summaries = []
+            for d in (o.logprobs or []):
+                summaries.append(sorted(
+                    (tid, lp.rank, round(lp.logprob, 4))
+                    for tid, lp in d.items()))
+            res[rid] = (list(o.token_ids), summaries)
+        return res
+    finally:
+        os.environ.pop("VTA_PIPELINE", None)
+        os.environ.pop("VTA_PIPELINE_MIN", None)
+
+
+def test_pipelined_logprobs_match_sync():
+    """Deferred (fused-path) logprob extraction == sync-path extraction:
+    same tokens, same top-N ids/ranks/values at every position, including
+    rows under an active min_tokens EOS ban."""
+    sync = _run_logprobs(False)
+    pipe = _run_logprobs(True)
+    assert set(sync) == set(pipe)
+    for rid in sync:
+        assert pipe[rid][0] == sync[rid][0], rid
+        assert len(pipe[rid][1]) == len(sync[rid][1]), rid
+        assert pipe[rid][1] == sync[rid][1], (rid, sync[rid][1][:2],
+                                              pipe[rid][1][:2])
+
+
+def test_deferred_logprob_extraction_matches_slow_path():
+    """finish_fused's deferred extraction == the original slow-path
+    extraction on identical logits (the slow path is forced with an
+    identity logits processor, which disqualifies the fused launch but
+    leaves values untouched)."""
+    from vllm_tgis_adapter_amd.engine.request import Request
+    from vllm_tgis_adapter_amd.engine.sampler import Sampler
+
+    torch.manual_seed(0)
+    n, vocab = 6, 97
+    logits = torch.randn(n, vocab)
+
+    def mk(i, procs):
+        p = SamplingParams(temperature=0.0, max_tokens=8, logprobs=4,
+                           min_tokens=2 if i % 2 else 0)
+        p.logits_processors = procs
+        r = Request(request_id=f"q{i}", prompt=None,
+                    prompt_token_ids=[1, 2, 3], sampling_params=p)
+        r.eos_token_id = 0
+        return r
+
+    s = Sampler(device="cpu")
+    fused_out = s.sample(logits.clone(), [mk(i, None) for i in range(n)])
+    slow_out = s.sample(logits.clone(),
+                        [mk(i, [lambda ids, row: row]) for i in range(n)])
+    assert fused_out.token_ids == slow_out.token_ids
+    for a, b in zip(fused_out.logprobs, slow_out.logprobs):
+        assert a is not None and b is not None
+        assert set(a) == set(b)
+        for tid in a:
+            assert a[tid].rank == b[tid].rank
+            assert abs(a[tid].logprob - b[tid].logprob) < 1e-4

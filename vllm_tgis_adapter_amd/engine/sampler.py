@@ -23,6 +23,12 @@ from .types import Logprob, PosLogprobs
 
 _NEG_INF = float("-inf")
 
+import os as _os_mod
+
+# deferred (async) logprob extraction on the fused path; VTA_DEFER_LP=0
+# falls back to the synchronous slow-path extraction
+_DEFER_LP = _os_mod.environ.get("VTA_DEFER_LP", "1") == "1"
+
 
 @dataclass
 class SamplerOutput:
@@ -57,6 +63,7 @@ class Sampler:
         # silently synchronous and would stall the pipelined step's launch
         self._pin: dict = {}
         self._pin_flip = 0
+        self._pending_lp = None  # deferred logprob tensors (finish_fused)
 
     def _h2d_list(self, name: str, data: list, dtype: torch.dtype) -> torch.Tensor:
         if self.device != "cuda":
@@ -88,9 +95,13 @@ class Sampler:
 
         Returns the device token-id tensor (caller syncs via .tolist()) when
         every row is greedy or plain temperature sampling — no top-k/top-p,
-        penalties, processors, guided masks or logprob requests — else None.
-        The pipelined engine step uses this to overlap host postprocessing
-        with the next step's GPU work.
+        penalties, processors or guided masks — else None.  Rows that
+        request logprobs/top-N stay ON this path: their log-softmax / top-k
+        / rank tensors are launched asynchronously here and synced later in
+        ``finish_fused`` (the TGIS wire always wants top-N details, and
+        dropping to the sync path would stall the pipelined step).  The
+        pipelined engine step uses this to overlap host postprocessing with
+        the next step's GPU work.
         """
         from .. import ops as _ops
 
@@ -99,16 +110,21 @@ class Sampler:
         any_sampling = False
         ban_rows: list[int] = []  # rows with min_tokens pending: EOS banned
         ban_eos: list[int] = []
+        lp_rows: list[int] = []   # rows that want logprob details (deferred)
         for i, req in enumerate(requests):
             p = req.sampling_params
             if (
                 p.repetition_penalty != 1.0
                 or p.logits_processors
                 or req.guided_state is not None
-                or p.logprobs is not None
             ):
                 simple = False
                 break
+            if p.logprobs is not None:
+                if not _DEFER_LP:  # kill switch: old behavior = slow path
+                    simple = False
+                    break
+                lp_rows.append(i)
             if (p.min_tokens and req.num_output_tokens < p.min_tokens
                     and req.eos_token_id is not None):
                 ban_rows.append(i)
@@ -119,6 +135,7 @@ class Sampler:
                     simple = False
                     break
         if not simple:
+            self._pending_lp = None
             return None
         import os as _os
         import time as _time
@@ -139,21 +156,25 @@ class Sampler:
         if not _ops.native_enabled(logits):
             # CPU / force-reference: plain argmax-or-race, still deferred-sync
             if not any_sampling:
-                return torch.argmax(logits, dim=-1)
-            probs = torch.softmax(
-                logits.float() / torch.tensor(
-                    [max(r.sampling_params.temperature, 1e-6) for r in requests]
-                ).unsqueeze(1),
-                dim=-1,
-            )
-            q = torch.empty_like(probs)
-            for i, req in enumerate(requests):
-                g = self._generator_for(req)
-                if g is None:
-                    q[i].exponential_()
-                else:
-                    q[i].exponential_(generator=g)
-            return torch.argmax(probs / q, dim=-1)
+                out = torch.argmax(logits, dim=-1)
+            else:
+                probs = torch.softmax(
+                    logits.float() / torch.tensor(
+                        [max(r.sampling_params.temperature, 1e-6)
+                         for r in requests]
+                    ).unsqueeze(1),
+                    dim=-1,
+                )
+                q = torch.empty_like(probs)
+                for i, req in enumerate(requests):
+                    g = self._generator_for(req)
+                    if g is None:
+                        q[i].exponential_()
+                    else:
+                        q[i].exponential_(generator=g)
+                out = torch.argmax(probs / q, dim=-1)
+            self._launch_deferred_logprobs(logits, out, lp_rows, requests)
+            return out
         temps = self._h2d_list(
             "temps", [r.sampling_params.temperature for r in requests],
             torch.float32,
@@ -188,7 +209,59 @@ class Sampler:
                       f"empty={d[1]/d[3]*1e3:.2f} kernel-launch={d[2]/d[3]*1e3:.2f}",
                       file=_sys.stderr, flush=True)
                 self._lt2 = [0.0, 0.0, 0.0, 0]
+        self._launch_deferred_logprobs(logits, out, lp_rows, requests)
         return out
+
+    def _launch_deferred_logprobs(
+        self, logits: torch.Tensor, out: torch.Tensor,
+        lp_rows: list[int], requests: list[Request],
+    ) -> None:
+        """Launch log-softmax/top-k/rank work for logprob rows WITHOUT a
+        host sync; ``finish_fused`` syncs and builds the wire dicts.  Uses
+        the post-min_tokens-ban logits — same semantics as the sync path
+        (``sample`` bans EOS before its log_softmax too)."""
+        self._pending_lp = None
+        if not lp_rows:
+            return
+        n, vocab = logits.shape
+        k = min(self.max_logprobs, vocab)
+        if len(lp_rows) == n:
+            rows = logits.float()
+            chosen = out
+        else:
+            idx = self._h2d_list("lp_rows", lp_rows, torch.long)
+            rows = logits.index_select(0, idx).float()
+            chosen = out.index_select(0, idx)
+        lp = torch.log_softmax(rows, dim=-1)
+        topv, topi = torch.topk(lp, k, dim=-1)
+        chosen_lp = lp.gather(1, chosen.unsqueeze(1)).squeeze(1)
+        ranks = (lp > chosen_lp.unsqueeze(1)).sum(dim=-1) + 1
+        self._pending_lp = (
+            topv, topi, chosen_lp, ranks, list(lp_rows),
+            [requests[i].sampling_params.logprobs for i in lp_rows],
+        )
+
+    def finish_fused(self, token_ids: list[int]) -> SamplerOutput:
+        """Sync the deferred logprob tensors (if any) and build the
+        per-position dicts the wire format needs."""
+        out_logprobs: list[Optional[PosLogprobs]] = [None] * len(token_ids)
+        pend = getattr(self, "_pending_lp", None)
+        self._pending_lp = None
+        if pend is not None:
+            topv, topi, chosen_lp, ranks, rows_idx, nums = pend
+            topv = topv.tolist()
+            topi = topi.tolist()
+            chosen_lp = chosen_lp.tolist()
+            ranks = ranks.tolist()
+            for row, i in enumerate(rows_idx):
+                d: PosLogprobs = {}
+                for j in range(min(nums[row], len(topi[row]))):
+                    d[topi[row][j]] = Logprob(logprob=topv[row][j], rank=j + 1)
+                tok = token_ids[i]
+                if tok not in d:
+                    d[tok] = Logprob(logprob=chosen_lp[row], rank=ranks[row])
+                out_logprobs[i] = d
+        return SamplerOutput(token_ids=token_ids, logprobs=out_logprobs)
 
     @torch.inference_mode()
     def sample(self, logits: torch.Tensor, requests: list[Request]) -> SamplerOutput:
@@ -203,7 +276,7 @@ class Sampler:
 
         fused = self.try_launch_fused(logits, requests)
         if fused is not None:
-            return SamplerOutput(token_ids=fused.tolist(), logprobs=[None] * n)
+            return self.finish_fused(fused.tolist())
 
         logits = logits.float()
 

@@ -48,18 +48,18 @@ class _DoneStep:
 
 
 class _PendingStep:
-    """Fused-sampler step in flight; finish() syncs the sampled token ids."""
+    """Fused-sampler step in flight; finish() syncs the sampled token ids
+    (and any deferred logprob tensors the sampler launched alongside)."""
 
-    __slots__ = ("_out",)
+    __slots__ = ("_out", "_sampler")
 
-    def __init__(self, out: torch.Tensor):
+    def __init__(self, out: torch.Tensor, sampler: Sampler):
         self._out = out
+        self._sampler = sampler
 
     def finish(self) -> ExecuteResult:
-        ids = self._out.tolist()
         return ExecuteResult(
-            sampler_output=SamplerOutput(token_ids=ids,
-                                         logprobs=[None] * len(ids)),
+            sampler_output=self._sampler.finish_fused(self._out.tolist()),
             spec_results=[],
         )
 
@@ -670,7 +670,7 @@ class Worker:
                               f"sampler-launch={d[2]/d[3]*1e3:.2f}",
                               file=_sys.stderr, flush=True)
                         self._lt = [0.0, 0.0, 0.0, 0]
-                return _PendingStep(fused)
+                return _PendingStep(fused, self.sampler)
         return _DoneStep(self._execute_finish(batch, logits))
 
     def execute(self, sched: SchedulerOutput) -> ExecuteResult:
