@@ -44,6 +44,9 @@ void launch_paged_prefill_mfma(__hip_bfloat16*, const __hip_bfloat16*,
                                int, int, int, int, int, int, hipStream_t);
 int decode_mfma_num_partitions(int nseq, int kvh, int max_context);
 int gemm_skinny_num_ksplit(int N, int K, int M);
+void launch_logsoftmax_topk(float*, int*, float*, int*,
+                            const __hip_bfloat16*, const long*, int, int,
+                            int, hipStream_t);
 void launch_gemm_tile(__hip_bfloat16*, const __hip_bfloat16*,
                       const __hip_bfloat16*, int, int, int, hipStream_t);
 void launch_xgmi_allreduce(const unsigned long long*, __hip_bfloat16*,
@@ -540,6 +543,32 @@ void sample_argmax(torch::Tensor out, torch::Tensor logits,
   });
 }
 
+void logsoftmax_topk(torch::Tensor topv, torch::Tensor topi,
+                     torch::Tensor chosen_lp, torch::Tensor ranks,
+                     torch::Tensor logits, torch::Tensor chosen) {
+  const int N = logits.size(0);
+  const int V = logits.size(1);
+  const int K = topv.size(1);
+  TORCH_CHECK(logits.scalar_type() == at::ScalarType::BFloat16 &&
+              logits.is_contiguous());
+  TORCH_CHECK(V % 8 == 0, "vocab must be a multiple of 8");
+  TORCH_CHECK(K <= 16, "K <= 16");
+  TORCH_CHECK(topv.scalar_type() == at::ScalarType::Float &&
+              topi.scalar_type() == at::ScalarType::Int &&
+              topv.is_contiguous() && topi.is_contiguous() &&
+              topi.size(1) == K && topv.size(0) == N && topi.size(0) == N);
+  TORCH_CHECK(chosen_lp.scalar_type() == at::ScalarType::Float &&
+              chosen_lp.size(0) == N);
+  TORCH_CHECK(ranks.scalar_type() == at::ScalarType::Int && ranks.size(0) == N);
+  TORCH_CHECK(chosen.scalar_type() == at::ScalarType::Long &&
+              chosen.is_contiguous() && chosen.size(0) == N);
+  launch_logsoftmax_topk(
+      topv.data_ptr<float>(), topi.data_ptr<int>(),
+      chosen_lp.data_ptr<float>(), ranks.data_ptr<int>(),
+      reinterpret_cast<const __hip_bfloat16*>(logits.data_ptr()),
+      chosen.data_ptr<long>(), N, V, K, current_stream());
+}
+
 void mfma_probe(torch::Tensor a, torch::Tensor b, torch::Tensor d, int64_t shape) {
   TORCH_CHECK(a.scalar_type() == at::ScalarType::BFloat16);
   auto* ap = reinterpret_cast<const unsigned short*>(a.data_ptr());
@@ -580,6 +609,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused gate/up skinny GEMM + SiLU-mul, M <= 64 (CDNA4 MFMA)");
   m.def("moe_gemm", &moe_gemm,
         "grouped MoE GEMM over expert segments (CDNA4 MFMA)");
+  m.def("logsoftmax_topk", &logsoftmax_topk,
+        "fused log-softmax + top-K + sampled-token rank (one HBM pass)");
   m.def("sample_argmax", &sample_argmax,
         "fused temperature/gumbel-race sampling + greedy argmax (CDNA4)");
   m.def("lora_bgmv", &lora_bgmv,

@@ -1,0 +1,183 @@
+// Fused log-softmax + top-K + sampled-token rank, one HBM pass (E8).
+//
+// The TGIS wire format wants, per generated token: the top-N token ids
+// with their logprobs and ranks, plus the sampled token's logprob and
+// rank (reference surface: grpc_server.py:701-756 response details).
+// The torch chain for this (float() -> log_softmax -> topk -> gather ->
+// rank count) reads and writes the [N, 128k] row set several times
+// (~0.5 GB of traffic per decode step at batch 512).  This kernel does
+// everything in ONE read of the bf16 logits:
+//
+//   per row (one 256-thread workgroup):
+//     - online max + sum-of-exp (lane-local, rescaled on new max)
+//     - lane-local sorted top-K candidates (insertion against the
+//       running K-th value; K <= 16)
+//     - count of elements strictly greater than the sampled token's
+//       logit (== rank - 1; log-softmax is monotonic so raw logits give
+//       the same ordering)
+//   then an LDS reduction for (max, sumexp) and the rank count, and a
+//   K-round selection over the 256*K LDS candidates for the global
+//   top-K (descending, ties resolved toward the lower candidate slot).
+//
+// Outputs: topv f32 [N,K] (log-softmax values), topi i32 [N,K],
+// chosen_lp f32 [N], ranks i32 [N].
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include "common.h"
+
+#define LSK_THREADS 256
+#define LSK_KMAX 16
+
+// 1/ln(2) for exp2-domain sums; log2 -> ln on the way out.
+#define LSK_LOG2E 1.44269504088896340736f
+#define LSK_LN2 0.69314718055994530942f
+
+typedef __attribute__((ext_vector_type(8))) short lsk_bf16x8;
+
+DEVINLINE float lsk_bf16_bits_to_f32(short b) {
+  return __uint_as_float(((unsigned)(unsigned short)b) << 16);
+}
+
+__global__ __launch_bounds__(LSK_THREADS) void logsoftmax_topk_kernel(
+    float* __restrict__ topv,        // [N, K]
+    int* __restrict__ topi,          // [N, K]
+    float* __restrict__ chosen_lp,   // [N]
+    int* __restrict__ ranks,         // [N]
+    const __hip_bfloat16* __restrict__ logits,  // [N, V]
+    const long* __restrict__ chosen, // [N]
+    const int V, const int K) {
+  const int row = blockIdx.x;
+  const int tid = threadIdx.x;
+  const __hip_bfloat16* x = logits + (long)row * V;
+
+  const float chosen_val = __bfloat162float(x[chosen[row]]);
+
+  // ---- lane-local single pass ------------------------------------------
+  float m = -INFINITY;   // running max (raw logits)
+  float s = 0.f;         // sum of exp2((x - m) * log2e)
+  int gt = 0;            // count of x[j] > chosen_val
+  float cv[LSK_KMAX];    // lane-local top-K values, descending
+  int ci[LSK_KMAX];
+#pragma unroll
+  for (int k = 0; k < LSK_KMAX; ++k) {
+    cv[k] = -INFINITY;
+    ci[k] = -1;
+  }
+  float kth = -INFINITY;  // cv[K-1]
+
+  for (int base = tid * 8; base < V; base += LSK_THREADS * 8) {
+    // V is padded to a multiple of 8 by the launcher contract
+    const lsk_bf16x8 v8 = *reinterpret_cast<const lsk_bf16x8*>(x + base);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int j = base + e;
+      const float xv = lsk_bf16_bits_to_f32(v8[e]);
+      if (xv > m) {
+        s = s * __builtin_amdgcn_exp2f((m - xv) * LSK_LOG2E);
+        m = xv;
+        s += 1.f;
+      } else {
+        s += __builtin_amdgcn_exp2f((xv - m) * LSK_LOG2E);
+      }
+      gt += (xv > chosen_val);
+      if (xv > kth) {
+        // insertion into the sorted lane-local list (rare after warmup)
+        int p = K - 1;
+        while (p > 0 && cv[p - 1] < xv) {
+          cv[p] = cv[p - 1];
+          ci[p] = ci[p - 1];
+          --p;
+        }
+        cv[p] = xv;
+        ci[p] = j;
+        kth = cv[K - 1];
+      }
+    }
+  }
+
+  // ---- (max, sumexp) + rank reduction over the workgroup ----------------
+  __shared__ float red_m[LSK_THREADS];
+  __shared__ float red_s[LSK_THREADS];
+  __shared__ int red_g[LSK_THREADS];
+  red_m[tid] = m;
+  red_s[tid] = s;
+  red_g[tid] = gt;
+  __syncthreads();
+  for (int off = LSK_THREADS / 2; off > 0; off >>= 1) {
+    if (tid < off) {
+      const float m2 = red_m[tid + off];
+      const float s2 = red_s[tid + off];
+      const float mm = fmaxf(red_m[tid], m2);
+      red_s[tid] = red_s[tid] * __builtin_amdgcn_exp2f((red_m[tid] - mm) * LSK_LOG2E) +
+                   s2 * __builtin_amdgcn_exp2f((m2 - mm) * LSK_LOG2E);
+      red_m[tid] = mm;
+      red_g[tid] += red_g[tid + off];
+    }
+    __syncthreads();
+  }
+  const float gmax = red_m[0];
+  // log(sumexp) in ln domain; lse = gmax + ln(sum)
+  const float lse = gmax + __logf(red_s[0]);
+
+  if (tid == 0) {
+    chosen_lp[row] = chosen_val - lse;
+    ranks[row] = red_g[0] + 1;
+  }
+
+  // ---- global top-K: K selection rounds over the LDS candidate pool ----
+  __shared__ float cand_v[LSK_THREADS * LSK_KMAX];
+  __shared__ int cand_i[LSK_THREADS * LSK_KMAX];
+#pragma unroll
+  for (int k = 0; k < LSK_KMAX; ++k) {
+    if (k < K) {
+      cand_v[tid * K + k] = cv[k];
+      cand_i[tid * K + k] = ci[k];
+    }
+  }
+  __syncthreads();
+
+  // one wave selects; others are done (their LDS writes are complete)
+  if (tid >= 64) return;
+  const int pool = LSK_THREADS * K;
+  for (int k = 0; k < K; ++k) {
+    float best = -INFINITY;
+    int best_p = -1;
+    for (int p = tid; p < pool; p += 64) {
+      const float v = cand_v[p];
+      if (v > best || (v == best && p < best_p)) {
+        best = v;
+        best_p = p;
+      }
+    }
+    // wave-local shfl reduce (single wave does the selection)
+    for (int off = 32; off > 0; off >>= 1) {
+      const float ov = __shfl_down(best, off, 64);
+      const int op = __shfl_down(best_p, off, 64);
+      if (ov > best || (ov == best && op != -1 && (best_p == -1 || op < best_p))) {
+        best = ov;
+        best_p = op;
+      }
+    }
+    best = __shfl(best, 0, 64);
+    best_p = __shfl(best_p, 0, 64);
+    if (tid == 0) {
+      topv[(long)row * K + k] = best - lse;
+      topi[(long)row * K + k] = (best_p >= 0) ? cand_i[best_p] : -1;
+    }
+    // retire the winner so the next round finds the next-largest
+    if (tid == 0 && best_p >= 0) cand_v[best_p] = -INFINITY;
+    __builtin_amdgcn_s_waitcnt(0);  // order the LDS retire before next scan
+    __builtin_amdgcn_wave_barrier();
+  }
+}
+
+void launch_logsoftmax_topk(float* topv, int* topi, float* chosen_lp,
+                            int* ranks, const __hip_bfloat16* logits,
+                            const long* chosen, int n, int vocab, int k,
+                            hipStream_t stream) {
+  hipLaunchKernelGGL(logsoftmax_topk_kernel, dim3(n), dim3(LSK_THREADS), 0,
+                     stream, topv, topi, chosen_lp, ranks, logits, chosen,
+                     vocab, k);
+}

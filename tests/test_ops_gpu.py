@@ -406,3 +406,41 @@ def test_gemm_tile(m, n, k):
     got = ops.gemm_tile(x, w)
     ref = (x.float() @ w.float().t())
     assert torch.allclose(got.float(), ref, atol=2e-1, rtol=2e-2)
+
+
+@pytest.mark.gpu
+def test_logsoftmax_topk_matches_torch():
+    """Fused E8 kernel vs plain fp32 torch: log-softmax top-K values,
+    chosen-token logprob, and rank, across ragged magnitudes."""
+    from vllm_tgis_adapter_amd import ops
+
+    torch.manual_seed(7)
+    n, v, k = 37, 128256, 11
+    logits = (torch.randn(n, v, device="cuda") * 4.0).to(torch.bfloat16)
+    chosen = torch.randint(0, v, (n,), device="cuda", dtype=torch.long)
+    # make a few rows adversarial: big spikes + the chosen token on a tie
+    logits[0, 12345] = 40.0
+    logits[1] = -10.0
+    logits[1, 777] = 25.0
+    chosen[1] = 777
+
+    topv, topi, chosen_lp, ranks = ops.logsoftmax_topk(logits, chosen, k)
+    torch.cuda.synchronize()
+
+    lp_ref = torch.log_softmax(logits.float(), dim=-1)
+    ref_v, ref_i = torch.topk(lp_ref, k, dim=-1)
+    ref_chosen = lp_ref.gather(1, chosen.unsqueeze(1)).squeeze(1)
+    ref_rank = (lp_ref > ref_chosen.unsqueeze(1)).sum(dim=-1) + 1
+
+    assert torch.allclose(topv, ref_v, atol=2e-3, rtol=1e-3), \
+        (topv - ref_v).abs().max()
+    assert torch.allclose(chosen_lp, ref_chosen, atol=2e-3, rtol=1e-3)
+    assert torch.equal(ranks.long(), ref_rank)
+    # ids may differ only where values tie at bf16 resolution
+    mism = topi.long() != ref_i
+    if bool(mism.any()):
+        rows, cols = mism.nonzero(as_tuple=True)
+        vals_k = topv[rows, cols]
+        vals_r = ref_v[rows, cols]
+        assert torch.allclose(vals_k, vals_r, atol=2e-3, rtol=1e-3), \
+            "top-K id mismatch beyond tie tolerance"

@@ -223,19 +223,26 @@ class Sampler:
         self._pending_lp = None
         if not lp_rows:
             return
+        from .. import ops as _ops
+
         n, vocab = logits.shape
         k = min(self.max_logprobs, vocab)
         if len(lp_rows) == n:
-            rows = logits.float()
+            rows = logits
             chosen = out
         else:
             idx = self._h2d_list("lp_rows", lp_rows, torch.long)
-            rows = logits.index_select(0, idx).float()
+            rows = logits.index_select(0, idx)
             chosen = out.index_select(0, idx)
-        lp = torch.log_softmax(rows, dim=-1)
-        topv, topi = torch.topk(lp, k, dim=-1)
-        chosen_lp = lp.gather(1, chosen.unsqueeze(1)).squeeze(1)
-        ranks = (lp > chosen_lp.unsqueeze(1)).sum(dim=-1) + 1
+        if _ops.logsoftmax_topk_usable(rows):
+            # one-HBM-pass HIP kernel (E8): log-softmax + top-K + rank
+            topv, topi, chosen_lp, ranks = _ops.logsoftmax_topk(
+                rows, chosen, k)
+        else:
+            lp = torch.log_softmax(rows.float(), dim=-1)
+            topv, topi = torch.topk(lp, k, dim=-1)
+            chosen_lp = lp.gather(1, chosen.unsqueeze(1)).squeeze(1)
+            ranks = (lp > chosen_lp.unsqueeze(1)).sum(dim=-1) + 1
         self._pending_lp = (
             topv, topi, chosen_lp, ranks, list(lp_rows),
             [requests[i].sampling_params.logprobs for i in lp_rows],

@@ -347,6 +347,31 @@ def sample_argmax(
     _C.sample_argmax(out, logits, temps, noise)
 
 
+def logsoftmax_topk_usable(logits: torch.Tensor) -> bool:
+    return (native_enabled(logits) and logits.dtype == torch.bfloat16
+            and logits.shape[1] % 8 == 0)
+
+
+def logsoftmax_topk(
+    logits: torch.Tensor,  # [N, V] bf16 (contiguous)
+    chosen: torch.Tensor,  # [N] int64 sampled/actual token per row
+    k: int,
+):
+    """Fused log-softmax + top-K + chosen-token logprob/rank in one HBM
+    pass (E8 wire details).  Returns (topv f32 [N,K], topi i32 [N,K],
+    chosen_lp f32 [N], ranks i32 [N]); launch only, no host sync."""
+    assert logsoftmax_topk_usable(logits), "logsoftmax_topk is the GPU path"
+    n = logits.shape[0]
+    dev = logits.device
+    topv = torch.empty((n, k), dtype=torch.float32, device=dev)
+    topi = torch.empty((n, k), dtype=torch.int32, device=dev)
+    chosen_lp = torch.empty((n,), dtype=torch.float32, device=dev)
+    ranks = torch.empty((n,), dtype=torch.int32, device=dev)
+    _C.logsoftmax_topk(topv, topi, chosen_lp, ranks,
+                       logits.contiguous(), chosen.contiguous())
+    return topv, topi, chosen_lp, ranks
+
+
 def moe_gemm(
     x: torch.Tensor,        # [T, K] rows sorted by expert
     w: torch.Tensor,        # [E, 2N, K] (gated) or [E, N, K]
