@@ -1,0 +1,39 @@
+"""Microbench: fused logsoftmax_topk kernel vs the torch chain (E8)."""
+import time
+
+import torch
+
+from vllm_tgis_adapter_amd import ops
+
+
+def main():
+    n, v, k = 512, 128256, 11
+    logits = (torch.randn(n, v, device="cuda") * 3).to(torch.bfloat16)
+    chosen = torch.randint(0, v, (n,), device="cuda", dtype=torch.long)
+
+    def torch_chain():
+        lp = torch.log_softmax(logits.float(), dim=-1)
+        topv, topi = torch.topk(lp, k, dim=-1)
+        clp = lp.gather(1, chosen.unsqueeze(1)).squeeze(1)
+        ranks = (lp > clp.unsqueeze(1)).sum(dim=-1) + 1
+        return topv, topi, clp, ranks
+
+    def fused():
+        return ops.logsoftmax_topk(logits, chosen, k)
+
+    for name, fn in (("torch_chain", torch_chain), ("fused_kernel", fused)):
+        for _ in range(5):
+            fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        iters = 50
+        for _ in range(iters):
+            fn()
+        torch.cuda.synchronize()
+        ms = (time.perf_counter() - t0) / iters * 1e3
+        gb = n * v * 2 / 1e9
+        print(f"{name}: {ms:.3f} ms  ({gb/ms*1e3:.2f} GB/s effective read)")
+
+
+if __name__ == "__main__":
+    main()
