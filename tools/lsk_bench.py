@@ -1,5 +1,9 @@
 """Microbench: fused logsoftmax_topk kernel vs the torch chain (E8)."""
+import sys
 import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 import torch
 
