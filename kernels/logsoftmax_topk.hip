@@ -54,18 +54,24 @@ __global__ __launch_bounds__(LSK_THREADS) void logsoftmax_topk_kernel(
 
   const float chosen_val = __bfloat162float(x[chosen[row]]);
 
+  // Per-lane top-K candidates live in LDS (lane-strided, sorted desc).
+  // NOTE: a register array here would be dynamically indexed by the
+  // insertion sort, which the compiler lowers to per-lane SCRATCH memory
+  // — measured 50x off the HBM wall.  LDS handles dynamic indexing.
+  __shared__ float cand_v[LSK_THREADS * LSK_KMAX];
+  __shared__ int cand_i[LSK_THREADS * LSK_KMAX];
+  float* lane_v = cand_v + tid * K;
+  int* lane_i = cand_i + tid * K;
+  for (int k = 0; k < K; ++k) {
+    lane_v[k] = -INFINITY;
+    lane_i[k] = -1;
+  }
+
   // ---- lane-local single pass ------------------------------------------
   float m = -INFINITY;   // running max (raw logits)
   float s = 0.f;         // sum of exp2((x - m) * log2e)
   int gt = 0;            // count of x[j] > chosen_val
-  float cv[LSK_KMAX];    // lane-local top-K values, descending
-  int ci[LSK_KMAX];
-#pragma unroll
-  for (int k = 0; k < LSK_KMAX; ++k) {
-    cv[k] = -INFINITY;
-    ci[k] = -1;
-  }
-  float kth = -INFINITY;  // cv[K-1]
+  float kth = -INFINITY;  // lane_v[K-1], kept in a register
 
   for (int base = tid * 8; base < V; base += LSK_THREADS * 8) {
     // V is padded to a multiple of 8 by the launcher contract
@@ -83,16 +89,17 @@ __global__ __launch_bounds__(LSK_THREADS) void logsoftmax_topk_kernel(
       }
       gt += (xv > chosen_val);
       if (xv > kth) {
-        // insertion into the sorted lane-local list (rare after warmup)
+        // insertion into the sorted lane-local LDS list (rare at steady
+        // state: expected ~K·ln(V/threads/K) inserts per lane)
         int p = K - 1;
-        while (p > 0 && cv[p - 1] < xv) {
-          cv[p] = cv[p - 1];
-          ci[p] = ci[p - 1];
+        while (p > 0 && lane_v[p - 1] < xv) {
+          lane_v[p] = lane_v[p - 1];
+          lane_i[p] = lane_i[p - 1];
           --p;
         }
-        cv[p] = xv;
-        ci[p] = j;
-        kth = cv[K - 1];
+        lane_v[p] = xv;
+        lane_i[p] = j;
+        kth = lane_v[K - 1];
       }
     }
   }
@@ -127,15 +134,6 @@ __global__ __launch_bounds__(LSK_THREADS) void logsoftmax_topk_kernel(
   }
 
   // ---- global top-K: K selection rounds over the LDS candidate pool ----
-  __shared__ float cand_v[LSK_THREADS * LSK_KMAX];
-  __shared__ int cand_i[LSK_THREADS * LSK_KMAX];
-#pragma unroll
-  for (int k = 0; k < LSK_KMAX; ++k) {
-    if (k < K) {
-      cand_v[tid * K + k] = cv[k];
-      cand_i[tid * K + k] = ci[k];
-    }
-  }
   __syncthreads();
 
   // one wave selects; others are done (their LDS writes are complete)
