@@ -68,25 +68,29 @@ __global__ __launch_bounds__(LSK_THREADS) void logsoftmax_topk_kernel(
   }
 
   // ---- lane-local single pass ------------------------------------------
-  float m = -INFINITY;   // running max (raw logits)
-  float s = 0.f;         // sum of exp2((x - m) * log2e)
+  // Two independent (max, sumexp) accumulator pairs break the serial
+  // exp-chain; 4 b128 loads in flight per outer iteration hide HBM
+  // latency (the grid is only ~2 WGs/CU at batch 512, so per-wave MLP is
+  // the only latency cover available).
+  float m0 = -INFINITY, s0 = 0.f;
+  float m1 = -INFINITY, s1 = 0.f;
   int gt = 0;            // count of x[j] > chosen_val
   float kth = -INFINITY;  // lane_v[K-1], kept in a register
 
-  for (int base = tid * 8; base < V; base += LSK_THREADS * 8) {
-    // V is padded to a multiple of 8 by the launcher contract
-    const lsk_bf16x8 v8 = *reinterpret_cast<const lsk_bf16x8*>(x + base);
+  const int stride = LSK_THREADS * 8;
+
+  auto process8 = [&](const lsk_bf16x8& v8, int base) {
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       const int j = base + e;
       const float xv = lsk_bf16_bits_to_f32(v8[e]);
-      if (xv > m) {
-        s = s * __builtin_amdgcn_exp2f((m - xv) * LSK_LOG2E);
-        m = xv;
-        s += 1.f;
-      } else {
-        s += __builtin_amdgcn_exp2f((xv - m) * LSK_LOG2E);
-      }
+      float& m = (e & 1) ? m1 : m0;
+      float& s = (e & 1) ? s1 : s0;
+      const float mn = fmaxf(m, xv);
+      // branchless online softmax: always one rescale + one term
+      s = s * __builtin_amdgcn_exp2f((m - mn) * LSK_LOG2E) +
+          __builtin_amdgcn_exp2f((xv - mn) * LSK_LOG2E);
+      m = mn;
       gt += (xv > chosen_val);
       if (xv > kth) {
         // insertion into the sorted lane-local LDS list (rare at steady
@@ -102,7 +106,29 @@ __global__ __launch_bounds__(LSK_THREADS) void logsoftmax_topk_kernel(
         kth = lane_v[K - 1];
       }
     }
+  };
+
+  for (int base = tid * 8; base < V; base += stride * 4) {
+    // V is a multiple of 8 (launcher contract), so every 8-chunk is
+    // either fully in range or fully out
+    const bool h1 = base + stride < V;
+    const bool h2 = base + 2 * stride < V;
+    const bool h3 = base + 3 * stride < V;
+    lsk_bf16x8 c0, c1, c2, c3;
+    c0 = *reinterpret_cast<const lsk_bf16x8*>(x + base);
+    if (h1) c1 = *reinterpret_cast<const lsk_bf16x8*>(x + base + stride);
+    if (h2) c2 = *reinterpret_cast<const lsk_bf16x8*>(x + base + 2 * stride);
+    if (h3) c3 = *reinterpret_cast<const lsk_bf16x8*>(x + base + 3 * stride);
+    process8(c0, base);
+    if (h1) process8(c1, base + stride);
+    if (h2) process8(c2, base + 2 * stride);
+    if (h3) process8(c3, base + 3 * stride);
   }
+
+  // fold the two accumulator pairs
+  const float m = fmaxf(m0, m1);
+  const float s = s0 * __builtin_amdgcn_exp2f((m0 - m) * LSK_LOG2E) +
+                  s1 * __builtin_amdgcn_exp2f((m1 - m) * LSK_LOG2E);
 
   // ---- (max, sumexp) + rank reduction over the workgroup ----------------
   __shared__ float red_m[LSK_THREADS];

@@ -348,8 +348,9 @@ def sample_argmax(
 
 
 def logsoftmax_topk_usable(logits: torch.Tensor) -> bool:
+    # V >= 2048 guarantees every lane feeds both accumulator pairs
     return (native_enabled(logits) and logits.dtype == torch.bfloat16
-            and logits.shape[1] % 8 == 0)
+            and logits.shape[1] % 8 == 0 and logits.shape[1] >= 2048)
 
 
 def logsoftmax_topk(
