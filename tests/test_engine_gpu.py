@@ -3,6 +3,8 @@
 
 from __future__ import annotations
 
+import os
+
 import pytest
 import torch
 
@@ -179,3 +181,50 @@ def test_draft_model_spec_gpu_exact():
     spec, spec_steps = run("tiny-llama")
     assert spec == base
     assert spec_steps < base_steps  # identical draft: full acceptance
+
+
+@pytest.mark.gpu
+def test_llama_engine_logprobs_gpu():
+    """Deferred fused-path logprob extraction (E8 kernel) at engine level:
+    GPU run with top-N logprobs matches the torch reference extraction on
+    the same model (VTA_DEFER_LP=0 forces the sync slow path)."""
+    import subprocess
+    import sys
+
+    code = """
+import os, json
+from vllm_tgis_adapter_amd.engine import (
+    EngineConfig, LLMEngine, ModelConfig, SamplingParams)
+from vllm_tgis_adapter_amd.engine.config import CacheConfig, SchedulerConfig
+mc = ModelConfig.from_model_arg("llama-1b", dtype="bfloat16")
+eng = LLMEngine(EngineConfig(
+    model_config=mc, cache_config=CacheConfig(block_size=16, num_gpu_blocks=512),
+    scheduler_config=SchedulerConfig(max_num_seqs=8, max_num_batched_tokens=2048),
+    device="cuda"))
+for i in range(4):
+    eng.add_request(f"r{i}", None, list(range(100 + i, 132 + i)),
+                    SamplingParams(temperature=0.0, max_tokens=8, logprobs=5))
+res = {}
+steps = 0
+while eng.has_unfinished():
+    for out in eng.step():
+        if out.finished:
+            o = out.outputs[0]
+            res[out.request_id] = [
+                sorted((int(t), int(lp.rank), round(float(lp.logprob), 2))
+                       for t, lp in d.items())
+                for d in o.logprobs]
+    steps += 1
+    assert steps < 100
+print("RESULT" + json.dumps(res, sort_keys=True))
+"""
+    outs = {}
+    for mode in ("1", "0"):
+        env = dict(os.environ)
+        env["VTA_DEFER_LP"] = mode
+        p = subprocess.run([sys.executable, "-c", code], env=env,
+                           capture_output=True, text=True, timeout=600)
+        assert p.returncode == 0, p.stderr[-2000:]
+        line = [ln for ln in p.stdout.splitlines() if ln.startswith("RESULT")][-1]
+        outs[mode] = line[len("RESULT"):]
+    assert outs["1"] == outs["0"]
