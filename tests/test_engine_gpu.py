@@ -185,46 +185,57 @@ def test_draft_model_spec_gpu_exact():
 
 @pytest.mark.gpu
 def test_llama_engine_logprobs_gpu():
-    """Deferred fused-path logprob extraction (E8 kernel) at engine level:
-    GPU run with top-N logprobs matches the torch reference extraction on
-    the same model (VTA_DEFER_LP=0 forces the sync slow path)."""
-    import subprocess
-    import sys
+    """Deferred fused-path logprob extraction (E8 kernel) at engine level.
 
-    code = """
-import os, json
-from vllm_tgis_adapter_amd.engine import (
-    EngineConfig, LLMEngine, ModelConfig, SamplingParams)
-from vllm_tgis_adapter_amd.engine.config import CacheConfig, SchedulerConfig
-mc = ModelConfig.from_model_arg("llama-1b", dtype="bfloat16")
-eng = LLMEngine(EngineConfig(
-    model_config=mc, cache_config=CacheConfig(block_size=16, num_gpu_blocks=512),
-    scheduler_config=SchedulerConfig(max_num_seqs=8, max_num_batched_tokens=2048),
-    device="cuda"))
-for i in range(4):
-    eng.add_request(f"r{i}", None, list(range(100 + i, 132 + i)),
-                    SamplingParams(temperature=0.0, max_tokens=8, logprobs=5))
-res = {}
-steps = 0
-while eng.has_unfinished():
-    for out in eng.step():
-        if out.finished:
-            o = out.outputs[0]
-            res[out.request_id] = [
-                sorted((int(t), int(lp.rank), round(float(lp.logprob), 2))
-                       for t, lp in d.items())
-                for d in o.logprobs]
-    steps += 1
-    assert steps < 100
-print("RESULT" + json.dumps(res, sort_keys=True))
-"""
-    outs = {}
-    for mode in ("1", "0"):
-        env = dict(os.environ)
-        env["VTA_DEFER_LP"] = mode
-        p = subprocess.run([sys.executable, "-c", code], env=env,
-                           capture_output=True, text=True, timeout=600)
-        assert p.returncode == 0, p.stderr[-2000:]
-        line = [ln for ln in p.stdout.splitlines() if ln.startswith("RESULT")][-1]
-        outs[mode] = line[len("RESULT"):]
-    assert outs["1"] == outs["0"]
+    Cross-process defer-vs-sync exactness is NOT tested: prefill GEMMs run
+    hipBLASLt stream-k kernels whose atomics make run-to-run logits differ
+    at bf16 ULP level (documented in docs/STATUS.md), which reshuffles the
+    near-tied top-N tail.  Instead this checks within-run invariants that
+    would catch any row/step misalignment in the deferred path: for greedy
+    decode the sampled token must be the rank-1 top entry of its own
+    position's dict, ranks must be consistent with descending values, and
+    every position must carry a dict.  (Value correctness of the kernel vs
+    torch is covered by test_ops_gpu.py::test_logsoftmax_topk_matches_torch;
+    defer-vs-slow-path parity is covered exactly on CPU in
+    test_pipelined_step.py.)
+    """
+    from vllm_tgis_adapter_amd.engine import (
+        EngineConfig, LLMEngine, ModelConfig, SamplingParams,
+    )
+    from vllm_tgis_adapter_amd.engine.config import CacheConfig, SchedulerConfig
+
+    mc = ModelConfig.from_model_arg("llama-1b", dtype="bfloat16")
+    eng = LLMEngine(EngineConfig(
+        model_config=mc,
+        cache_config=CacheConfig(block_size=16, num_gpu_blocks=512),
+        scheduler_config=SchedulerConfig(max_num_seqs=8,
+                                         max_num_batched_tokens=2048),
+        device="cuda",
+    ))
+    n_req, max_tokens, top_n = 4, 8, 5
+    for i in range(n_req):
+        eng.add_request(f"r{i}", None, list(range(100 + i, 132 + i)),
+                        SamplingParams(temperature=0.0, max_tokens=max_tokens,
+                                       logprobs=top_n))
+    finished = {}
+    steps = 0
+    while eng.has_unfinished():
+        for out in eng.step():
+            if out.finished:
+                finished[out.request_id] = out
+        steps += 1
+        assert steps < 100
+    assert len(finished) == n_req
+    for rid, out in finished.items():
+        o = out.outputs[0]
+        assert len(o.logprobs) == len(o.token_ids) == max_tokens, rid
+        for pos, (tok, d) in enumerate(zip(o.token_ids, o.logprobs)):
+            assert d and tok in d, (rid, pos)
+            # greedy: the sampled token IS the argmax -> rank 1 and the
+            # highest value in the dict
+            assert d[tok].rank == 1, (rid, pos, d[tok].rank)
+            entries = sorted(d.values(), key=lambda lp: -lp.logprob)
+            assert entries[0].logprob == d[tok].logprob
+            ranks = [lp.rank for lp in entries]
+            assert ranks == sorted(ranks), (rid, pos, ranks)
+            assert len(d) <= top_n + 1
