@@ -165,3 +165,36 @@ def test_draft_model_spec_exact_with_divergent_draft():
     base, _ = _run_engine(None, [3, 4, 5], seed=7)
     spec, _ = _run_engine("tiny-llama", [3, 4, 5], seed=7)
     assert spec == base
+
+
+def test_spec_with_logprobs_falls_back_cleanly():
+    """Requests with logprobs are spec-ineligible by design (spec.eligible
+    excludes them: multi-token verify does not produce per-position top-N);
+    they must still decode correctly alongside an enabled speculator, one
+    token per step, with a complete dict at every position."""
+    from vllm_tgis_adapter_amd.engine import (
+        EngineConfig, LLMEngine, ModelConfig, SamplingParams,
+    )
+    from vllm_tgis_adapter_amd.engine.config import (
+        CacheConfig, SchedulerConfig,
+    )
+
+    mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+    eng = LLMEngine(EngineConfig(
+        model_config=mc, cache_config=CacheConfig(block_size=16),
+        scheduler_config=SchedulerConfig(max_num_seqs=8,
+                                         max_num_batched_tokens=512),
+        device="cpu", seed=0, speculative_model="[ngram]"))
+    eng.add_request("r0", None, [5, 6, 7, 8] * 6,
+                    SamplingParams(temperature=0.0, max_tokens=12, logprobs=3))
+    finals = {}
+    steps = 0
+    while eng.has_unfinished() and steps < 100:
+        for out in eng.step():
+            if out.finished:
+                finals[out.request_id] = out
+        steps += 1
+    o = finals["r0"].outputs[0]
+    assert len(o.logprobs) == len(o.token_ids) == 12
+    for tok, d in zip(o.token_ids, o.logprobs):
+        assert d and tok in d and d[tok].rank == 1
