@@ -40,6 +40,7 @@ DEVINLINE float lsk_bf16_bits_to_f32(short b) {
   return __uint_as_float(((unsigned)(unsigned short)b) << 16);
 }
 
+template <int KT>
 __global__ __launch_bounds__(LSK_THREADS) void logsoftmax_topk_kernel(
     float* __restrict__ topv,        // [N, K]
     int* __restrict__ topi,          // [N, K]
@@ -47,24 +48,28 @@ __global__ __launch_bounds__(LSK_THREADS) void logsoftmax_topk_kernel(
     int* __restrict__ ranks,         // [N]
     const __hip_bfloat16* __restrict__ logits,  // [N, V]
     const long* __restrict__ chosen, // [N]
-    const int V, const int K) {
+    const int V, const int kout) {  // kout <= KT: output columns/rounds
+  constexpr int K = KT;
   const int row = blockIdx.x;
   const int tid = threadIdx.x;
   const __hip_bfloat16* x = logits + (long)row * V;
 
   const float chosen_val = __bfloat162float(x[chosen[row]]);
 
-  // Per-lane top-K candidates live in LDS (lane-strided, sorted desc).
-  // NOTE: a register array here would be dynamically indexed by the
-  // insertion sort, which the compiler lowers to per-lane SCRATCH memory
-  // — measured 50x off the HBM wall.  LDS handles dynamic indexing.
-  __shared__ float cand_v[LSK_THREADS * LSK_KMAX];
-  __shared__ int cand_i[LSK_THREADS * LSK_KMAX];
-  float* lane_v = cand_v + tid * K;
-  int* lane_i = cand_i + tid * K;
+  // Per-lane top-K candidates in REGISTERS with a fully static-index
+  // insertion cascade.  (v1 used a dynamically-indexed register array —
+  // the compiler lowered it to per-lane scratch, 50x off the wall; v2
+  // kept the list in LDS — the data-dependent shift loop cost dependent
+  // LDS round-trips on virtually every wave-step, since with V/lane=500
+  // SOME lane inserts almost every iteration.  The cascade is ~6 VALU
+  // per slot with no memory traffic, so the wave-divergent insert costs
+  // ~K*6 VALU instead of ~K LDS round trips.)
+  float cv[K];
+  int ci[K];
+#pragma unroll
   for (int k = 0; k < K; ++k) {
-    lane_v[k] = -INFINITY;
-    lane_i[k] = -1;
+    cv[k] = -INFINITY;
+    ci[k] = -1;
   }
 
   // ---- lane-local single pass ------------------------------------------
@@ -93,17 +98,21 @@ __global__ __launch_bounds__(LSK_THREADS) void logsoftmax_topk_kernel(
       m = mn;
       gt += (xv > chosen_val);
       if (xv > kth) {
-        // insertion into the sorted lane-local LDS list (rare at steady
-        // state: expected ~K·ln(V/threads/K) inserts per lane)
-        int p = K - 1;
-        while (p > 0 && lane_v[p - 1] < xv) {
-          lane_v[p] = lane_v[p - 1];
-          lane_i[p] = lane_i[p - 1];
-          --p;
+        // static-index insertion cascade into the descending list:
+        // position k takes its upper neighbour if xv ranks above it,
+        // xv itself if this is the insertion point, else keeps its value
+#pragma unroll
+        for (int k = K - 1; k >= 1; --k) {
+          const bool shift = xv > cv[k - 1];
+          const bool place = !shift && (xv > cv[k]);
+          cv[k] = shift ? cv[k - 1] : (place ? xv : cv[k]);
+          ci[k] = shift ? ci[k - 1] : (place ? j : ci[k]);
         }
-        lane_v[p] = xv;
-        lane_i[p] = j;
-        kth = lane_v[K - 1];
+        if (xv > cv[0]) {
+          ci[0] = j;
+          cv[0] = xv;
+        }
+        kth = cv[K - 1];
       }
     }
   };
@@ -160,12 +169,19 @@ __global__ __launch_bounds__(LSK_THREADS) void logsoftmax_topk_kernel(
   }
 
   // ---- global top-K: K selection rounds over the LDS candidate pool ----
+  __shared__ float cand_v[LSK_THREADS * K];
+  __shared__ int cand_i[LSK_THREADS * K];
+#pragma unroll
+  for (int k = 0; k < K; ++k) {
+    cand_v[tid * K + k] = cv[k];
+    cand_i[tid * K + k] = ci[k];
+  }
   __syncthreads();
 
   // one wave selects; others are done (their LDS writes are complete)
   if (tid >= 64) return;
   const int pool = LSK_THREADS * K;
-  for (int k = 0; k < K; ++k) {
+  for (int k = 0; k < kout; ++k) {
     float best = -INFINITY;
     int best_p = -1;
     for (int p = tid; p < pool; p += 64) {
@@ -187,8 +203,8 @@ __global__ __launch_bounds__(LSK_THREADS) void logsoftmax_topk_kernel(
     best = __shfl(best, 0, 64);
     best_p = __shfl(best_p, 0, 64);
     if (tid == 0) {
-      topv[(long)row * K + k] = best - lse;
-      topi[(long)row * K + k] = (best_p >= 0) ? cand_i[best_p] : -1;
+      topv[(long)row * kout + k] = best - lse;
+      topi[(long)row * kout + k] = (best_p >= 0) ? cand_i[best_p] : -1;
     }
     // retire the winner so the next round finds the next-largest
     if (tid == 0 && best_p >= 0) cand_v[best_p] = -INFINITY;
@@ -201,7 +217,16 @@ void launch_logsoftmax_topk(float* topv, int* topi, float* chosen_lp,
                             int* ranks, const __hip_bfloat16* logits,
                             const long* chosen, int n, int vocab, int k,
                             hipStream_t stream) {
-  hipLaunchKernelGGL(logsoftmax_topk_kernel, dim3(n), dim3(LSK_THREADS), 0,
-                     stream, topv, topi, chosen_lp, ranks, logits, chosen,
-                     vocab, k);
+  // K is a template constant so the insertion cascade stays static-index;
+  // the sampler always asks for max_logprobs = 11 (validation caps top_n
+  // at 10, +1 for the sampled token).  16 covers the headroom case.
+  if (k == 11) {
+    hipLaunchKernelGGL(logsoftmax_topk_kernel<11>, dim3(n),
+                       dim3(LSK_THREADS), 0, stream, topv, topi, chosen_lp,
+                       ranks, logits, chosen, vocab, 11);
+  } else if (k <= 16) {
+    hipLaunchKernelGGL(logsoftmax_topk_kernel<16>, dim3(n),
+                       dim3(LSK_THREADS), 0, stream, topv, topi, chosen_lp,
+                       ranks, logits, chosen, vocab, k);
+  }
 }
