@@ -76,3 +76,22 @@ def test_metrics(http_base):
     assert "tgis_amd:request_success" in body
     assert "tgis_amd:generation_tokens" in body
     assert "tgis_amd:time_to_first_token_seconds" in body
+
+
+def test_completions_logprobs(http_base):
+    """OpenAI completions `logprobs: N` returns the tokens /
+    token_logprobs / top_logprobs block (reference surface: vLLM's
+    OpenAI app behind `http.py`)."""
+    resp = json.load(_post_json(
+        f"{http_base}/v1/completions",
+        {"model": "m", "prompt": "hello", "max_tokens": 4, "logprobs": 3},
+    ))
+    lp = resp["choices"][0]["logprobs"]
+    assert lp is not None
+    assert len(lp["tokens"]) == 4
+    assert len(lp["token_logprobs"]) == 4
+    assert all(isinstance(v, float) for v in lp["token_logprobs"])
+    assert len(lp["top_logprobs"]) == 4
+    for top in lp["top_logprobs"]:
+        assert 1 <= len(top) <= 4  # up to N + the sampled token
+        assert all(isinstance(v, float) for v in top.values())

@@ -176,6 +176,28 @@ async def build_http_server(args: "argparse.Namespace", engine: "AsyncLLMEngine"
         except Exception as e:
             return JSONResponse(status_code=500, content={"error": {"message": str(e)}})
 
+        tokenizer = None
+        if params.logprobs is not None:
+            tokenizer = await engine.get_tokenizer()
+
+        def lp_object(o):
+            # OpenAI completions logprobs block (reference surface is
+            # vLLM's OpenAI app: tokens / token_logprobs / top_logprobs)
+            if params.logprobs is None or o.logprobs is None:
+                return None
+            toks, tok_lps, top_lps = [], [], []
+            for tok_id, d in zip(o.token_ids, o.logprobs):
+                tok = tokenizer.convert_ids_to_tokens([tok_id])[0]
+                toks.append(tok)
+                lp = d.get(tok_id)
+                tok_lps.append(float(lp.logprob) if lp is not None else None)
+                top_lps.append({
+                    tokenizer.convert_ids_to_tokens([t])[0]: float(v.logprob)
+                    for t, v in d.items()
+                })
+            return {"tokens": toks, "token_logprobs": tok_lps,
+                    "top_logprobs": top_lps}
+
         choices = []
         prompt_tokens = completion_tokens = 0
         for i, out in enumerate(results):
@@ -184,7 +206,7 @@ async def build_http_server(args: "argparse.Namespace", engine: "AsyncLLMEngine"
                 "index": i,
                 "text": o.text,
                 "finish_reason": o.finish_reason,
-                "logprobs": None,
+                "logprobs": lp_object(o),
             })
             prompt_tokens += len(out.prompt_token_ids)
             completion_tokens += len(o.token_ids)
