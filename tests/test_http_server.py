@@ -95,3 +95,24 @@ def test_completions_logprobs(http_base):
     for top in lp["top_logprobs"]:
         assert 1 <= len(top) <= 4  # up to N + the sampled token
         assert all(isinstance(v, float) for v in top.values())
+
+
+def test_chat_completions_streaming(http_base):
+    """`stream: true` on chat completions yields OpenAI
+    chat.completion.chunk SSE events: a role-priming first delta, content
+    deltas, and a terminating [DONE]."""
+    r = _post_json(
+        f"{http_base}/v1/chat/completions",
+        {"model": "m", "messages": [{"role": "user", "content": "hi"}],
+         "max_tokens": 4, "stream": True},
+    )
+    body = r.read().decode()
+    events = [line for line in body.split("\n\n") if line.startswith("data: ")]
+    assert events[-1] == "data: [DONE]"
+    chunks = [json.loads(e[len("data: "):]) for e in events[:-1]]
+    assert chunks[0]["object"] == "chat.completion.chunk"
+    assert chunks[0]["choices"][0]["delta"].get("role") == "assistant"
+    text = "".join(c["choices"][0]["delta"].get("content", "")
+                   for c in chunks)
+    assert len(text) > 0
+    assert any(c["choices"][0]["finish_reason"] for c in chunks)

@@ -235,8 +235,37 @@ async def build_http_server(args: "argparse.Namespace", engine: "AsyncLLMEngine"
             params = _sampling_params_from_completion(body, model_config.max_model_len)
         except ValueError as e:
             return JSONResponse(status_code=400, content={"error": {"message": str(e)}})
-        params.output_kind = RequestOutputKind.FINAL_ONLY
         rid = f"chatcmpl-{uuid.uuid4().hex}"
+        if bool(body.get("stream", False)):
+            params.output_kind = RequestOutputKind.DELTA
+
+            async def chat_event_stream():
+                def chunk(delta, finish=None):
+                    return "data: " + json.dumps({
+                        "id": rid,
+                        "object": "chat.completion.chunk",
+                        "created": created,
+                        "model": model_name,
+                        "choices": [{"index": 0, "delta": delta,
+                                     "finish_reason": finish}],
+                    }) + "\n\n"
+
+                yield chunk({"role": "assistant", "content": ""})
+                async for out in engine.generate(
+                        prompt=prompt, sampling_params=params,
+                        request_id=rid):
+                    delta = out.outputs[0]
+                    if delta.text or delta.finish_reason is None:
+                        yield chunk({"content": delta.text},
+                                    delta.finish_reason)
+                    elif delta.finish_reason is not None:
+                        yield chunk({}, delta.finish_reason)
+                yield "data: [DONE]\n\n"
+
+            return StreamingResponse(chat_event_stream(),
+                                     media_type="text/event-stream")
+
+        params.output_kind = RequestOutputKind.FINAL_ONLY
         final = None
         async for out in engine.generate(prompt=prompt, sampling_params=params, request_id=rid):
             final = out
