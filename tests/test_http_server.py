@@ -116,3 +116,16 @@ def test_chat_completions_streaming(http_base):
                    for c in chunks)
     assert len(text) > 0
     assert any(c["choices"][0]["finish_reason"] for c in chunks)
+
+
+def test_completions_n_choices(http_base):
+    """OpenAI `n` produces n choices per prompt (independent engine
+    requests; seeded requests get derived per-choice seeds)."""
+    resp = json.load(_post_json(
+        f"{http_base}/v1/completions",
+        {"model": "m", "prompt": ["a", "b"], "max_tokens": 3, "n": 2,
+         "temperature": 0.8, "seed": 7},
+    ))
+    assert len(resp["choices"]) == 4
+    assert {c["index"] for c in resp["choices"]} == {0, 1, 2, 3}
+    assert resp["usage"]["completion_tokens"] == 12

@@ -123,12 +123,28 @@ async def build_http_server(args: "argparse.Namespace", engine: "AsyncLLMEngine"
         if isinstance(prompts, str):
             prompts = [prompts]
         stream = bool(body.get("stream", False))
-        n_prompts = len(prompts)
+        n_choices = int(body.get("n", 1) or 1)
+        if n_choices < 1 or n_choices > 64:
+            return JSONResponse(status_code=400, content={
+                "error": {"message": "n must be in [1, 64]"}})
+        # n > 1: one engine request per choice (seeded requests get
+        # distinct derived seeds so choices differ, as in the reference)
+        gen_prompts = [p for p in prompts for _ in range(n_choices)]
+        n_prompts = len(gen_prompts)
         try:
             params = _sampling_params_from_completion(body, model_config.max_model_len)
         except ValueError as e:
             return JSONResponse(status_code=400, content={"error": {"message": str(e)}})
         base_id = request.headers.get("x-request-id") or f"cmpl-{uuid.uuid4().hex}"
+
+        def params_for(i: int):
+            if n_choices == 1 or params.seed is None:
+                return params
+            import copy as _copy
+
+            p2 = _copy.copy(params)
+            p2.seed = params.seed + (i % n_choices)
+            return p2
 
         if stream:
             params.output_kind = RequestOutputKind.DELTA
@@ -137,10 +153,10 @@ async def build_http_server(args: "argparse.Namespace", engine: "AsyncLLMEngine"
                 gens = [
                     engine.generate(
                         prompt=p,
-                        sampling_params=params,
+                        sampling_params=params_for(i),
                         request_id=f"cmpl-{base_id}-{i}",
                     )
-                    for i, p in enumerate(prompts)
+                    for i, p in enumerate(gen_prompts)
                 ]
                 from .engine.types import merge_async_iterators
 
@@ -167,12 +183,13 @@ async def build_http_server(args: "argparse.Namespace", engine: "AsyncLLMEngine"
 
         async def run(i: int, p: str):
             async for out in engine.generate(
-                prompt=p, sampling_params=params, request_id=f"cmpl-{base_id}-{i}"
+                prompt=p, sampling_params=params_for(i),
+                request_id=f"cmpl-{base_id}-{i}"
             ):
                 results[i] = out
 
         try:
-            await asyncio.gather(*(run(i, p) for i, p in enumerate(prompts)))
+            await asyncio.gather(*(run(i, p) for i, p in enumerate(gen_prompts)))
         except Exception as e:
             return JSONResponse(status_code=500, content={"error": {"message": str(e)}})
 
