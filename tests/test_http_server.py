@@ -129,3 +129,11 @@ def test_completions_n_choices(http_base):
     assert len(resp["choices"]) == 4
     assert {c["index"] for c in resp["choices"]} == {0, 1, 2, 3}
     assert resp["usage"]["completion_tokens"] == 12
+
+
+def test_completions_echo(http_base):
+    resp = json.load(_post_json(
+        f"{http_base}/v1/completions",
+        {"model": "m", "prompt": "abc", "max_tokens": 2, "echo": True},
+    ))
+    assert resp["choices"][0]["text"].startswith("abc")

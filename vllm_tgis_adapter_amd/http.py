@@ -215,13 +215,14 @@ async def build_http_server(args: "argparse.Namespace", engine: "AsyncLLMEngine"
             return {"tokens": toks, "token_logprobs": tok_lps,
                     "top_logprobs": top_lps}
 
+        echo = bool(body.get("echo", False))
         choices = []
         prompt_tokens = completion_tokens = 0
         for i, out in enumerate(results):
             o = out.outputs[0]
             choices.append({
                 "index": i,
-                "text": o.text,
+                "text": (gen_prompts[i] + o.text) if echo else o.text,
                 "finish_reason": o.finish_reason,
                 "logprobs": lp_object(o),
             })
