@@ -81,3 +81,17 @@ def test_bench_engine_tp2_torchrun_cpu():
     assert out["metric"] == "engine_output_tokens_per_s"
     assert out["n_gpus"] == 2
     assert out["value"] > 0
+
+
+def test_bench_serve_top_n_dry_run():
+    """--top-n drives the wire logprob/token-detail path end to end."""
+    proc = subprocess.run(
+        [sys.executable, str(REPO / "bench.py"),
+         "--steps", "6", "--warmup", "2", "--batch", "8", "--top-n", "5",
+         "--grpc-port", "18919", "--http-port", "18920",
+         "--server-log", "/tmp/test_bench_serve_topn_server.log"],
+        capture_output=True, text=True, timeout=300, cwd=str(REPO),
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    out = json.loads(proc.stdout.strip().splitlines()[-1])
+    assert out["value"] > 0
