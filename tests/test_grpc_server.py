@@ -325,3 +325,68 @@ def test_guided_bad_schema_aborts(grpc_client):
     with pytest.raises(_grpc.RpcError) as e:
         grpc_client.Generate(req, timeout=30)
     assert e.value.code() == _grpc.StatusCode.INVALID_ARGUMENT
+
+
+def test_stop_reason_token_limit_mapping():
+    """Server-capped max_tokens -> TOKEN_LIMIT, client-set -> MAX_TOKENS
+    (reference: grpc_server.py:676-678,787-798 semantics)."""
+    from types import SimpleNamespace
+
+    from vllm_tgis_adapter_amd.grpc.convert import resolve_stop
+
+    out = SimpleNamespace(finish_reason="length", stop_reason=None)
+    assert resolve_stop(out, capped=True, deadline_hit=False,
+                        tokenizer=None).reason == SR.TOKEN_LIMIT
+    assert resolve_stop(out, capped=False, deadline_hit=False,
+                        tokenizer=None).reason == SR.MAX_TOKENS
+    unfinished = SimpleNamespace(finish_reason=None, stop_reason=None)
+    assert resolve_stop(unfinished, capped=False, deadline_hit=True,
+                        tokenizer=None).reason == SR.TIME_LIMIT
+    assert resolve_stop(unfinished, capped=False, deadline_hit=False,
+                        tokenizer=None).reason == SR.NOT_FINISHED
+
+
+def test_add_special_tokens_env_toggle(grpc_client, monkeypatch):
+    """ADD_SPECIAL_TOKENS (default true) controls BOS in Tokenize
+    (reference: grpc_server.py:88-91,850)."""
+    from vllm_tgis_adapter_amd.grpc import service as svc
+
+    req = proto.BatchedTokenizeRequest(
+        model_id="m", requests=[proto.TokenizeRequest(text="abc")])
+    with_special = grpc_client.Tokenize(req, timeout=30).responses[0].token_count
+    monkeypatch.setattr(svc, "ADD_SPECIAL_TOKENS", False)
+    without = grpc_client.Tokenize(req, timeout=30).responses[0].token_count
+    assert with_special == without + 1  # BOS dropped
+
+
+def test_lora_unique_ids_start_at_1000001():
+    """Reference invariant: adapter unique-id counter starts at 1000001
+    (adapters.py:59)."""
+    import asyncio
+
+    from types import SimpleNamespace
+
+    from vllm_tgis_adapter_amd.grpc.adapters import (
+        AdapterStore, validate_adapters,
+    )
+
+    store = AdapterStore(cache_path="tests/fixtures/adapters", adapters={})
+    assert store.next_unique_id == 1000001
+
+    from vllm_tgis_adapter_amd.engine.types import LoRARequest
+
+    class _Models:
+        lora_requests: dict = {}
+
+        async def load_lora_adapter(self, lora_name, lora_path,
+                                    base_model_name=None):
+            self.lora_requests[lora_name] = LoRARequest(
+                lora_name=lora_name, lora_int_id=1, lora_path=lora_path)
+            return None
+
+    req = SimpleNamespace(adapter_id="tiny-lora", prefix_id="",
+                          model_id="m")
+    kwargs = asyncio.run(validate_adapters(req, store, _Models()))
+    assert "lora_request" in kwargs
+    # the unique-id counter advanced exactly once from its 1000001 start
+    assert store.next_unique_id == 1000002
