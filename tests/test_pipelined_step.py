@@ -212,3 +212,71 @@ def test_deferred_logprob_extraction_matches_slow_path():
         for tid in a:
             assert a[tid].rank == b[tid].rank
             assert abs(a[tid].logprob - b[tid].logprob) < 1e-4
+
+
+def _run_churn(pipeline: bool):
+    """Deterministic add/abort churn: requests join and get aborted at
+    fixed step indices while others run with stops/logprobs/min_tokens."""
+    os.environ["VTA_PIPELINE"] = "1" if pipeline else "0"
+    os.environ["VTA_PIPELINE_MIN"] = "1"
+    try:
+        mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+        eng = LLMEngine(EngineConfig(
+            model_config=mc, cache_config=CacheConfig(block_size=16),
+            scheduler_config=SchedulerConfig(max_num_seqs=16,
+                                             max_num_batched_tokens=512),
+            device="cpu", seed=0,
+        ))
+
+        def add(i):
+            sp = SamplingParams(
+                temperature=0.0, max_tokens=16,
+                stop=["zq"] if i % 3 == 0 else None,
+                logprobs=2 if i % 4 == 0 else None,
+                min_tokens=3 if i % 5 == 0 else 0,
+            )
+            eng.add_request(f"c{i}", None, [15 + i, 25 + i, 35 + i], sp)
+
+        for i in range(6):
+            add(i)
+        finals = {}
+        steps = 0
+        next_new = 6
+        while eng.has_unfinished() and steps < 300:
+            if steps == 4:
+                eng.abort_request("c1")
+            if steps == 7:
+                eng.abort_request("c3")     # may already be finished — no-op
+            if steps in (5, 9) and next_new < 10:
+                add(next_new)
+                next_new += 1
+            for out in eng.step():
+                if out.finished:
+                    o = out.outputs[0]
+                    lp = None
+                    if o.logprobs:
+                        lp = [sorted((t, d[t].rank) for t in d)
+                              for d in o.logprobs]
+                    finals[out.request_id] = (
+                        o.text, tuple(o.token_ids), o.finish_reason,
+                        o.stop_reason, lp)
+            steps += 1
+        return finals
+    finally:
+        os.environ.pop("VTA_PIPELINE", None)
+        os.environ.pop("VTA_PIPELINE_MIN", None)
+
+
+def test_pipelined_churn_matches_sync():
+    """Pipelined == sync under add/abort churn with stops, logprobs and
+    min_tokens mixed in (aborted requests may surface one step later in
+    the pipelined path, so only non-aborted outcomes are compared)."""
+    sync = _run_churn(False)
+    pipe = _run_churn(True)
+    aborted = ("c1", "c3")  # abort-vs-finish ordering differs by design
+    sync_done = {k: v for k, v in sync.items() if k not in aborted}
+    pipe_done = {k: v for k, v in pipe.items() if k not in aborted}
+    assert set(sync_done) == set(pipe_done)
+    for rid in sync_done:
+        assert pipe_done[rid] == sync_done[rid], (
+            rid, sync_done[rid], pipe_done[rid])
