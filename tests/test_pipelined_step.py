@@ -57,10 +57,10 @@ def test_pipelined_matches_sync(stops):
     sync = _run(False, stops)
     pipe = _run(True, stops)
     assert set(sync) == set(pipe)
-    # FINAL_ONLY-equivalent comparison: same finish reasons; cumulative
-    # output identical (DELTA streams concatenate to the same text)
     for rid in sync:
-        assert pipe[rid][1:] == sync[rid][1:], (rid, sync[rid], pipe[rid])
+        # full comparison INCLUDING text — a drain-phase skip once dropped
+        # the final token's text from every cheap (length/EOS) finish
+        assert pipe[rid] == sync[rid], (rid, sync[rid], pipe[rid])
 
 
 def test_pipelined_cumulative_text_matches():

@@ -267,13 +267,19 @@ class LLMEngine:
         if d is None:
             return []
         items, pairs = d
+        from .request import RequestStatus
+
         for it, token_id in pairs:
             req = it.request
-            if req.status.is_finished:
+            if req.status == RequestStatus.FINISHED_ABORTED:
                 # aborted since; detok state no longer matters
                 continue
             new_text = self.detokenizer.append_token(req, token_id)
-            self.stop_checker.check_text(req, new_text)
+            if not req.status.is_finished:
+                self.stop_checker.check_text(req, new_text)
+            # length/EOS finishes decided in the sync-append phase still
+            # need this (their final) token's text in the output — skipping
+            # them dropped the last text piece from every cheap finish
 
         outputs: list[RequestOutput] = []
         seen = set()
