@@ -214,7 +214,7 @@ def test_deferred_logprob_extraction_matches_slow_path():
             assert abs(a[tid].logprob - b[tid].logprob) < 1e-4
 
 
-def _run_churn(pipeline: bool):
+def _run_churn(pipeline: bool, lengths=None):
     """Deterministic add/abort churn: requests join and get aborted at
     fixed step indices while others run with stops/logprobs/min_tokens."""
     os.environ["VTA_PIPELINE"] = "1" if pipeline else "0"
@@ -229,9 +229,11 @@ def _run_churn(pipeline: bool):
         ))
 
         def add(i):
+            mt = (lengths or {}).get(i, 16)
             sp = SamplingParams(
-                temperature=0.0, max_tokens=16,
-                stop=["zq"] if i % 3 == 0 else None,
+                temperature=0.0, max_tokens=mt,
+                stop=(["zq"] if i % 3 == 0 else
+                      (["@@", "AA"] if i % 3 == 1 else None)),
                 logprobs=2 if i % 4 == 0 else None,
                 min_tokens=3 if i % 5 == 0 else 0,
             )
@@ -265,6 +267,22 @@ def _run_churn(pipeline: bool):
     finally:
         os.environ.pop("VTA_PIPELINE", None)
         os.environ.pop("VTA_PIPELINE_MIN", None)
+
+
+def test_pipelined_churn_varied_lengths_matches_sync():
+    """Second churn schedule: staggered max_tokens so finishes land on
+    many different steps, plus stop strings that actually FIRE (the tiny
+    model's greedy output repeats the prompt byte, so "@@"/"AA" hit)."""
+    lens = {0: 5, 1: 9, 2: 13, 3: 16, 4: 7, 5: 21, 6: 11, 7: 6, 8: 18, 9: 4}
+    sync = _run_churn(False, lens)
+    pipe = _run_churn(True, lens)
+    aborted = ("c1", "c3")
+    sync_done = {k: v for k, v in sync.items() if k not in aborted}
+    pipe_done = {k: v for k, v in pipe.items() if k not in aborted}
+    assert set(sync_done) == set(pipe_done)
+    for rid in sync_done:
+        assert pipe_done[rid] == sync_done[rid], (
+            rid, sync_done[rid], pipe_done[rid])
 
 
 def test_pipelined_churn_matches_sync():
