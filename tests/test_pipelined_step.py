@@ -298,3 +298,39 @@ def test_pipelined_churn_matches_sync():
     for rid in sync_done:
         assert pipe_done[rid] == sync_done[rid], (
             rid, sync_done[rid], pipe_done[rid])
+
+
+def test_pipelined_seeded_sampling_matches_sync():
+    """Seeded (temperature>0) sampling draws identically in pipelined and
+    sync stepping: the per-request generators advance once per step in
+    the same order on both paths."""
+    def run(pipeline):
+        os.environ["VTA_PIPELINE"] = "1" if pipeline else "0"
+        os.environ["VTA_PIPELINE_MIN"] = "1"
+        try:
+            mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+            eng = LLMEngine(EngineConfig(
+                model_config=mc, cache_config=CacheConfig(block_size=16),
+                scheduler_config=SchedulerConfig(max_num_seqs=8,
+                                                 max_num_batched_tokens=512),
+                device="cpu", seed=0,
+            ))
+            for i in range(5):
+                eng.add_request(
+                    f"s{i}", None, [40 + i, 50 + i],
+                    SamplingParams(temperature=0.9, seed=1234 + i,
+                                   max_tokens=10))
+            finals = {}
+            steps = 0
+            while eng.has_unfinished() and steps < 200:
+                for out in eng.step():
+                    if out.finished:
+                        finals[out.request_id] = tuple(
+                            out.outputs[0].token_ids)
+                steps += 1
+            return finals
+        finally:
+            os.environ.pop("VTA_PIPELINE", None)
+            os.environ.pop("VTA_PIPELINE_MIN", None)
+
+    assert run(False) == run(True)
