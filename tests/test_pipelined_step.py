@@ -334,3 +334,40 @@ def test_pipelined_seeded_sampling_matches_sync():
             os.environ.pop("VTA_PIPELINE_MIN", None)
 
     assert run(False) == run(True)
+
+
+def test_pipelined_prefix_caching_matches_sync():
+    """Prefix caching (shared prompt prefixes, block reuse) composes with
+    pipelined stepping: outputs equal the sync path's."""
+    def run(pipeline):
+        os.environ["VTA_PIPELINE"] = "1" if pipeline else "0"
+        os.environ["VTA_PIPELINE_MIN"] = "1"
+        try:
+            mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+            eng = LLMEngine(EngineConfig(
+                model_config=mc,
+                cache_config=CacheConfig(block_size=16,
+                                         enable_prefix_caching=True),
+                scheduler_config=SchedulerConfig(max_num_seqs=8,
+                                                 max_num_batched_tokens=512),
+                device="cpu", seed=0,
+            ))
+            shared = list(range(60, 60 + 40))  # > 2 blocks of shared prefix
+            for i in range(6):
+                eng.add_request(
+                    f"p{i}", None, shared + [100 + i],
+                    SamplingParams(temperature=0.0, max_tokens=12))
+            finals = {}
+            steps = 0
+            while eng.has_unfinished() and steps < 200:
+                for out in eng.step():
+                    if out.finished:
+                        o = out.outputs[0]
+                        finals[out.request_id] = (o.text, tuple(o.token_ids))
+                steps += 1
+            return finals
+        finally:
+            os.environ.pop("VTA_PIPELINE", None)
+            os.environ.pop("VTA_PIPELINE_MIN", None)
+
+    assert run(False) == run(True)
