@@ -371,3 +371,49 @@ def test_pipelined_prefix_caching_matches_sync():
             os.environ.pop("VTA_PIPELINE_MIN", None)
 
     assert run(False) == run(True)
+
+
+def test_pipelined_guided_matches_sync():
+    """Guided (regex-constrained) requests take the slow sampler inside
+    execute_begin but still flow through the pipelined drain; outputs and
+    constraint satisfaction match the sync path."""
+    import re
+
+    from vllm_tgis_adapter_amd.engine.types import StructuredOutputsParams
+
+    def run(pipeline):
+        os.environ["VTA_PIPELINE"] = "1" if pipeline else "0"
+        os.environ["VTA_PIPELINE_MIN"] = "1"
+        try:
+            mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+            eng = LLMEngine(EngineConfig(
+                model_config=mc, cache_config=CacheConfig(block_size=16),
+                scheduler_config=SchedulerConfig(max_num_seqs=8,
+                                                 max_num_batched_tokens=512),
+                device="cpu", seed=0,
+            ))
+            for i in range(4):
+                sp = SamplingParams(temperature=0.0, max_tokens=10)
+                if i % 2 == 0:
+                    sp.structured_outputs = StructuredOutputsParams(
+                        regex="[0-9]{4}")
+                eng.add_request(f"g{i}", None, [70 + i, 80 + i], sp)
+            finals = {}
+            steps = 0
+            while eng.has_unfinished() and steps < 200:
+                for out in eng.step():
+                    if out.finished:
+                        o = out.outputs[0]
+                        finals[out.request_id] = (o.text, tuple(o.token_ids),
+                                                  o.finish_reason)
+                steps += 1
+            return finals
+        finally:
+            os.environ.pop("VTA_PIPELINE", None)
+            os.environ.pop("VTA_PIPELINE_MIN", None)
+
+    sync = run(False)
+    pipe = run(True)
+    assert sync == pipe
+    for rid in ("g0", "g2"):
+        assert re.fullmatch(r"[0-9]{4}", sync[rid][0]), sync[rid]
