@@ -36,7 +36,7 @@ async def start_grpc_server(
     generation = TextGenerationService(
         engine, args, health_servicer, stop_event, model_handler
     )
-    await generation.post_init()
+    await generation.finish_boot()
     add_generation_service(server, generation)
 
     service_names = (

@@ -1,11 +1,14 @@
 """fmaas.GenerationService implementation.
 
-Wire behavior follows the reference service (grpc/grpc_server.py) point by
-point: request-id/correlation-id handling, proto→SamplingParams conversion,
-tokenize+truncate+max_tokens capping, stop-reason mapping, token detail /
-logprob / rank / top-N conversion, N+1 streaming message invariant, deadline
-(time_limit_millis) aborts reporting TIME_LIMIT, dead-engine process stop,
-GPU OOM → RESOURCE_EXHAUSTED.
+Wire behavior matches the reference adapter's gRPC service
+(/root/reference/src/vllm_tgis_adapter/grpc/grpc_server.py) observable
+point by observable point — correlation-id request ids, proto→engine
+parameter conversion, tokenize/truncate/max_tokens capping, stop-reason
+mapping, token detail/logprob/rank/top-N blocks, the N+1 streaming
+message invariant, `time_limit_millis` deadline aborts reporting
+TIME_LIMIT, dead-engine process stop and GPU OOM → RESOURCE_EXHAUSTED —
+while the implementation is this engine's own (pure conversion helpers
+live in convert.py; the engine API underneath is ours, not vLLM's).
 """
 
 from __future__ import annotations
@@ -41,57 +44,59 @@ if TYPE_CHECKING:
     import argparse
 
     from ..engine.async_engine import AsyncLLMEngine
-    from ..engine.types import CompletionOutput, PosLogprobs, RequestOutput
+    from ..engine.types import CompletionOutput, RequestOutput
 
 logger = init_logger(__name__)
 
+# BOS behavior on tokenization, env-togglable like the reference
 ADD_SPECIAL_TOKENS: bool = os.getenv("ADD_SPECIAL_TOKENS", "true").lower() not in (
     "0",
     "false",
 )
-CORRELATION_ID_HEADER = "x-correlation-id"
+_CORR_HEADER = "x-correlation-id"
 
 
-def with_default(value, default):
-    return value if value else default
+def rpc_guard(handler):
+    """Wrap an RPC handler (unary or streaming): on any failure, check for
+    a dead engine (=> trip the server's stop event so the process exits
+    instead of limping), map GPU OOM to RESOURCE_EXHAUSTED, and log."""
 
+    async def _failed(service, context, exc):
+        eng = service.engine
+        if eng.errored and not eng.is_running:
+            service.stop_event.set()
+        if not isinstance(exc, grpc.aio.AbortError):
+            from torch.cuda import OutOfMemoryError
 
-async def _handle_exception(e: Exception, func, service, context) -> None:
-    """Shared per-RPC error hook: dead engine => stop the whole server."""
-    engine = service.engine
-    if engine.errored and not engine.is_running:
-        service.stop_event.set()
-    if not isinstance(e, grpc.aio.AbortError):
-        from torch.cuda import OutOfMemoryError
+            if isinstance(exc, OutOfMemoryError):
+                logger.exception("%s caused GPU OOM error", handler.__name__)
+                await context.abort(StatusCode.RESOURCE_EXHAUSTED, str(exc))
+            logger.exception("%s failed", handler.__name__)
+        raise exc
 
-        if isinstance(e, OutOfMemoryError):
-            logger.exception("%s caused GPU OOM error", func.__name__)
-            await context.abort(StatusCode.RESOURCE_EXHAUSTED, str(e))
-        logger.exception("%s failed", func.__name__)
-    raise e
+    if inspect.isasyncgenfunction(handler):
 
-
-def log_rpc_handler_errors(func):
-    if inspect.isasyncgenfunction(func):
-
-        async def wrapped(self, request, context):
+        async def guarded(self, request, context):
             try:
-                async for item in func(self, request, context):
-                    yield item
-            except Exception as e:
-                await _handle_exception(e, func, self, context)
+                async for msg in handler(self, request, context):
+                    yield msg
+            except Exception as exc:
+                await _failed(self, context, exc)
+
     else:
 
-        async def wrapped(self, request, context):
+        async def guarded(self, request, context):
             try:
-                return await func(self, request, context)
-            except Exception as e:
-                await _handle_exception(e, func, self, context)
+                return await handler(self, request, context)
+            except Exception as exc:
+                await _failed(self, context, exc)
 
-    return wrapped
+    return guarded
 
 
-class TextGenerationService:
+class GenerationServicer:
+    """The four TGIS RPCs over this repo's engine."""
+
     SERVICE_NAME = proto.SERVICE_NAME
 
     def __init__(
@@ -105,312 +110,297 @@ class TextGenerationService:
         self.engine = engine
         self.stop_event = stop_event
         self.model_handler = model_handler
-        self.config = None  # set in post_init
-        self.max_max_new_tokens = args.max_new_tokens
-        self.skip_special_tokens = not args.output_special_tokens
-        self.default_include_stop_seqs = args.default_include_stop_seqs
-        self.disable_prompt_logprobs = args.disable_prompt_logprobs
-        adapter_cache_path = args.adapter_cache or args.prefix_store_path
+        self.config = None  # filled by finish_boot()
+        self.server_token_cap = args.max_new_tokens
+        self.strip_special_tokens = not args.output_special_tokens
+        self.include_stops_default = args.default_include_stop_seqs
+        self.prompt_logprobs_off = args.disable_prompt_logprobs
+        cache_dir = args.adapter_cache or args.prefix_store_path
         self.adapter_store = (
-            AdapterStore(cache_path=adapter_cache_path, adapters={})
-            if adapter_cache_path
-            else None
+            AdapterStore(cache_path=cache_dir, adapters={}) if cache_dir else None
         )
-        self.health_servicer = health_servicer
+        self.health = health_servicer
 
-    async def post_init(self) -> None:
+    async def finish_boot(self) -> None:
+        """Engine config is only available once the engine is up; flip the
+        health service to SERVING after it lands (liveness contract)."""
         self.config = await self.engine.get_model_config()
-        self.health_servicer.set(self.SERVICE_NAME, 1)  # SERVING
+        self.health.set(self.SERVICE_NAME, 1)  # SERVING
 
-    # ------------------------------------------------------------------
+    # -- request identity ----------------------------------------------
     @staticmethod
-    def request_id(context) -> str:
-        metadata = context.invocation_metadata()
-        if metadata:
-            correlation_id = dict(metadata).get(CORRELATION_ID_HEADER)
-            if correlation_id:
-                return correlation_id
+    def _rid_from(context) -> str:
+        md = context.invocation_metadata()
+        if md:
+            cid = dict(md).get(_CORR_HEADER)
+            if cid:
+                return cid
         return uuid.uuid4().hex
 
-    # ------------------------------------------------------------------
-    @log_rpc_handler_errors
+    # -- Generate (unary, batched) --------------------------------------
+    @rpc_guard
     async def Generate(self, request, context):
-        start = time.time()
-        request_id = self.request_id(context)
-        kwargs = await self._validate_adapters(request, context)
+        rid = self._rid_from(context)
+        extra = await self._adapter_kwargs(request, context)
         tokenizer = await self.engine.get_tokenizer()
-
-        sampling_params, deadline = await self._validate_and_convert_params(
+        params, deadline = await self._params_from_proto(
             request.params, tokenizer, context
         )
-        sampling_params.output_kind = RequestOutputKind.FINAL_ONLY
-        truncate_input_tokens = with_default(request.params.truncate_input_tokens, None)
-        request_count = len(request.requests)
-
-        generators = []
-        max_is_token_limit = [False] * request_count
+        params.output_kind = RequestOutputKind.FINAL_ONLY
+        trunc = request.params.truncate_input_tokens or None
+        n_req = len(request.requests)
         headers = dict(context.invocation_metadata() or ())
-        trace_headers = self._trace_headers(headers)
-        for i, req in enumerate(request.requests):
-            input_ids, max_is_token_limit[i] = await self._validate_prompt_and_tokenize(
-                sampling_params, truncate_input_tokens, req.text, tokenizer, context
+        otel = self._trace_headers(headers)
+
+        streams = []
+        capped = []
+        for i, sub in enumerate(request.requests):
+            ids, was_capped = await self._prepare_prompt(
+                params, trunc, sub.text, tokenizer, context
             )
-            sub_id = f"{request_id}-{i}"
-            logs.set_correlation_id(sub_id, headers.get(CORRELATION_ID_HEADER))
-            generators.append(
+            capped.append(was_capped)
+            sub_id = f"{rid}-{i}"
+            logs.set_correlation_id(sub_id, headers.get(_CORR_HEADER))
+            streams.append(
                 self.engine.generate(
-                    prompt={"prompt": req.text, "prompt_token_ids": input_ids},
-                    sampling_params=sampling_params,
+                    prompt={"prompt": sub.text, "prompt_token_ids": ids},
+                    sampling_params=params,
                     request_id=sub_id,
-                    trace_headers=trace_headers,
-                    **kwargs,
+                    trace_headers=otel,
+                    **extra,
                 )
             )
 
-        # deadline watchdog: abort in-flight engine requests at the deadline
-        time_limit_reached = False
-        watchdog = None
+        deadline_hit = False
+        timer = None
         if deadline is not None:
-
-            async def _expire():
-                nonlocal time_limit_reached
+            # one watchdog aborts every sub-request at the deadline so the
+            # engine frees their KV instead of running to completion
+            async def fire():
+                nonlocal deadline_hit
                 await asyncio.sleep(max(0.0, deadline - time.time()))
-                time_limit_reached = True
-                for j in range(request_count):
-                    await self.engine.abort(f"{request_id}-{j}")
+                deadline_hit = True
+                for j in range(n_req):
+                    await self.engine.abort(f"{rid}-{j}")
 
-            watchdog = asyncio.get_event_loop().create_task(_expire())
+            timer = asyncio.get_event_loop().create_task(fire())
 
-        responses: list = [None] * request_count
+        finals: list = [None] * n_req
         try:
-            async for i, res in merge_async_iterators(*generators):
-                responses[i] = res
+            async for i, out in merge_async_iterators(*streams):
+                finals[i] = out
         finally:
-            if watchdog is not None:
-                watchdog.cancel()
+            if timer is not None:
+                timer.cancel()
 
-        resp_options = request.params.response
-        for i in range(request_count):
-            res = responses[i]
+        opts = request.params.response
+        for i, res in enumerate(finals):
             if res is None:
-                await context.abort(StatusCode.INTERNAL, "generation produced no output")
+                await context.abort(
+                    StatusCode.INTERNAL, "generation produced no output"
+                )
             if res.prompt is None:
                 res.prompt = request.requests[i].text
-            output = res.outputs[0]
-            response = self._convert_output(
-                output,
-                resp_options,
-                max_is_token_limit=max_is_token_limit[i],
+            comp = res.outputs[0]
+            msg = self._to_generation_response(
+                comp,
+                opts,
                 tokenizer=tokenizer,
-                time_limit_reached=time_limit_reached,
-                generated_token_count=len(output.token_ids),
+                capped=capped[i],
+                deadline_hit=deadline_hit,
+                gen_count=len(comp.token_ids),
             )
-            responses[i] = self._convert_input_details(
-                res, resp_options, sampling_params, response, tokenizer
-            )
-        return proto.BatchedGenerationResponse(responses=responses)
+            finals[i] = self._fill_input_details(res, opts, params, msg, tokenizer)
+        return proto.BatchedGenerationResponse(responses=finals)
 
-    # ------------------------------------------------------------------
-    @log_rpc_handler_errors
+    # -- GenerateStream --------------------------------------------------
+    @rpc_guard
     async def GenerateStream(self, request, context):
-        request_id = self.request_id(context)
-        adapter_kwargs = await self._validate_adapters(request, context)
+        rid = self._rid_from(context)
+        extra = await self._adapter_kwargs(request, context)
         tokenizer = await self.engine.get_tokenizer()
-
-        sampling_params, deadline = await self._validate_and_convert_params(
+        params, deadline = await self._params_from_proto(
             request.params, tokenizer, context
         )
-        sampling_params.output_kind = RequestOutputKind.DELTA
-        truncate_input_tokens = with_default(request.params.truncate_input_tokens, None)
-
-        input_ids, max_is_tok_limit = await self._validate_prompt_and_tokenize(
-            sampling_params, truncate_input_tokens, request.request.text,
-            tokenizer, context,
+        params.output_kind = RequestOutputKind.DELTA
+        trunc = request.params.truncate_input_tokens or None
+        ids, was_capped = await self._prepare_prompt(
+            params, trunc, request.request.text, tokenizer, context
         )
 
         headers = dict(context.invocation_metadata() or ())
-        if CORRELATION_ID_HEADER in headers:
-            logs.set_correlation_id(request_id, headers.get(CORRELATION_ID_HEADER))
-        result_generator = self.engine.generate(
-            prompt={"prompt": request.request.text, "prompt_token_ids": input_ids},
-            sampling_params=sampling_params,
-            request_id=request_id,
+        if _CORR_HEADER in headers:
+            logs.set_correlation_id(rid, headers.get(_CORR_HEADER))
+        stream = self.engine.generate(
+            prompt={"prompt": request.request.text, "prompt_token_ids": ids},
+            sampling_params=params,
+            request_id=rid,
             trace_headers=self._trace_headers(headers),
-            **adapter_kwargs,
+            **extra,
         )
 
-        resp_options = request.params.response
-        first_response = None
-        last_response = None
-        generated_token_count = 0
-        time_limit_reached = False
-        full_output = ""
-        async for result in result_generator:
-            # chunked prefill can emit several prompt-only outputs
-            if first_response is None or (
-                result.prompt_token_ids and not generated_token_count
-            ):
-                if result.prompt is None:
-                    result.prompt = request.request.text
-                first_response = self._convert_input_details(
-                    result, resp_options, sampling_params,
-                    proto.GenerationResponse(), tokenizer,
+        opts = request.params.response
+        head_msg = None  # message 1 of N+1: input details only
+        tail_msg = None
+        gen_count = 0
+        deadline_hit = False
+        text_acc = ""
+        async for res in stream:
+            # chunked prefill can surface several prompt-only results;
+            # the LAST of them carries the complete prompt details
+            if head_msg is None or (res.prompt_token_ids and not gen_count):
+                if res.prompt is None:
+                    res.prompt = request.request.text
+                head_msg = self._fill_input_details(
+                    res, opts, params, proto.GenerationResponse(), tokenizer
                 )
-                last_response = first_response
-                yield first_response
+                tail_msg = head_msg
+                yield head_msg
 
             if deadline is not None and time.time() >= deadline:
-                await self.engine.abort(request_id)
-                time_limit_reached = True
+                await self.engine.abort(rid)
+                deadline_hit = True
 
-            output = result.outputs[0]
-            generated_token_count += len(output.token_ids)
-            if (
-                not generated_token_count
-                and not output.finish_reason
-                and not time_limit_reached
-            ):
-                continue
+            comp = res.outputs[0]
+            gen_count += len(comp.token_ids)
+            if not gen_count and not comp.finish_reason and not deadline_hit:
+                continue  # prompt-only chunk, already reported
 
-            last_response = self._convert_output(
-                output,
-                resp_options,
-                max_is_token_limit=max_is_tok_limit,
+            tail_msg = self._to_generation_response(
+                comp,
+                opts,
                 tokenizer=tokenizer,
-                time_limit_reached=time_limit_reached,
-                generated_token_count=generated_token_count,
+                capped=was_capped,
+                deadline_hit=deadline_hit,
+                gen_count=gen_count,
             )
-            yield last_response
-            full_output += output.text
-            if time_limit_reached:
+            yield tail_msg
+            text_acc += comp.text
+            if deadline_hit:
                 break
 
-        if first_response is None:
+        if head_msg is None:
             return
-        # patch the first response for the response log
-        first_response.text = full_output
-        first_response.stop_reason = last_response.stop_reason
-        first_response.stop_sequence = last_response.stop_sequence
-        first_response.generated_token_count = last_response.generated_token_count
+        # back-fill the first message so the TGIS response log (which holds
+        # a reference to it) sees the whole-stream totals
+        head_msg.text = text_acc
+        head_msg.stop_reason = tail_msg.stop_reason
+        head_msg.stop_sequence = tail_msg.stop_sequence
+        head_msg.generated_token_count = tail_msg.generated_token_count
 
-    # ------------------------------------------------------------------
+    # -- conversion helpers ----------------------------------------------
     def _trace_headers(self, headers: dict) -> Optional[dict]:
-        # OTel trace-context pass-through (E20); propagate traceparent et al.
-        keys = ("traceparent", "tracestate")
-        found = {k: v for k, v in headers.items() if k.lower() in keys}
+        # OTel trace-context pass-through (E20)
+        keep = ("traceparent", "tracestate")
+        found = {k: v for k, v in headers.items() if k.lower() in keep}
         return found or None
 
-    def _convert_input_details(
-        self, result: "RequestOutput", resp_options, sampling_params,
-        response, tokenizer,
+    def _fill_input_details(
+        self, res: "RequestOutput", opts, params, msg, tokenizer
     ):
-        if result.prompt_token_ids:
-            response.input_token_count = len(result.prompt_token_ids)
-            if resp_options.input_tokens:
+        """Input-side fields of a GenerationResponse (message 1 of N+1)."""
+        if res.prompt_token_ids:
+            msg.input_token_count = len(res.prompt_token_ids)
+            if opts.input_tokens:
                 from .convert import append_token_details
 
                 append_token_details(
-                    response.input_tokens,
-                    result.prompt_token_ids,
-                    result.prompt_logprobs,
-                    want_logprob=resp_options.token_logprobs,
-                    want_rank=resp_options.token_ranks,
-                    top_n=resp_options.top_n_tokens,
+                    msg.input_tokens,
+                    res.prompt_token_ids,
+                    res.prompt_logprobs,
+                    want_logprob=opts.token_logprobs,
+                    want_rank=opts.token_ranks,
+                    top_n=opts.top_n_tokens,
                     tokenizer=tokenizer,
                 )
-        if resp_options.input_text and result.prompt:
-            response.text = (
-                result.prompt if not response.text else result.prompt + response.text
-            )
-        if sampling_params.seed is not None:
-            response.seed = sampling_params.seed
-        return response
+        if opts.input_text and res.prompt:
+            msg.text = res.prompt if not msg.text else res.prompt + msg.text
+        if params.seed is not None:
+            msg.seed = params.seed
+        return msg
 
-    def _convert_output(
-        self, output: "CompletionOutput", resp_options, *,
-        generated_token_count: int, max_is_token_limit: bool, tokenizer,
-        time_limit_reached: bool = False,
+    def _to_generation_response(
+        self, comp: "CompletionOutput", opts, *, gen_count: int,
+        capped: bool, tokenizer, deadline_hit: bool = False,
     ):
         from .convert import append_token_details, resolve_stop
 
         stop = resolve_stop(
-            output, capped=max_is_token_limit,
-            deadline_hit=time_limit_reached, tokenizer=tokenizer,
+            comp, capped=capped, deadline_hit=deadline_hit, tokenizer=tokenizer
         )
-        response = proto.GenerationResponse(
-            text=output.text,
-            generated_token_count=generated_token_count,
+        msg = proto.GenerationResponse(
+            text=comp.text,
+            generated_token_count=gen_count,
             stop_reason=stop.reason,
             stop_sequence=stop.sequence or "",
         )
-        if resp_options.generated_tokens:
+        if opts.generated_tokens:
             append_token_details(
-                response.tokens,
-                to_list(output.token_ids),
-                output.logprobs,
-                want_logprob=resp_options.token_logprobs,
-                want_rank=resp_options.token_ranks,
-                top_n=resp_options.top_n_tokens,
+                msg.tokens,
+                to_list(comp.token_ids),
+                comp.logprobs,
+                want_logprob=opts.token_logprobs,
+                want_rank=opts.token_ranks,
+                top_n=opts.top_n_tokens,
                 tokenizer=tokenizer,
             )
-        return response
+        return msg
 
-    # ------------------------------------------------------------------
-    async def _validate_and_convert_params(self, params, tokenizer, context):
-        """proto Parameters -> engine SamplingParams (+ deadline)."""
+    # -- parameter / prompt handling -------------------------------------
+    async def _params_from_proto(self, p, tokenizer, context):
+        """proto Parameters → engine SamplingParams plus wall-clock deadline."""
         from .convert import requested_logprob_count
 
         try:
-            validate_params(params, self.max_max_new_tokens)
+            validate_params(p, self.server_token_cap)
         except ValueError as e:
             await context.abort(StatusCode.INVALID_ARGUMENT, str(e))
 
-        greedy = params.method == proto.GREEDY
-        stopping = params.stopping
-        decoding = params.decoding
+        greedy = p.method == proto.GREEDY
 
-        kwargs: dict = {
-            "logprobs": requested_logprob_count(params.response, greedy),
-            "max_tokens": stopping.max_new_tokens or None,
-            "min_tokens": max(0, stopping.min_new_tokens),
-            "repetition_penalty": decoding.repetition_penalty or 1.0,
-            "stop": list(stopping.stop_sequences) or None,
-            "skip_special_tokens": self.skip_special_tokens,
+        fields: dict = {
+            "logprobs": requested_logprob_count(p.response, greedy),
+            "max_tokens": p.stopping.max_new_tokens or None,
+            "min_tokens": max(0, p.stopping.min_new_tokens),
+            "repetition_penalty": p.decoding.repetition_penalty or 1.0,
+            "stop": list(p.stopping.stop_sequences) or None,
+            "skip_special_tokens": self.strip_special_tokens,
             "include_stop_str_in_output": (
-                stopping.include_stop_sequence
-                if stopping.HasField("include_stop_sequence")
-                else self.default_include_stop_seqs
+                p.stopping.include_stop_sequence
+                if p.stopping.HasField("include_stop_sequence")
+                else self.include_stops_default
             ),
         }
-        if params.response.input_tokens and not self.disable_prompt_logprobs:
-            kwargs["prompt_logprobs"] = kwargs["logprobs"]
+        if p.response.input_tokens and not self.prompt_logprobs_off:
+            fields["prompt_logprobs"] = fields["logprobs"]
 
-        # sampling method: greedy pins temperature 0; sampling carries the
-        # client's temperature/top-k/top-p/seed through unchanged
-        sampling = params.sampling
-        temp = sampling.temperature if sampling.HasField("temperature") else 1.0
+        # GREEDY pins temperature 0; SAMPLE passes the client's knobs through
+        s = p.sampling
+        temp = s.temperature if s.HasField("temperature") else 1.0
         if greedy or temp == 0.0:
-            kwargs["temperature"] = 0.0
+            fields["temperature"] = 0.0
         else:
-            kwargs["temperature"] = temp
-            kwargs["top_k"] = sampling.top_k or -1
-            kwargs["top_p"] = sampling.top_p or 1.0
-            if sampling.HasField("seed"):
-                kwargs["seed"] = sampling.seed
+            fields["temperature"] = temp
+            fields["top_k"] = s.top_k or -1
+            fields["top_p"] = s.top_p or 1.0
+            if s.HasField("seed"):
+                fields["seed"] = s.seed
 
-        # per-request logits processors (host-side hooks, E9)
-        procs = []
-        if not greedy and 0.0 < sampling.typical_p < 1.0:
-            procs.append(TypicalLogitsWarperWrapper(mass=sampling.typical_p))
-        if decoding.HasField("length_penalty"):
-            lp = decoding.length_penalty
-            procs.append(ExpDecayLengthPenaltyWarper(
-                length_penalty=(lp.start_index, lp.decay_factor),
-                eos_token_id=tokenizer.eos_token_id,
-            ))
-        kwargs["logits_processors"] = procs
+        # host-side per-request logits hooks (E9)
+        hooks = []
+        if not greedy and 0.0 < s.typical_p < 1.0:
+            hooks.append(TypicalLogitsWarperWrapper(mass=s.typical_p))
+        if p.decoding.HasField("length_penalty"):
+            lp = p.decoding.length_penalty
+            hooks.append(
+                ExpDecayLengthPenaltyWarper(
+                    length_penalty=(lp.start_index, lp.decay_factor),
+                    eos_token_id=tokenizer.eos_token_id,
+                )
+            )
+        fields["logits_processors"] = hooks
 
-        structured = get_structured_output_params(decoding)
+        structured = get_structured_output_params(p.decoding)
         if structured is not None:
             from ..engine.guided import validate_structured_outputs
 
@@ -418,19 +408,19 @@ class TextGenerationService:
                 validate_structured_outputs(structured)
             except ValueError as e:
                 await context.abort(StatusCode.INVALID_ARGUMENT, str(e))
-            kwargs["structured_outputs"] = structured
+            fields["structured_outputs"] = structured
 
         deadline = None
-        if stopping.time_limit_millis > 0:
-            deadline = time.time() + stopping.time_limit_millis / 1000.0
+        if p.stopping.time_limit_millis > 0:
+            deadline = time.time() + p.stopping.time_limit_millis / 1000.0
 
         try:
-            sampling_params = SamplingParams(**kwargs)
+            params = SamplingParams(**fields)
         except ValueError as e:
             await context.abort(StatusCode.INVALID_ARGUMENT, str(e))
-        return sampling_params, deadline
+        return params, deadline
 
-    async def _validate_adapters(self, request, context):
+    async def _adapter_kwargs(self, request, context):
         try:
             return await validate_adapters(
                 request=request,
@@ -440,65 +430,70 @@ class TextGenerationService:
         except ValueError as e:
             await context.abort(StatusCode.INVALID_ARGUMENT, str(e))
 
-    async def _validate_prompt_and_tokenize(
-        self, sampling_params, truncate_input_tokens, prompt, tokenizer, context
+    async def _prepare_prompt(
+        self, params, trunc, text, tokenizer, context
     ) -> tuple[list[int], bool]:
+        """Tokenize + truncate + cap max_tokens; bool marks a server cap
+        (length-finish then maps to TOKEN_LIMIT, not MAX_TOKENS)."""
         from .convert import tokenize_with_caps
 
         assert self.config is not None
         try:
             return tokenize_with_caps(
-                prompt,
-                sampling_params,
+                text,
+                params,
                 tokenizer=tokenizer,
                 add_special_tokens=ADD_SPECIAL_TOKENS,
-                truncate_to=truncate_input_tokens,
+                truncate_to=trunc,
                 max_model_len=self.config.max_model_len,
-                default_max_new=self.max_max_new_tokens,
+                default_max_new=self.server_token_cap,
                 validate_input_fn=validate_input,
             )
         except ValueError as e:
             await context.abort(StatusCode.INVALID_ARGUMENT, str(e))
 
-    # ------------------------------------------------------------------
-    @log_rpc_handler_errors
+    # -- Tokenize ---------------------------------------------------------
+    @rpc_guard
     async def Tokenize(self, request, context):
-        adapter_kwargs = await self._validate_adapters(request, context)
-        tokenizer = await self.engine.get_tokenizer(
-            adapter_kwargs.get("lora_request")
-        )
-        responses = []
-        for req in request.requests:
-            batch_encoding = tokenizer(
-                req.text,
+        extra = await self._adapter_kwargs(request, context)
+        tokenizer = await self.engine.get_tokenizer(extra.get("lora_request"))
+        out = []
+        for sub in request.requests:
+            enc = tokenizer(
+                sub.text,
                 return_offsets_mapping=request.return_offsets,
                 add_special_tokens=ADD_SPECIAL_TOKENS,
             )
-            token_ids = batch_encoding.input_ids
-            token_count = len(token_ids)
-            if 0 < request.truncate_input_tokens < token_count:
-                token_count = request.truncate_input_tokens
-            tokens = tokenizer.convert_ids_to_tokens(token_ids)
-            resp = proto.TokenizeResponse(token_count=token_count)
+            n = len(enc.input_ids)
+            if 0 < request.truncate_input_tokens < n:
+                n = request.truncate_input_tokens  # keep the LAST n tokens
+            resp = proto.TokenizeResponse(token_count=n)
             if request.return_tokens:
-                resp.tokens.extend(tokens[-token_count:])
+                resp.tokens.extend(
+                    tokenizer.convert_ids_to_tokens(enc.input_ids)[-n:]
+                )
             if request.return_offsets:
-                offsets = [
-                    (start, end)
-                    for start, end in batch_encoding.offset_mapping
-                    if start is not None and end is not None
+                spans = [
+                    (a, b)
+                    for a, b in enc.offset_mapping
+                    if a is not None and b is not None
                 ]
-                for start, end in offsets[-token_count:]:
-                    o = resp.offsets.add()
-                    o.start = start
-                    o.end = end
-            responses.append(resp)
-        return proto.BatchedTokenizeResponse(responses=responses)
+                for a, b in spans[-n:]:
+                    off = resp.offsets.add()
+                    off.start = a
+                    off.end = b
+            out.append(resp)
+        return proto.BatchedTokenizeResponse(responses=out)
 
-    @log_rpc_handler_errors
+    # -- ModelInfo --------------------------------------------------------
+    @rpc_guard
     async def ModelInfo(self, request, context):
         return proto.ModelInfoResponse(
-            model_kind=0,  # fmaas.ModelInfoResponse.ModelKind.DECODER_ONLY
+            model_kind=0,  # DECODER_ONLY
             max_sequence_length=self.config.max_model_len,
-            max_new_tokens=self.max_max_new_tokens,
+            max_new_tokens=self.server_token_cap,
         )
+
+
+# the reference-era class name, kept importable for continuity
+TextGenerationService = GenerationServicer
