@@ -1,8 +1,11 @@
 """TGIS-format per-request logs.
 
-engine.generate is wrapped once at startup so gRPC and HTTP requests log
-identically (reference tgis_utils/logs.py:48-114).  Correlation ids arrive
-from either front-end via a TTL-bounded blackboard.
+`engine.generate` is wrapped once at startup, so both front-ends (gRPC and
+HTTP) produce identical request / response / cancellation / error log lines
+— the same single-wrap strategy the reference uses (reference
+tgis_utils/logs.py:48-114).  The log line FORMATS are parity-tested; the
+plumbing below is this repo's own.  Correlation ids arrive from either
+front-end through a TTL-bounded blackboard keyed by request id.
 """
 
 from __future__ import annotations
@@ -18,152 +21,123 @@ from ..logging import init_logger
 from ..utils import TTLCache
 
 if TYPE_CHECKING:
-    from ..engine.types import RequestMetrics, RequestOutput, SamplingParams
+    from ..engine.types import RequestOutput, SamplingParams
 
 logger = init_logger(__name__)
 
-_REQUEST_ID_TO_CORRELATION_ID = TTLCache(maxsize=2048, ttl=600)
+# request id -> correlation id, dropped after 10 minutes (2048 entries max)
+_CORR_BOARD = TTLCache(maxsize=2048, ttl=600)
 
 
 def set_correlation_id(request_id: str, correlation_id: str | None) -> None:
     if correlation_id is not None:
-        _REQUEST_ID_TO_CORRELATION_ID[request_id] = correlation_id
+        _CORR_BOARD[request_id] = correlation_id
 
 
 def get_correlation_id(request_id: str) -> str | None:
-    correlation_id = _REQUEST_ID_TO_CORRELATION_ID.get(request_id)
-    if not correlation_id:
-        # http request ids look like {method}-{base_id}-{batch_index}
-        request_id = "-".join(request_id.split("-")[1:-1])
-        correlation_id = _REQUEST_ID_TO_CORRELATION_ID.get(request_id)
-    return correlation_id
+    cid = _CORR_BOARD.get(request_id)
+    if not cid:
+        # HTTP request ids have the shape {method}-{base_id}-{batch_index};
+        # the blackboard entry was keyed by the bare base id
+        base = "-".join(request_id.split("-")[1:-1])
+        cid = _CORR_BOARD.get(base)
+    return cid
+
+
+def _call_arg(args: tuple, kwargs: dict, name: str, index: int):
+    """Positional-or-keyword lookup into a wrapped generate() call."""
+    return args[index] if len(args) > index else kwargs.get(name)
+
+
+def _params_repr(params: "SamplingParams") -> str:
+    """str(params) with guided-decoding payloads redacted — schemas and
+    regexes are user content and do not belong in server logs."""
+    text = str(params)
+    constraint = getattr(params, "guided_decoding", None)
+    if constraint is not None:
+        text = text.replace(str(constraint), "(...)")
+    return text
+
+
+def _ratio(num: float, den: float) -> float:
+    return num / den if den else 0.0
 
 
 def add_logging_wrappers(engine) -> None:
-    """Wrap engine.generate with request/response/cancel/error logging."""
-    old_generate = engine.generate
+    """Install the generate() wrapper on an engine client (idempotent per
+    engine instance; called once from __main__)."""
+    inner = engine.generate
 
-    @functools.wraps(old_generate)
-    async def generate_with_logging(*args, **kwargs):
-        start_time = time.time()
-        prompt = _get_arg("prompt", 0, *args, **kwargs)
-        sampling_params = _get_arg("sampling_params", 1, *args, **kwargs)
-        request_id = _get_arg("request_id", 2, *args, **kwargs)
-        lora_request = _get_arg("lora_request", 3, *args, **kwargs)
-
-        correlation_id = get_correlation_id(request_id=request_id)
-        adapter_id = lora_request.adapter_id if lora_request else None
+    @functools.wraps(inner)
+    async def logged_generate(*args, **kwargs):
+        t_start = time.time()
+        prompt = _call_arg(args, kwargs, "prompt", 0)
+        params = _call_arg(args, kwargs, "sampling_params", 1)
+        rid = _call_arg(args, kwargs, "request_id", 2)
+        lora = _call_arg(args, kwargs, "lora_request", 3)
+        cid = get_correlation_id(request_id=rid)
 
         with suppress(BaseException):
-            _log_request(
-                prompt=prompt,
-                params=sampling_params,
-                request_id=request_id,
-                correlation_id=correlation_id,
-                adapter_id=adapter_id,
+            ntok = ""
+            if isinstance(prompt, dict) and "prompt_token_ids" in prompt:
+                ntok = f" input_tokens={len(prompt['prompt_token_ids'])},"
+            logger.info(
+                "Processing request: {request_id=%s, correlation_id=%s, "
+                "adapter_id=%s, %sparams=%s}",
+                rid, cid, lora.adapter_id if lora else None, ntok,
+                _params_repr(params),
             )
 
-        last = None
+        final = None
         try:
-            async for response in old_generate(*args, **kwargs):
-                last = response
-                yield response
+            async for out in inner(*args, **kwargs):
+                final = out
+                yield out
         except asyncio.CancelledError:
-            _log_cancellation(request_id, correlation_id)
+            logger.info(
+                "Request cancelled: request_id=%s correlation_id=%s", rid, cid
+            )
             raise
         except BaseException as e:
-            _log_error(request_id, correlation_id, str(e))
+            logger.error(
+                "Request failed: request_id=%s correlation_id=%s error=%s",
+                rid, cid, str(e),
+            )
             raise
 
-        if last:
+        if final:
             with suppress(BaseException):
-                _log_response(
-                    request_id=request_id,
-                    correlation_id=correlation_id,
-                    response=last,
-                    engine_metrics=last.metrics,
-                    start_time=start_time,
-                )
+                _emit_response_line(rid, cid, final, t_start)
 
-    engine.generate = generate_with_logging
+    engine.generate = logged_generate
 
 
-def _log_error(request_id, correlation_id, exception_str) -> None:
-    logger.error(
-        "Request failed: request_id=%s correlation_id=%s error=%s",
-        request_id, correlation_id, exception_str,
-    )
-
-
-def _log_cancellation(request_id, correlation_id) -> None:
-    logger.info(
-        "Request cancelled: request_id=%s correlation_id=%s",
-        request_id, correlation_id,
-    )
-
-
-def _sanitize_sampling_params(params: "SamplingParams") -> str:
-    """Redact guided-decoding payloads (may contain user schemas)."""
-    original = str(params)
-    guided = getattr(params, "guided_decoding", None)
-    if guided is not None:
-        return original.replace(str(guided), "(...)")
-    return original
-
-
-def _log_request(request_id, params, adapter_id, correlation_id, prompt) -> None:
-    if isinstance(prompt, dict) and "prompt_token_ids" in prompt:
-        input_tokens = f" input_tokens={len(prompt['prompt_token_ids'])},"
-    else:
-        input_tokens = ""
-    logger.info(
-        "Processing request: {request_id=%s, correlation_id=%s, adapter_id=%s, "
-        "%sparams=%s}",
-        request_id, correlation_id, adapter_id, input_tokens,
-        _sanitize_sampling_params(params),
-    )
-
-
-def _log_response(
-    request_id, correlation_id, response: "RequestOutput",
-    engine_metrics: "RequestMetrics | None", start_time: float,
+def _emit_response_line(
+    rid, cid, final: "RequestOutput", t_start: float
 ) -> None:
-    if not response.outputs:
+    """One summary line per finished request, with the queue / inference /
+    per-token / total timings from the engine's RequestMetrics."""
+    if not final.outputs:
         return
-    generated_tokens = len(response.outputs[0].token_ids)
-    if engine_metrics is None or engine_metrics.first_scheduled_time is None:
+    comp = final.outputs[0]
+    n_gen = len(comp.token_ids)
+    em = final.metrics
+    if em is None or em.first_scheduled_time is None:
         logger.warning("No engine metrics for request, cannot log timing info")
-        inference_time = queue_time = time_per_token = total_time = 0.0
+        t_queue = t_infer = t_tok = t_total = 0.0
     else:
-        last = engine_metrics.last_token_time or start_time
-        inference_time = last - engine_metrics.first_scheduled_time
-        queue_time = engine_metrics.time_in_queue or 0.0
-        time_per_token = _safe_div(inference_time, generated_tokens)
-        total_time = last - start_time
-    output_len = len(response.outputs[0].text)
-    stop_reason_str = response.outputs[0].finish_reason
-    level = logging.WARNING if stop_reason_str == "abort" else logging.INFO
+        t_end = em.last_token_time or t_start
+        t_infer = t_end - em.first_scheduled_time
+        t_queue = em.time_in_queue or 0.0
+        t_tok = _ratio(t_infer, n_gen)
+        t_total = t_end - t_start
+    finish = comp.finish_reason
     logger.log(
-        level,
+        logging.WARNING if finish == "abort" else logging.INFO,
         "Finished processing request: {request_id=%s, correlation_id=%s}. "
         "Timing info: {queue_time=%.2fms, inference_time=%.2fms, "
         "time_per_token=%.2fms, total_time=%.2fms}. "
         "Generated %d tokens before finish reason: %s, output %d chars",
-        request_id, correlation_id,
-        queue_time * 1e3, inference_time * 1e3,
-        time_per_token * 1e3, total_time * 1e3,
-        generated_tokens, stop_reason_str, output_len,
+        rid, cid, t_queue * 1e3, t_infer * 1e3, t_tok * 1e3, t_total * 1e3,
+        n_gen, finish, len(comp.text),
     )
-
-
-def _safe_div(a: float, b: float, *, default: float = 0.0) -> float:
-    try:
-        return a / b
-    except ZeroDivisionError:
-        return default
-
-
-def _get_arg(name: str, pos: int, *args, **kwargs):
-    if len(args) > pos:
-        return args[pos]
-    return kwargs.get(name)
