@@ -417,3 +417,42 @@ def test_pipelined_guided_matches_sync():
     assert sync == pipe
     for rid in ("g0", "g2"):
         assert re.fullmatch(r"[0-9]{4}", sync[rid][0]), sync[rid]
+
+
+def test_pipelined_preemption_matches_sync():
+    """A tiny KV pool forces preemption (free + recompute) mid-run; the
+    pipelined path must recover to the same outputs as the sync path."""
+    def run(pipeline):
+        os.environ["VTA_PIPELINE"] = "1" if pipeline else "0"
+        os.environ["VTA_PIPELINE_MIN"] = "1"
+        try:
+            mc = ModelConfig.from_model_arg("tiny-llama", dtype="float32")
+            eng = LLMEngine(EngineConfig(
+                model_config=mc,
+                cache_config=CacheConfig(block_size=16, num_gpu_blocks=10),
+                scheduler_config=SchedulerConfig(max_num_seqs=8,
+                                                 max_num_batched_tokens=512),
+                device="cpu", seed=0,
+            ))
+            for i in range(4):
+                eng.add_request(
+                    f"k{i}", None, list(range(90 + 20 * i, 90 + 20 * i + 30)),
+                    SamplingParams(temperature=0.0, max_tokens=24))
+            finals = {}
+            steps = 0
+            while eng.has_unfinished() and steps < 400:
+                for out in eng.step():
+                    if out.finished:
+                        o = out.outputs[0]
+                        finals[out.request_id] = (o.text, tuple(o.token_ids),
+                                                  o.finish_reason)
+                steps += 1
+            assert not eng.has_unfinished()
+            return finals
+        finally:
+            os.environ.pop("VTA_PIPELINE", None)
+            os.environ.pop("VTA_PIPELINE_MIN", None)
+
+    sync = run(False)
+    pipe = run(True)
+    assert sync == pipe and len(sync) == 4
