@@ -71,8 +71,10 @@ def _engine_proc_main(config: EngineConfig, cmd_conn, out_conn) -> None:
     thread preserves send order; the step loop never waits on the pipe.
     """
     import collections
+    import os
     import threading
 
+    use_thread = os.environ.get("VTA_SEND_THREAD", "1") == "1"
     sendq: collections.deque = collections.deque()
     send_evt = threading.Event()
     _STOP_SENTINEL = object()
@@ -93,16 +95,20 @@ def _engine_proc_main(config: EngineConfig, cmd_conn, out_conn) -> None:
             except (BrokenPipeError, OSError):
                 return  # parent gone; engine loop will notice via cmds
 
-    sender_thread = threading.Thread(target=_sender, daemon=True,
-                                     name="mp-engine-sender")
-    sender_thread.start()
+    sender_thread = None
+    if use_thread:
+        sender_thread = threading.Thread(target=_sender, daemon=True,
+                                         name="mp-engine-sender")
+        sender_thread.start()
 
-    def send(item) -> None:
-        sendq.append(item)
-        send_evt.set()
+        def send(item) -> None:
+            sendq.append(item)
+            send_evt.set()
+    else:
+        def send(item) -> None:
+            out_conn.send(item)
 
     try:
-        import os
 
         if os.environ.get("VTA_NULL_ENGINE", "0") == "1":
             from .null_engine import NullEngine as LLMEngine
@@ -167,15 +173,17 @@ def _engine_proc_main(config: EngineConfig, cmd_conn, out_conn) -> None:
                 cmd_conn.poll(0.02)
         engine.shutdown()
         send(("stopped", None))
-        sendq.append(_STOP_SENTINEL)
-        send_evt.set()
-        sender_thread.join(timeout=10)
+        if sender_thread is not None:
+            sendq.append(_STOP_SENTINEL)
+            send_evt.set()
+            sender_thread.join(timeout=10)
     except BaseException:
         try:
             # drain-then-report: the fatal must not race queued sends
-            sendq.append(_STOP_SENTINEL)
-            send_evt.set()
-            sender_thread.join(timeout=5)
+            if sender_thread is not None:
+                sendq.append(_STOP_SENTINEL)
+                send_evt.set()
+                sender_thread.join(timeout=5)
             out_conn.send(("fatal", traceback.format_exc()))
         except Exception:
             pass
