@@ -63,57 +63,57 @@ class TGISValidationError(str, Enum):
 
 def validate_input(sampling_params, token_num: int, max_model_len: int) -> None:
     """Reject prompts that cannot fit the model context."""
+    n_min = sampling_params.min_tokens
     if token_num >= max_model_len:
         TGISValidationError.InputLength2.error(token_num, 0, max_model_len)
-    if token_num + sampling_params.min_tokens > max_model_len:
-        TGISValidationError.InputLength.error(
-            token_num, 0, sampling_params.min_tokens, max_model_len
-        )
+    if token_num + n_min > max_model_len:
+        TGISValidationError.InputLength.error(token_num, 0, n_min, max_model_len)
 
 
-def validate_params(params, max_max_new_tokens: int) -> None:
-    """Raise ValueError (TGIS strings) when Parameters is invalid."""
-    resp = params.response
-    sampling = params.sampling
+def validate_params(params, server_cap: int) -> None:
+    """Raise ValueError carrying the exact TGIS string when any field of a
+    proto Parameters violates the documented TGIS limits.  Check ORDER is
+    part of observable behavior (the first violated limit names the error)
+    and follows the reference."""
+    err = TGISValidationError
+
+    if params.decoding.HasField("length_penalty"):
+        decay = params.decoding.length_penalty.decay_factor
+        if decay < 1.0 or decay > 10.0:
+            err.LengthPenalty.error()
+
+    rep = params.decoding.repetition_penalty  # 0 == unset == no penalty
+    if rep < 0 or rep > 2:
+        err.RepetitionPenalty.error()
+
     stopping = params.stopping
-    decoding = params.decoding
+    if stopping.max_new_tokens > server_cap:
+        err.MaxNewTokens.error(server_cap)
+    if stopping.min_new_tokens > (stopping.max_new_tokens or server_cap):
+        err.MinNewTokens.error()
 
-    if decoding.HasField("length_penalty"):
-        decay = decoding.length_penalty.decay_factor
-        if not (1.0 <= decay <= 10.0):
-            TGISValidationError.LengthPenalty.error()
+    seqs = list(stopping.stop_sequences)
+    ok_seqs = len(seqs) <= MAX_STOP_SEQS and all(
+        0 < len(s) <= MAX_STOP_SEQ_LENGTH for s in seqs
+    )
+    if not ok_seqs:
+        err.StopSequences.error(MAX_STOP_SEQS, MAX_STOP_SEQ_LENGTH)
 
-    if not (0 <= decoding.repetition_penalty <= 2):
-        # 0 means unset / no penalty
-        TGISValidationError.RepetitionPenalty.error()
-
-    if stopping.max_new_tokens > max_max_new_tokens:
-        TGISValidationError.MaxNewTokens.error(max_max_new_tokens)
-
-    if stopping.min_new_tokens > (stopping.max_new_tokens or max_max_new_tokens):
-        TGISValidationError.MinNewTokens.error()
-
-    stop_seqs = list(stopping.stop_sequences)
-    if (stop_seqs and len(stop_seqs) > MAX_STOP_SEQS) or not all(
-        0 < len(s) <= MAX_STOP_SEQ_LENGTH for s in stop_seqs
-    ):
-        TGISValidationError.StopSequences.error(MAX_STOP_SEQS, MAX_STOP_SEQ_LENGTH)
-
+    resp = params.response
     if resp.top_n_tokens > MAX_TOP_N_TOKENS:
-        TGISValidationError.TopN.error(resp.top_n_tokens, MAX_TOP_N_TOKENS)
+        err.TopN.error(resp.top_n_tokens, MAX_TOP_N_TOKENS)
+    detail = resp.token_logprobs or resp.token_ranks or resp.top_n_tokens
+    if detail and not (resp.input_tokens or resp.generated_tokens):
+        err.TokenDetail.error()
 
-    wants_detail = resp.token_logprobs or resp.token_ranks or resp.top_n_tokens
-    if wants_detail and not (resp.input_tokens or resp.generated_tokens):
-        TGISValidationError.TokenDetail.error()
-
-    greedy = params.method == proto.GREEDY
-    if STRICT_PARAMETER_VALIDATION and greedy and (
-        sampling.temperature or sampling.top_k or sampling.top_p or sampling.typical_p
+    s = params.sampling
+    if STRICT_PARAMETER_VALIDATION and params.method == proto.GREEDY and (
+        s.temperature or s.top_k or s.top_p or s.typical_p
     ):
-        TGISValidationError.SampleParametersGreedy.error()
-    if sampling.top_k < 0:
-        TGISValidationError.TopK.error()
-    if not (0 <= sampling.top_p <= 1):
-        TGISValidationError.TopP.error()
-    if sampling.typical_p > 1:
-        TGISValidationError.TypicalP.error()
+        err.SampleParametersGreedy.error()
+    if s.top_k < 0:
+        err.TopK.error()
+    if s.top_p < 0 or s.top_p > 1:
+        err.TopP.error()
+    if s.typical_p > 1:
+        err.TypicalP.error()
