@@ -4,16 +4,16 @@ Behavioral contract from the reference (grpc/adapters.py): per-adapter
 asyncio locks, unique ids starting at 1000001, path-traversal rejection,
 ``adapter_config.json`` read in a small thread pool, ``peft_type == "LORA"``
 registered with the serving-models handler, anything else rejected with the
-TGIS error strings.
+TGIS error strings.  The resolution flow below is this repo's own.
 """
 
 from __future__ import annotations
 
 import asyncio
-import concurrent.futures
 import dataclasses
 import json
 import re
+from concurrent.futures import ThreadPoolExecutor
 from pathlib import Path
 from typing import TYPE_CHECKING, Optional
 
@@ -23,11 +23,13 @@ from .validation import TGISValidationError
 if TYPE_CHECKING:
     from ..engine.types import LoRARequest
 
-_thread_pool: Optional[concurrent.futures.ThreadPoolExecutor] = None
-
-VALID_ADAPTER_ID_PATTERN = re.compile("[/\\w\\-]+")
-
 logger = init_logger(__name__)
+
+# alphanumerics, underscore, dash and slash only — everything else (and any
+# id that resolves outside the cache dir) is a traversal attempt
+_ID_RE = re.compile(r"[/\w\-]+")
+
+_config_pool: Optional[ThreadPoolExecutor] = None
 
 
 @dataclasses.dataclass
@@ -46,93 +48,94 @@ class AdapterStore:
     load_locks: dict[str, asyncio.Lock] = dataclasses.field(default_factory=dict)
 
 
+def _screen_id(adapter_id: str) -> None:
+    """Path-traversal defense: charset allow-list, then a resolved-path
+    containment check against the working directory."""
+    if not _ID_RE.fullmatch(adapter_id):
+        TGISValidationError.InvalidAdapterID.error(adapter_id)
+    if not Path(adapter_id).resolve().is_relative_to(Path.cwd()):
+        TGISValidationError.InvalidAdapterID.error(adapter_id)
+
+
+def _read_adapter_config(adapter_id: str, where: str, uid: int) -> AdapterMetadata:
+    """Blocking filesystem work — runs in the 2-worker pool."""
+    root = Path(where)
+    if not root.exists():
+        TGISValidationError.AdapterNotFound.error(
+            adapter_id, "directory does not exist"
+        )
+    cfg_file = root / "adapter_config.json"
+    if not cfg_file.exists():
+        TGISValidationError.AdapterNotFound.error(
+            adapter_id, "invalid adapter: no adapter_config.json found"
+        )
+    cfg = json.loads(cfg_file.read_text())
+    return AdapterMetadata(
+        unique_id=uid,
+        adapter_type=cfg.get("peft_type", None),
+        full_path=where,
+        full_config=cfg,
+    )
+
+
+async def _register_lora(
+    request, adapter_id: str, meta: AdapterMetadata, model_handler
+) -> "LoRARequest":
+    """Hand the adapter to the engine's model handler and return its
+    LoRARequest (the handler owns the live registry)."""
+    problem = await model_handler.load_lora_adapter(
+        lora_name=adapter_id,
+        lora_path=meta.full_path,
+        base_model_name=request.model_id,
+    )
+    if problem is not None:
+        raise ValueError(problem)
+    lr = model_handler.lora_requests.get(adapter_id)
+    if lr is None:
+        raise RuntimeError("engine failed to load LoRA adapter")
+    return lr
+
+
 async def validate_adapters(
     request,
     adapter_store: Optional[AdapterStore],
     model_handler,
 ) -> dict[str, "LoRARequest"]:
-    """Resolve the request's adapter_id to engine kwargs, loading on demand."""
-    global _thread_pool
-    adapter_id = request.adapter_id
-    if not adapter_id and request.prefix_id:
-        adapter_id = request.prefix_id  # legacy prefix_id support
+    """Resolve the request's adapter_id into engine kwargs, loading from the
+    on-disk cache the first time an id is seen."""
+    global _config_pool
 
-    if adapter_id and not adapter_store:
-        TGISValidationError.AdaptersDisabled.error()
-    if not adapter_id or not adapter_store:
+    wanted = request.adapter_id or request.prefix_id  # prefix_id: legacy alias
+    if not wanted:
         return {}
+    if adapter_store is None:
+        TGISValidationError.AdaptersDisabled.error()
 
-    async with adapter_store.load_locks.setdefault(adapter_id, asyncio.Lock()):
-        existing = model_handler.lora_requests.get(adapter_id)
-        if existing is not None:
-            return {"lora_request": existing}
+    lock = adapter_store.load_locks.setdefault(wanted, asyncio.Lock())
+    async with lock:
+        live = model_handler.lora_requests.get(wanted)
+        if live is not None:
+            return {"lora_request": live}
 
-        adapter_metadata = adapter_store.adapters.get(adapter_id)
-        if adapter_metadata is None:
-            _reject_bad_adapter_id(adapter_id)
-            local_path = str(Path(adapter_store.cache_path) / adapter_id)
-
-            loop = asyncio.get_running_loop()
-            if _thread_pool is None:
-                _thread_pool = concurrent.futures.ThreadPoolExecutor(max_workers=2)
-
-            # allocate the unique id in async land (no thread-safety concerns)
-            unique_id = adapter_store.next_unique_id
+        meta = adapter_store.adapters.get(wanted)
+        if meta is None:
+            _screen_id(wanted)
+            if _config_pool is None:
+                _config_pool = ThreadPoolExecutor(max_workers=2)
+            # ids are allocated here, in async land, so the counter needs
+            # no locking (reference behavior: count from 1000001)
+            uid = adapter_store.next_unique_id
             adapter_store.next_unique_id += 1
-
-            adapter_metadata = await loop.run_in_executor(
-                _thread_pool, _load_adapter_metadata, adapter_id, local_path, unique_id
+            meta = await asyncio.get_running_loop().run_in_executor(
+                _config_pool,
+                _read_adapter_config,
+                wanted,
+                str(Path(adapter_store.cache_path) / wanted),
+                uid,
             )
+            if meta.adapter_type == "LORA":
+                lr = await _register_lora(request, wanted, meta, model_handler)
+                return {"lora_request": lr}
+            adapter_store.adapters[wanted] = meta
 
-            if adapter_metadata.adapter_type == "LORA":
-                lora_request = await _load_lora_adapter(
-                    request, adapter_id, adapter_metadata, model_handler
-                )
-                return {"lora_request": lora_request}
-            adapter_store.adapters[adapter_id] = adapter_metadata
-
-    TGISValidationError.AdapterUnsupported.error(adapter_metadata.adapter_type)
-
-
-async def _load_lora_adapter(
-    request, adapter_id: str, adapter_metadata: AdapterMetadata, model_handler
-) -> "LoRARequest":
-    err = await model_handler.load_lora_adapter(
-        lora_name=adapter_id,
-        lora_path=adapter_metadata.full_path,
-        base_model_name=request.model_id,
-    )
-    if err is not None:
-        raise ValueError(err)
-    existing = model_handler.lora_requests.get(adapter_id)
-    if existing is not None:
-        return existing
-    raise RuntimeError("engine failed to load LoRA adapter")
-
-
-def _load_adapter_metadata(adapter_id: str, adapter_path: str, unique_id: int) -> AdapterMetadata:
-    """Filesystem access to deduce the adapter type (runs in the pool)."""
-    if not Path(adapter_path).exists():
-        TGISValidationError.AdapterNotFound.error(adapter_id, "directory does not exist")
-    config_path = Path(adapter_path) / "adapter_config.json"
-    if not config_path.exists():
-        TGISValidationError.AdapterNotFound.error(
-            adapter_id, "invalid adapter: no adapter_config.json found"
-        )
-    with open(config_path) as f:
-        adapter_config = json.load(f)
-    return AdapterMetadata(
-        unique_id=unique_id,
-        adapter_type=adapter_config.get("peft_type", None),
-        full_path=adapter_path,
-        full_config=adapter_config,
-    )
-
-
-def _reject_bad_adapter_id(adapter_id: str) -> None:
-    """Reject ids with path traversal or invalid characters."""
-    if not VALID_ADAPTER_ID_PATTERN.fullmatch(adapter_id):
-        TGISValidationError.InvalidAdapterID.error(adapter_id)
-    cwd = Path().cwd()
-    if not Path(adapter_id).resolve().is_relative_to(cwd):
-        TGISValidationError.InvalidAdapterID.error(adapter_id)
+    TGISValidationError.AdapterUnsupported.error(meta.adapter_type)
