@@ -13,6 +13,12 @@ from .grpc.proto import HealthCheckRequest
 from .grpc.stubs import HealthStub
 
 
+def _open_channel(target: str, insecure: bool):
+    if insecure:
+        return grpc.insecure_channel(target)
+    return grpc.secure_channel(target, grpc.ssl_channel_credentials())
+
+
 def health_check(
     *,
     server_url: str = "localhost:8033",
@@ -21,38 +27,35 @@ def health_check(
     timeout: float = 1,
 ) -> bool:
     print("health check...", end="")
-    request = HealthCheckRequest(service=service or "")
     try:
-        if insecure:
-            channel = grpc.insecure_channel(server_url)
-        else:
-            channel = grpc.secure_channel(server_url, grpc.ssl_channel_credentials())
-        with channel:
-            response = HealthStub(channel).Check(request, timeout=timeout)
+        with _open_channel(server_url, insecure) as ch:
+            reply = HealthStub(ch).Check(
+                HealthCheckRequest(service=service or ""), timeout=timeout
+            )
     except grpc.RpcError as e:
         print(f"Health.Check failed: code={e.code()}, details={e.details()}")
         return False
-    print(str(response).strip())
-    return response.status == 1  # SERVING
+    print(str(reply).strip())
+    return reply.status == 1  # SERVING
 
 
 def parse_args() -> argparse.Namespace:
-    parser = argparse.ArgumentParser()
-    parser.formatter_class = argparse.ArgumentDefaultsHelpFormatter
-    group = parser.add_mutually_exclusive_group(required=False)
-    group.add_argument("--insecure", dest="insecure", action="store_true",
-                       help="Use an insecure connection")
-    group.add_argument("--secure", dest="secure", action="store_true",
-                       help="Use a secure connection")
-    group.set_defaults(insecure=True, secure=False)
-    parser.add_argument("--server-url", type=str, default="localhost:8033",
-                        help="grpc server url (`host:port`)")
-    parser.add_argument("--timeout", type=float, default=1,
-                        help="Timeout for healthcheck request")
-    parser.add_argument("--service-name", type=str, required=False,
-                        default="fmaas.GenerationService",
-                        help="Name of the service to check")
-    return parser.parse_args()
+    ap = argparse.ArgumentParser(
+        formatter_class=argparse.ArgumentDefaultsHelpFormatter)
+    tls = ap.add_mutually_exclusive_group(required=False)
+    tls.add_argument("--insecure", dest="insecure", action="store_true",
+                     help="Use an insecure connection")
+    tls.add_argument("--secure", dest="secure", action="store_true",
+                     help="Use a secure connection")
+    tls.set_defaults(insecure=True, secure=False)
+    ap.add_argument("--server-url", type=str, default="localhost:8033",
+                    help="grpc server url (`host:port`)")
+    ap.add_argument("--timeout", type=float, default=1,
+                    help="Timeout for healthcheck request")
+    ap.add_argument("--service-name", type=str, required=False,
+                    default="fmaas.GenerationService",
+                    help="Name of the service to check")
+    return ap.parse_args()
 
 
 def cli() -> None:
