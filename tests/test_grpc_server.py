@@ -390,3 +390,54 @@ def test_lora_unique_ids_start_at_1000001():
     assert "lora_request" in kwargs
     # the unique-id counter advanced exactly once from its 1000001 start
     assert store.next_unique_id == 1000002
+
+
+def test_rpc_guard_engine_death_and_oom():
+    """Unit coverage of the per-RPC error hook (reference analog:
+    tests/test_grpc_server.py DummyEngine tests of _handle_exception):
+    dead engine trips the server stop event; GPU OOM aborts the RPC with
+    RESOURCE_EXHAUSTED; a healthy-engine failure just re-raises."""
+    import asyncio
+
+    from types import SimpleNamespace
+
+    from torch.cuda import OutOfMemoryError
+
+    from vllm_tgis_adapter_amd.grpc.service import rpc_guard
+
+    class Ctx:
+        def __init__(self):
+            self.aborted = None
+
+        async def abort(self, code, details):
+            self.aborted = (code, details)
+            raise grpc.aio.AbortError()
+
+    def make_service(errored):
+        return SimpleNamespace(
+            engine=SimpleNamespace(errored=errored, is_running=not errored),
+            stop_event=asyncio.Event(),
+        )
+
+    @rpc_guard
+    async def boom(self, request, context):
+        raise request  # the "request" carries the exception to raise
+
+    # dead engine -> stop event set, original exception re-raised
+    svc = make_service(errored=True)
+    with pytest.raises(RuntimeError):
+        asyncio.run(boom(svc, RuntimeError("engine died"), Ctx()))
+    assert svc.stop_event.is_set()
+
+    # healthy engine -> re-raised, stop event untouched
+    svc = make_service(errored=False)
+    with pytest.raises(ValueError):
+        asyncio.run(boom(svc, ValueError("bad"), Ctx()))
+    assert not svc.stop_event.is_set()
+
+    # GPU OOM -> abort(RESOURCE_EXHAUSTED)
+    svc = make_service(errored=False)
+    ctx = Ctx()
+    with pytest.raises(grpc.aio.AbortError):
+        asyncio.run(boom(svc, OutOfMemoryError("oom"), ctx))
+    assert ctx.aborted[0] == grpc.StatusCode.RESOURCE_EXHAUSTED
