@@ -280,3 +280,27 @@ def test_ep_moe_matches_dense():
     results = _run_spawn(_ep_moe_worker, 2)
     for rank, payload in results.items():
         assert payload["ok"], (rank, payload)
+
+
+def _engine_tp_pipelined_worker(rank, world, port, q):
+    """Same as _engine_tp_worker but with pipelined stepping forced on
+    (VTA_PIPELINE_MIN=1) — the combination the driver's multi-GPU scale
+    bench runs (batch 512 >= the default engage threshold with TP>1)."""
+    import os
+
+    os.environ["VTA_PIPELINE"] = "1"
+    os.environ["VTA_PIPELINE_MIN"] = "1"
+    try:
+        _engine_tp_worker(rank, world, port, q)
+    finally:
+        os.environ.pop("VTA_PIPELINE", None)
+        os.environ.pop("VTA_PIPELINE_MIN", None)
+
+
+def test_tp2_pipelined_engine_matches_tp1():
+    """TP-2 sharded engine with pipelined stepping == the TP-1 sync
+    reference (gloo world-2; broadcast/launch/drain interleaving across
+    ranks must not change outputs)."""
+    ref = _engine_tp1_reference()
+    res = _run_spawn(_engine_tp_pipelined_worker, 2, timeout=600)
+    assert res[0] == ref, (res[0], ref)
