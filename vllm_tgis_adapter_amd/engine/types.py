@@ -9,8 +9,8 @@ the MI355X-native engine.
 from __future__ import annotations
 
 import enum
-from dataclasses import dataclass, field
-from typing import Any, Callable, Optional, Sequence, Union
+from dataclasses import dataclass
+from typing import Any, Callable, Optional, Union
 
 
 class RequestOutputKind(enum.Enum):

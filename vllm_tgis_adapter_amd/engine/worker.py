@@ -21,7 +21,6 @@ from ..parallel import get_tp_rank, get_tp_world_size, tp_broadcast_object
 from . import roctx
 from .config import EngineConfig
 from .metadata import AttnMetadata
-from .request import Request
 from .sampler import Sampler, SamplerOutput, prompt_logprob_dicts
 from .scheduler import ScheduledItem, SchedulerOutput
 
